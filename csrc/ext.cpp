@@ -1,0 +1,282 @@
+// Torch extension bindings for the dlaf_amd CDNA4 kernels.
+//
+// All entry points take CUDA (ROCm) tensors, run on the CURRENT torch stream
+// (so the Python runtime's stream/event scheduling applies), and are
+// asynchronous. The per-tile Cholesky (`potrf_tile`) is host-orchestrated here
+// in C++: potrf_block (single-WG LDS kernel) + column-parallel block inverse +
+// fused-GEMM panel/trailing updates, with all GemmDescs staged to the device in
+// ONE H2D copy per call.
+#include <torch/extension.h>
+#include <c10/cuda/CUDAStream.h>
+#include <c10/cuda/CUDACachingAllocator.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+#include "kernels.h"
+
+namespace {
+
+#define HIP_CHECK(expr)                                                     \
+  do {                                                                      \
+    hipError_t _e = (expr);                                                 \
+    TORCH_CHECK(_e == hipSuccess, "HIP error: ", hipGetErrorString(_e));    \
+  } while (0)
+
+hipStream_t cur_stream() {
+  return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+}
+
+int potrf_bsz(at::ScalarType t) {
+  return (t == at::kComplexDouble || t == at::kComplexFloat) ? 64 : 128;
+}
+
+void check_gemm_args(const torch::Tensor& desc, const torch::Tensor& A,
+                     const torch::Tensor& B, const torch::Tensor& C) {
+  TORCH_CHECK(C.is_cuda() && A.is_cuda() && B.is_cuda(), "tensors must be on GPU");
+  TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == at::kLong && desc.is_contiguous(),
+              "desc must be contiguous int64 on GPU");
+  TORCH_CHECK(desc.dim() == 2 && desc.size(1) == 6, "desc must be [n, 6]");
+  TORCH_CHECK(A.scalar_type() == C.scalar_type() && B.scalar_type() == C.scalar_type(),
+              "dtype mismatch");
+}
+
+// C[desc] = alpha * op(A) op(B) + beta * C ; desc rows are
+// (c_off, a_off, b_off, ktiles, a_kstride, b_kstride) in element units.
+void batch_gemm(torch::Tensor C, torch::Tensor A, torch::Tensor B,
+                torch::Tensor desc, int64_t M, int64_t N, int64_t K,
+                int64_t lda, int64_t ldb, int64_t ldc, int64_t opA, int64_t opB,
+                double alpha_re, double alpha_im, double beta_re,
+                double beta_im) {
+  check_gemm_args(desc, A, B, C);
+  const int nd = (int)desc.size(0);
+  if (nd == 0) return;
+  auto descs = reinterpret_cast<const GemmDesc*>(desc.data_ptr<int64_t>());
+  auto s = cur_stream();
+  switch (C.scalar_type()) {
+    case at::kDouble:
+      gemm_tiles_f64(descs, nd, A.data_ptr<double>(), B.data_ptr<double>(),
+                     C.data_ptr<double>(), M, N, K, lda, ldb, ldc, opA, opB,
+                     alpha_re, beta_re, s);
+      break;
+    case at::kFloat:
+      gemm_tiles_f32(descs, nd, A.data_ptr<float>(), B.data_ptr<float>(),
+                     C.data_ptr<float>(), M, N, K, lda, ldb, ldc, opA, opB,
+                     (float)alpha_re, (float)beta_re, s);
+      break;
+    case at::kComplexDouble:
+      gemm_tiles_c128(descs, nd, (const double*)A.data_ptr(),
+                      (const double*)B.data_ptr(), (double*)C.data_ptr(), M, N,
+                      K, lda, ldb, ldc, opA, opB, alpha_re, alpha_im, beta_re,
+                      beta_im, s);
+      break;
+    case at::kComplexFloat:
+      gemm_tiles_c64(descs, nd, (const float*)A.data_ptr(),
+                     (const float*)B.data_ptr(), (float*)C.data_ptr(), M, N, K,
+                     lda, ldb, ldc, opA, opB, (float)alpha_re, (float)alpha_im,
+                     (float)beta_re, (float)beta_im, s);
+      break;
+    default:
+      TORCH_CHECK(false, "unsupported dtype");
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+void potrf_block(torch::Tensor A, int64_t n, int64_t ld) {
+  TORCH_CHECK(A.is_cuda());
+  auto s = cur_stream();
+  switch (A.scalar_type()) {
+    case at::kDouble:
+      TORCH_CHECK(n <= 128);
+      potrf_block128_f64(A.data_ptr<double>(), n, ld, s);
+      break;
+    case at::kFloat:
+      TORCH_CHECK(n <= 128);
+      potrf_block128_f32(A.data_ptr<float>(), n, ld, s);
+      break;
+    case at::kComplexDouble:
+      TORCH_CHECK(n <= 64);
+      potrf_block128_c128((double*)A.data_ptr(), n, ld, s);
+      break;
+    case at::kComplexFloat:
+      TORCH_CHECK(n <= 64);
+      potrf_block128_c64((float*)A.data_ptr(), n, ld, s);
+      break;
+    default:
+      TORCH_CHECK(false, "unsupported dtype");
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+// T = tril(L)^-1 for an n x n view; T must not alias L.
+void trtri_lower(torch::Tensor L, torch::Tensor T, int64_t n, int64_t ldl,
+                 int64_t ldt, bool unit_diag) {
+  TORCH_CHECK(L.is_cuda() && T.is_cuda());
+  auto s = cur_stream();
+  switch (L.scalar_type()) {
+    case at::kDouble:
+      trtri_lower_f64(L.data_ptr<double>(), T.data_ptr<double>(), n, ldl, ldt,
+                      unit_diag, s);
+      break;
+    case at::kFloat:
+      trtri_lower_f32(L.data_ptr<float>(), T.data_ptr<float>(), n, ldl, ldt,
+                      unit_diag, s);
+      break;
+    case at::kComplexDouble:
+      trtri_lower_c128((const double*)L.data_ptr(), (double*)T.data_ptr(), n,
+                       ldl, ldt, unit_diag, s);
+      break;
+    case at::kComplexFloat:
+      trtri_lower_c64((const float*)L.data_ptr(), (float*)T.data_ptr(), n, ldl,
+                      ldt, unit_diag, s);
+      break;
+    default:
+      TORCH_CHECK(false, "unsupported dtype");
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+// In-place Cholesky (Lower) of the leading n x n of a tile with row stride ld.
+// Also fills `dinv` ([nblocks, bsz, bsz] contiguous) with the inverses of the
+// bsz x bsz diagonal blocks of the factor — the panel TRSM then becomes GEMMs.
+void potrf_tile(torch::Tensor A, int64_t n, int64_t ld, torch::Tensor dinv) {
+  TORCH_CHECK(A.is_cuda() && dinv.is_cuda() && dinv.is_contiguous());
+  const int bsz = potrf_bsz(A.scalar_type());
+  const int nblocks = (int)((n + bsz - 1) / bsz);
+  TORCH_CHECK(dinv.numel() >= (int64_t)nblocks * bsz * bsz, "dinv too small");
+  auto s = cur_stream();
+  const bool cplx = A.is_complex();
+  const int opConj = cplx ? OP_C : OP_T;
+
+  // stage all descs in one upload: 2 per diagonal block (panel, trailing)
+  std::vector<int64_t> hdesc;
+  hdesc.reserve(nblocks * 12);
+  for (int d = 0; d < nblocks; ++d) {
+    // desc offsets are relative to per-launch base pointers; we use base = A
+    // for C and A operands, base = dinv block for B.
+    const int64_t c0 = (int64_t)d * bsz;
+    const int64_t panel_off = (c0 + bsz) * ld + c0;  // A21 block
+    const int64_t trail_off = (c0 + bsz) * ld + (c0 + bsz);
+    // panel: X = A21 * dinv^H (in-place; N <= BN so single column block)
+    hdesc.insert(hdesc.end(), {panel_off, panel_off, 0, 1, 0, 0});
+    // trailing: A22 -= X X^H
+    hdesc.insert(hdesc.end(), {trail_off, panel_off, panel_off, 1, 0, 0});
+  }
+  auto opts = torch::TensorOptions().dtype(at::kLong).device(A.device());
+  auto ddesc = torch::empty({(int64_t)hdesc.size()}, opts);
+  HIP_CHECK(hipMemcpyAsync(ddesc.data_ptr<int64_t>(), hdesc.data(),
+                           hdesc.size() * sizeof(int64_t), hipMemcpyHostToDevice,
+                           s));
+  auto descs = reinterpret_cast<const GemmDesc*>(ddesc.data_ptr<int64_t>());
+
+  for (int d = 0; d < nblocks; ++d) {
+    const int64_t c0 = (int64_t)d * bsz;
+    const int bs = (int)std::min<int64_t>(bsz, n - c0);
+    // 1) factor diagonal block
+    {
+      const int64_t off = c0 * ld + c0;
+      switch (A.scalar_type()) {
+        case at::kDouble:
+          potrf_block128_f64(A.data_ptr<double>() + off, bs, ld, s);
+          break;
+        case at::kFloat:
+          potrf_block128_f32(A.data_ptr<float>() + off, bs, ld, s);
+          break;
+        case at::kComplexDouble:
+          potrf_block128_c128((double*)A.data_ptr() + 2 * off, bs, ld, s);
+          break;
+        case at::kComplexFloat:
+          potrf_block128_c64((float*)A.data_ptr() + 2 * off, bs, ld, s);
+          break;
+        default:
+          TORCH_CHECK(false);
+      }
+    }
+    // 2) inverse of the diagonal block -> dinv[d]
+    {
+      const int64_t off = c0 * ld + c0;
+      const int64_t doff = (int64_t)d * bsz * bsz;
+      switch (A.scalar_type()) {
+        case at::kDouble:
+          trtri_lower_f64(A.data_ptr<double>() + off,
+                          dinv.data_ptr<double>() + doff, bs, ld, bsz, 0, s);
+          break;
+        case at::kFloat:
+          trtri_lower_f32(A.data_ptr<float>() + off,
+                          dinv.data_ptr<float>() + doff, bs, ld, bsz, 0, s);
+          break;
+        case at::kComplexDouble:
+          trtri_lower_c128((double*)A.data_ptr() + 2 * off,
+                           (double*)dinv.data_ptr() + 2 * doff, bs, ld, bsz, 0,
+                           s);
+          break;
+        case at::kComplexFloat:
+          trtri_lower_c64((float*)A.data_ptr() + 2 * off,
+                          (float*)dinv.data_ptr() + 2 * doff, bs, ld, bsz, 0, s);
+          break;
+        default:
+          TORCH_CHECK(false);
+      }
+    }
+    const int64_t rows_below = n - c0 - bs;
+    if (rows_below <= 0) continue;
+    const GemmDesc* dp = descs + 2 * d;
+    const int64_t doff = (int64_t)d * bsz * bsz;
+    // 3) panel: X = A21 * dinv^H  (in place)
+    // 4) trailing: A22 -= X X^H
+    switch (A.scalar_type()) {
+      case at::kDouble: {
+        auto Ap = A.data_ptr<double>();
+        auto Dp = dinv.data_ptr<double>() + doff;
+        gemm_tiles_f64(dp, 1, Ap, Dp, Ap, rows_below, bs, bs, ld, bsz, ld, OP_N,
+                       OP_T, 1.0, 0.0, s);
+        gemm_tiles_f64(dp + 1, 1, Ap, Ap, Ap, rows_below, rows_below, bs, ld,
+                       ld, ld, OP_N, OP_T, -1.0, 1.0, s);
+        break;
+      }
+      case at::kFloat: {
+        auto Ap = A.data_ptr<float>();
+        auto Dp = dinv.data_ptr<float>() + doff;
+        gemm_tiles_f32(dp, 1, Ap, Dp, Ap, rows_below, bs, bs, ld, bsz, ld, OP_N,
+                       OP_T, 1.0f, 0.0f, s);
+        gemm_tiles_f32(dp + 1, 1, Ap, Ap, Ap, rows_below, rows_below, bs, ld,
+                       ld, ld, OP_N, OP_T, -1.0f, 1.0f, s);
+        break;
+      }
+      case at::kComplexDouble: {
+        auto Ap = (double*)A.data_ptr();
+        auto Dp = (double*)dinv.data_ptr() + 2 * doff;
+        gemm_tiles_c128(dp, 1, Ap, Dp, Ap, rows_below, bs, bs, ld, bsz, ld,
+                        OP_N, OP_C, 1.0, 0.0, 0.0, 0.0, s);
+        gemm_tiles_c128(dp + 1, 1, Ap, Ap, Ap, rows_below, rows_below, bs, ld,
+                        ld, ld, OP_N, OP_C, -1.0, 0.0, 1.0, 0.0, s);
+        break;
+      }
+      case at::kComplexFloat: {
+        auto Ap = (float*)A.data_ptr();
+        auto Dp = (float*)dinv.data_ptr() + 2 * doff;
+        gemm_tiles_c64(dp, 1, Ap, Dp, Ap, rows_below, bs, bs, ld, bsz, ld, OP_N,
+                       OP_C, 1.0f, 0.0f, 0.0f, 0.0f, s);
+        gemm_tiles_c64(dp + 1, 1, Ap, Ap, Ap, rows_below, rows_below, bs, ld,
+                       ld, ld, OP_N, OP_C, -1.0f, 0.0f, 1.0f, 0.0f, s);
+        break;
+      }
+      default:
+        TORCH_CHECK(false);
+    }
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("batch_gemm", &batch_gemm,
+        "fused batched tile GEMM: C[d] = alpha*op(A[d])op(B[d]) + beta*C[d]");
+  m.def("potrf_block", &potrf_block, "single-workgroup Cholesky block factor");
+  m.def("trtri_lower", &trtri_lower, "lower-triangular block inverse");
+  m.def("potrf_tile", &potrf_tile,
+        "in-place tile Cholesky + diagonal-block inverses");
+  m.attr("POTRF_BSZ_REAL") = 128;
+  m.attr("POTRF_BSZ_CPLX") = 64;
+}

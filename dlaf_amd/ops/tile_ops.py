@@ -1,0 +1,184 @@
+"""Mid-level tile operations with CPU (torch) / GPU (native HIP) dispatch.
+
+The GPU path is descriptor-driven and FUSED: one call covers a list of tile
+triples and becomes one kernel launch (csrc/gemm_tiles.hip). The CPU path loops
+over tiles with torch.linalg — it is the reference backend (the analog of the
+reference's Backend::MC per-tile blaspp/lapackpp calls) and the
+oracle the GPU kernels are tested against.
+
+Conventions: tiles are row-major views; ``op`` follows dlaf_amd.types.Op;
+for real dtypes ConjTrans == Trans. Triangular solves on GPU are performed
+against block inverses (see csrc/factor.hip docstring).
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from ..types import Op, is_complex
+from ._ext import get_ext
+
+_OP_CODE = {Op.NoTrans: 0, Op.Trans: 1, Op.ConjTrans: 2}
+
+
+def _opc(op: Op) -> int:
+    return _OP_CODE[op]
+
+
+def _alpha_parts(alpha) -> Tuple[float, float]:
+    c = complex(alpha)
+    return (c.real, c.imag)
+
+
+def make_descs(
+    c_offs: Sequence[int],
+    a_offs: Sequence[int],
+    b_offs: Sequence[int],
+    ktiles: int = 1,
+    a_kstride: int = 0,
+    b_kstride: int = 0,
+) -> np.ndarray:
+    n = len(c_offs)
+    d = np.empty((n, 6), dtype=np.int64)
+    d[:, 0] = np.asarray(c_offs, dtype=np.int64)
+    d[:, 1] = np.asarray(a_offs, dtype=np.int64)
+    d[:, 2] = np.asarray(b_offs, dtype=np.int64)
+    d[:, 3] = ktiles
+    d[:, 4] = a_kstride
+    d[:, 5] = b_kstride
+    return d
+
+
+def gemm_fused(
+    C_base: torch.Tensor,
+    A_base: torch.Tensor,
+    B_base: torch.Tensor,
+    descs,
+    M: int,
+    N: int,
+    K: int,
+    lda: int,
+    ldb: int,
+    ldc: int,
+    opA: Op,
+    opB: Op,
+    alpha,
+    beta,
+) -> None:
+    """GPU fused batched GEMM. ``descs``: np.ndarray [n,6] or device tensor."""
+    ext = get_ext()
+    if isinstance(descs, np.ndarray):
+        descs = torch.from_numpy(descs).to(C_base.device, non_blocking=True)
+    ar, ai = _alpha_parts(alpha)
+    br, bi = _alpha_parts(beta)
+    ext.batch_gemm(
+        C_base.view(-1), A_base.view(-1), B_base.view(-1), descs,
+        M, N, K, lda, ldb, ldc, _opc(opA), _opc(opB), ar, ai, br, bi,
+    )
+
+
+def _t(x: torch.Tensor, op: Op) -> torch.Tensor:
+    if op is Op.NoTrans:
+        return x
+    if op is Op.Trans:
+        return x.mT
+    return x.mH
+
+
+def gemm_tile(C: torch.Tensor, A: torch.Tensor, B: torch.Tensor, opA: Op, opB: Op, alpha, beta) -> None:
+    """CPU single-tile GEMM: C = alpha*op(A)op(B) + beta*C."""
+    prod = _t(A, opA) @ _t(B, opB)
+    if beta == 0:
+        C.copy_(alpha * prod)
+    else:
+        C.mul_(beta).add_(prod, alpha=alpha)
+
+
+# ---------------- factorization tile ops ----------------
+
+def potrf_bsz(dtype: torch.dtype) -> int:
+    return 64 if is_complex(dtype) else 128
+
+
+def dinv_workspace(nb: int, dtype: torch.dtype, device) -> torch.Tensor:
+    bsz = potrf_bsz(dtype)
+    nblocks = (nb + bsz - 1) // bsz
+    return torch.empty((nblocks, bsz, bsz), dtype=dtype, device=device)
+
+
+def potrf_tile(tile: torch.Tensor, dinv: Optional[torch.Tensor] = None) -> Optional[torch.Tensor]:
+    """In-place lower Cholesky of a padded tile.
+
+    On GPU also fills/returns ``dinv`` (diagonal-block inverses) for the panel
+    TRSM-as-GEMM. On CPU returns None (panel solve uses solve_triangular).
+    """
+    n = tile.shape[0]
+    assert tile.shape[0] == tile.shape[1] and tile.stride(1) == 1
+    if tile.is_cuda:
+        if dinv is None:
+            dinv = dinv_workspace(n, tile.dtype, tile.device)
+        get_ext().potrf_tile(tile, n, tile.stride(0), dinv)
+        return dinv
+    L = torch.linalg.cholesky(tile)
+    tile.copy_(L)
+    return None
+
+
+def trtri_tile(L: torch.Tensor, out: torch.Tensor, unit_diag: bool = False) -> None:
+    """out = tril(L)^-1 (full tile written: upper part zeroed)."""
+    n = L.shape[0]
+    if L.is_cuda:
+        get_ext().trtri_lower(L, out, n, L.stride(0), out.stride(0), unit_diag)
+        return
+    eye = torch.eye(n, dtype=L.dtype, device=L.device)
+    T = torch.linalg.solve_triangular(
+        torch.tril(L) if not unit_diag else torch.tril(L, -1) + eye,
+        eye, upper=False, unitriangular=unit_diag,
+    )
+    out.copy_(torch.tril(T))
+
+
+def trsm_panel_right_lowerH(
+    panel_base: torch.Tensor,
+    tile_offs: Sequence[int],
+    L_diag: torch.Tensor,
+    dinv: torch.Tensor,
+    mb: int,
+    nb: int,
+    ld: int,
+) -> None:
+    """Solve X * op(L)^H = X for every panel tile in place (L lower-triangular).
+
+    GPU blocked algorithm per inner block d (bsz = dinv block size):
+        X[:, d] -= X[:, :d] @ L[d, :d]^H        (fused over all panel tiles)
+        X[:, d]  = X[:, d] @ dinv[d]^H          (fused, in place)
+    This is the panel step of right-looking Cholesky (reference
+    ``factorization/cholesky/impl.h:151-189`` trsmPanelTile).
+    """
+    assert panel_base.is_cuda
+    opc = Op.ConjTrans if is_complex(panel_base.dtype) else Op.Trans
+    bsz = dinv.shape[-1]
+    nblocks = (nb + bsz - 1) // bsz
+    offs = np.asarray(tile_offs, dtype=np.int64)
+    ld_l = L_diag.stride(0)
+    for d in range(nblocks):
+        c0 = d * bsz
+        bs = min(bsz, nb - c0)
+        if d > 0:
+            # X[:, c0:c0+bs] -= X[:, :c0] @ (L[c0:c0+bs, :c0])^H
+            descs = make_descs(offs + c0, offs, [c0 * ld_l] * len(offs))
+            gemm_fused(
+                panel_base, panel_base, L_diag, descs,
+                mb, bs, c0, ld, ld_l, ld,
+                Op.NoTrans, opc, -1.0, 1.0,
+            )
+        # X[:, c0:c0+bs] @= dinv[d]^H  (in place; single column block per WG)
+        descs = make_descs(offs + c0, offs + c0, [0] * len(offs))
+        gemm_fused(
+            panel_base, panel_base, dinv[d], descs,
+            mb, bs, bs, ld, bsz, ld,
+            Op.NoTrans, opc, 1.0, 0.0,
+        )
