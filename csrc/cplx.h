@@ -5,7 +5,7 @@
 template <typename T>
 struct cplx {
   T re, im;
-  __device__ __host__ cplx() : re(0), im(0) {}
+  cplx() = default;
   __device__ __host__ cplx(T r, T i) : re(r), im(i) {}
   __device__ inline cplx operator+(const cplx& o) const { return {re + o.re, im + o.im}; }
   __device__ inline cplx operator-(const cplx& o) const { return {re - o.re, im - o.im}; }

@@ -1,0 +1,157 @@
+#include "hip/hip_runtime.h"
+// Single-tile factorization building blocks for CDNA4.
+//
+// * potrf_block: single-workgroup LDS-resident Cholesky of a leading n x n
+//   block (n <= 128 real / n <= 64 complex, limited by 160 KiB LDS). The tile
+//   Cholesky of an nb x nb tile is orchestrated on the host (ext.cpp) as
+//   potrf_block + column-parallel block inverse + fused GEMM panel/trailing
+//   updates — the per-tile analog of the right-looking algorithm, replacing the
+//   reference's rocSOLVER potrf call.
+// * trtri_lower: column-parallel lower-triangular inverse (thread-per-column
+//   forward substitution). Triangular solves everywhere in the library are
+//   GEMMs against these block inverses — the standard GPU TRSM trade.
+#include "kernels.h"
+#include "cplx.h"
+
+namespace {
+
+template <typename S, int BSZ>
+__launch_bounds__(256) __global__ void potrf_block_k(S* A, int n, int ld) {
+  using TR = ScalarTraits<S>;
+  using RT = typename TR::real_t;
+  __shared__ S L[BSZ][BSZ + 1];
+  const int tid = threadIdx.x;
+
+  // load (full block; only the lower triangle is referenced)
+  for (int e = tid; e < BSZ * BSZ; e += 256) {
+    const int i = e / BSZ, j = e % BSZ;
+    L[i][j] = (i < n && j < n) ? A[(int64_t)i * ld + j] : TR::zero();
+  }
+  __syncthreads();
+
+  const int ncb = (n + 15) / 16;
+  for (int cb = 0; cb < ncb; ++cb) {
+    const int c0 = cb * 16;
+    const int bsz = min(16, n - c0);
+    // 1) factor the 16x16 diagonal block
+    for (int p = 0; p < bsz; ++p) {
+      if (tid == 0) {
+        RT d = TR::real(L[c0 + p][c0 + p]);
+        L[c0 + p][c0 + p] = TR::from_real(sqrt(d));
+      }
+      __syncthreads();
+      if (tid > p && tid < bsz) {
+        RT d = TR::real(L[c0 + p][c0 + p]);
+        L[c0 + tid][c0 + p] = L[c0 + tid][c0 + p] * (RT(1) / d);
+      }
+      __syncthreads();
+      {
+        const int i = tid / 16, j = tid % 16;
+        if (i < bsz && j < bsz && j > p && j <= i)
+          L[c0 + i][c0 + j] -= L[c0 + i][c0 + p] * TR::conj(L[c0 + j][c0 + p]);
+      }
+      __syncthreads();
+    }
+    // 2) panel solve: rows below vs the diagonal block (X * D^H = A)
+    {
+      const int r = c0 + 16 + tid;
+      if (r < n) {
+        for (int j = 0; j < bsz; ++j) {
+          S x = L[r][c0 + j];
+          for (int p = 0; p < j; ++p) x -= L[r][c0 + p] * TR::conj(L[c0 + j][c0 + p]);
+          L[r][c0 + j] = x * (RT(1) / TR::real(L[c0 + j][c0 + j]));
+        }
+      }
+      __syncthreads();
+    }
+    // 3) trailing update (lower part only)
+    {
+      const int t0 = c0 + 16;
+      const int nt = n - t0;
+      if (nt > 0) {
+        for (int e = tid; e < nt * nt; e += 256) {
+          const int i = e / nt, j = e % nt;
+          if (j <= i) {
+            S s = L[t0 + i][c0];
+            S acc = s * TR::conj(L[t0 + j][c0]);
+            for (int p = 1; p < bsz; ++p)
+              acc += L[t0 + i][c0 + p] * TR::conj(L[t0 + j][c0 + p]);
+            L[t0 + i][t0 + j] -= acc;
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // store lower triangle (incl. diagonal)
+  for (int e = tid; e < BSZ * BSZ; e += 256) {
+    const int i = e / BSZ, j = e % BSZ;
+    if (i < n && j < n && j <= i) A[(int64_t)i * ld + j] = L[i][j];
+  }
+}
+
+template <typename S>
+__global__ void trtri_lower_k(const S* L, S* T, int n, int ldl, int ldt,
+                              int unit_diag) {
+  using TR = ScalarTraits<S>;
+  const int j = blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= n) return;
+  // zero the upper part of column j (full tiles feed fused GEMMs)
+  for (int i = 0; i < j; ++i) T[(int64_t)i * ldt + j] = TR::zero();
+  const S djj = unit_diag ? TR::from_real(1) : TR::recip(L[(int64_t)j * ldl + j]);
+  T[(int64_t)j * ldt + j] = djj;
+  for (int i = j + 1; i < n; ++i) {
+    S acc = TR::zero();
+    for (int p = j; p < i; ++p)
+      acc += L[(int64_t)i * ldl + p] * T[(int64_t)p * ldt + j];
+    const S dii = unit_diag ? TR::from_real(1) : TR::recip(L[(int64_t)i * ldl + i]);
+    T[(int64_t)i * ldt + j] = -(dii * acc);
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+void potrf_block128_f64(double* A, int n, int ld, hipStream_t stream) {
+ hipLaunchKernelGGL(( potrf_block_k<double, 128>), dim3(1), dim3(256), 0, stream, A, n, ld);
+}
+void potrf_block128_f32(float* A, int n, int ld, hipStream_t stream) {
+ hipLaunchKernelGGL(( potrf_block_k<float, 128>), dim3(1), dim3(256), 0, stream, A, n, ld);
+}
+// complex: block size 64 (LDS limit); `A` is the interleaved base, ld in
+// complex elements.
+void potrf_block128_c128(double* A, int n, int ld, hipStream_t stream) {
+ hipLaunchKernelGGL(( potrf_block_k<cplx<double>, 64>)
+      , dim3(1), dim3(256), 0, stream, reinterpret_cast<cplx<double>*>(A), n, ld);
+}
+void potrf_block128_c64(float* A, int n, int ld, hipStream_t stream) {
+ hipLaunchKernelGGL(( potrf_block_k<cplx<float>, 64>)
+      , dim3(1), dim3(256), 0, stream, reinterpret_cast<cplx<float>*>(A), n, ld);
+}
+
+void trtri_lower_f64(const double* L, double* T, int n, int ldl, int ldt,
+                     int unit_diag, hipStream_t stream) {
+ hipLaunchKernelGGL(( trtri_lower_k<double>)
+      , dim3((n + 255) / 256), dim3(256), 0, stream, L, T, n, ldl, ldt, unit_diag);
+}
+void trtri_lower_f32(const float* L, float* T, int n, int ldl, int ldt,
+                     int unit_diag, hipStream_t stream) {
+ hipLaunchKernelGGL(( trtri_lower_k<float>)
+      , dim3((n + 255) / 256), dim3(256), 0, stream, L, T, n, ldl, ldt, unit_diag);
+}
+void trtri_lower_c128(const double* L, double* T, int n, int ldl, int ldt,
+                      int unit_diag, hipStream_t stream) {
+ hipLaunchKernelGGL(( trtri_lower_k<cplx<double>>), dim3((n + 255) / 256), dim3(256), 0, stream, 
+      reinterpret_cast<const cplx<double>*>(L), reinterpret_cast<cplx<double>*>(T),
+      n, ldl, ldt, unit_diag);
+}
+void trtri_lower_c64(const float* L, float* T, int n, int ldl, int ldt,
+                     int unit_diag, hipStream_t stream) {
+ hipLaunchKernelGGL(( trtri_lower_k<cplx<float>>), dim3((n + 255) / 256), dim3(256), 0, stream, 
+      reinterpret_cast<const cplx<float>*>(L), reinterpret_cast<cplx<float>*>(T),
+      n, ldl, ldt, unit_diag);
+}
+
+}  // extern "C"

@@ -1,0 +1,463 @@
+#include "hip/hip_runtime.h"
+// Fused batched tile GEMM for CDNA4 (gfx950) on fp64/fp32 MFMA.
+//
+// One kernel launch processes a LIST of tile-triples (GemmDesc): this is how a
+// whole trailing update (SYRK/HERK sweep), panel solve, or back-transform step
+// becomes a single launch with enough workgroups to fill 256 CUs. Each
+// workgroup computes one BMxBN block of one C tile:
+//   real    (f64/f32): BM=BN=128, BK=16, 4 waves, 64x64 per wave, 4x4 frags of
+//                      mfma_{f64,f32}_16x16x4 (exact fp64/fp32, the CDNA4
+//                      "SGEMM-class" MFMA at the vector-f64/f32 rate).
+//   complex (c128/c64): BM=BN=64, BK=16, 4 waves, 32x32 per wave, 2x2 frags,
+//                      4 MFMA per fragment pair (re/im cross terms).
+//
+// K-tiling pipeline: single LDS buffer; global loads of K-step s+1 issue while
+// MFMAs of step s run from LDS (register staging, write-after-barrier — the
+// pattern cdna_hip_programming.md §5 recommends when not using glds).
+//
+// Operand layout: row-major tiles. op(A) is resolved during LDS staging, so the
+// inner loop is layout-independent. Out-of-range rows/cols (edge blocks when
+// M,N,K are not multiples of the block sizes) stage zeros and stores are
+// guarded, so arbitrary sizes are supported.
+#include "kernels.h"
+
+typedef double v4d __attribute__((ext_vector_type(4)));
+typedef float v4f __attribute__((ext_vector_type(4)));
+
+namespace {
+
+template <typename T>
+struct Mfma;
+template <>
+struct Mfma<double> {
+  using acc_t = v4d;
+  static __device__ inline acc_t mma(double a, double b, acc_t c) {
+    return __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c, 0, 0, 0);
+  }
+};
+template <>
+struct Mfma<float> {
+  using acc_t = v4f;
+  static __device__ inline acc_t mma(float a, float b, acc_t c) {
+    return __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, c, 0, 0, 0);
+  }
+};
+
+// ---------------- real kernel ----------------
+
+template <typename T, int OPA, int OPB>
+__launch_bounds__(256) __global__ void gemm_tiles_k(
+    const GemmDesc* __restrict__ descs, const T* __restrict__ A,
+    const T* __restrict__ B, T* __restrict__ C, int M, int N, int K, int lda,
+    int ldb, int ldc, T alpha, T beta, int mblocks, int nblocks) {
+  constexpr int BM = 128, BN = 128, BK = 16;
+  constexpr int LA = BM * BK / 256;  // elements staged per thread (A)
+  constexpr int LB = BK * BN / 256;
+  using acc_t = typename Mfma<T>::acc_t;
+
+  __shared__ T As[BM][BK + 1];
+  __shared__ T Bs[BK][BN + 2];
+
+  const int wg = blockIdx.x;
+  const int per_desc = mblocks * nblocks;
+  const GemmDesc d = descs[wg / per_desc];
+  const int rem = wg % per_desc;
+  const int bi = rem / nblocks, bj = rem % nblocks;
+  const int i0 = bi * BM, j0 = bj * BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, w = tid >> 6;
+  const int wrow = (w >> 1) * 64, wcol = (w & 1) * 64;
+  const int li = lane & 15, lk = lane >> 4;
+
+  const int steps_per_tile = (K + BK - 1) / BK;
+  const int total_steps = (int)d.ktiles * steps_per_tile;
+
+  acc_t acc[4][4];
+#pragma unroll
+  for (int a = 0; a < 4; ++a)
+#pragma unroll
+    for (int b = 0; b < 4; ++b) acc[a][b] = {0, 0, 0, 0};
+
+  T ra[LA], rb[LB];
+
+  auto load_step = [&](int s, T* va, T* vb) {
+    const int kt = s / steps_per_tile;
+    const int k0 = (s % steps_per_tile) * BK;
+    const T* Ab = A + d.a_off + (int64_t)kt * d.a_kstride;
+    const T* Bb = B + d.b_off + (int64_t)kt * d.b_kstride;
+#pragma unroll
+    for (int j = 0; j < LA; ++j) {
+      const int e = j * 256 + tid;
+      int i, k;
+      if (OPA == OP_N) {
+        i = e / BK;
+        k = e % BK;
+      } else {
+        k = e / BM;
+        i = e % BM;
+      }
+      const int gi = i0 + i, gk = k0 + k;
+      T v = T(0);
+      if (gi < M && gk < K)
+        v = (OPA == OP_N) ? Ab[(int64_t)gi * lda + gk] : Ab[(int64_t)gk * lda + gi];
+      va[j] = v;
+    }
+#pragma unroll
+    for (int j = 0; j < LB; ++j) {
+      const int e = j * 256 + tid;
+      int k, c;
+      if (OPB == OP_N) {
+        k = e / BN;
+        c = e % BN;
+      } else {
+        c = e / BK;
+        k = e % BK;
+      }
+      const int gk = k0 + k, gc = j0 + c;
+      T v = T(0);
+      if (gk < K && gc < N)
+        v = (OPB == OP_N) ? Bb[(int64_t)gk * ldb + gc] : Bb[(int64_t)gc * ldb + gk];
+      vb[j] = v;
+    }
+  };
+
+  load_step(0, ra, rb);
+
+  for (int s = 0; s < total_steps; ++s) {
+    // regs -> LDS
+#pragma unroll
+    for (int j = 0; j < LA; ++j) {
+      const int e = j * 256 + tid;
+      int i, k;
+      if (OPA == OP_N) {
+        i = e / BK;
+        k = e % BK;
+      } else {
+        k = e / BM;
+        i = e % BM;
+      }
+      As[i][k] = ra[j];
+    }
+#pragma unroll
+    for (int j = 0; j < LB; ++j) {
+      const int e = j * 256 + tid;
+      int k, c;
+      if (OPB == OP_N) {
+        k = e / BN;
+        c = e % BN;
+      } else {
+        c = e / BK;
+        k = e % BK;
+      }
+      Bs[k][c] = rb[j];
+    }
+    __syncthreads();
+    if (s + 1 < total_steps) load_step(s + 1, ra, rb);  // overlaps MFMA below
+
+#pragma unroll
+    for (int ks = 0; ks < BK / 4; ++ks) {
+      T af[4], bf[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) af[mi] = As[wrow + mi * 16 + li][ks * 4 + lk];
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) bf[ni] = Bs[ks * 4 + lk][wcol + ni * 16 + li];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = Mfma<T>::mma(af[mi], bf[ni], acc[mi][ni]);
+    }
+    __syncthreads();  // LDS reuse barrier
+  }
+
+  // epilogue: C = alpha*acc + beta*C  (C/D map: col = lane&15, row = (lane>>4)*4+r)
+  T* Cb = C + d.c_off;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const acc_t v = acc[mi][ni];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = i0 + wrow + mi * 16 + lk * 4 + r;
+        const int col = j0 + wcol + ni * 16 + li;
+        if (row < M && col < N) {
+          const int64_t off = (int64_t)row * ldc + col;
+          T out = alpha * (T)v[r];
+          if (beta != T(0)) out += beta * Cb[off];
+          Cb[off] = out;
+        }
+      }
+    }
+}
+
+// ---------------- complex kernel ----------------
+// Interleaved (re, im); offsets/strides from the descriptor are in COMPLEX
+// elements. Conjugation (OP_C) is applied to the staged fragments.
+
+template <typename T, int OPA, int OPB>
+__launch_bounds__(256) __global__ void gemm_tiles_cplx_k(
+    const GemmDesc* __restrict__ descs, const T* __restrict__ A,
+    const T* __restrict__ B, T* __restrict__ C, int M, int N, int K, int lda,
+    int ldb, int ldc, T alpha_re, T alpha_im, T beta_re, T beta_im, int mblocks,
+    int nblocks) {
+  constexpr int BM = 64, BN = 64, BK = 16;
+  constexpr int LA = BM * BK / 256;  // complex elements per thread
+  constexpr int LB = BK * BN / 256;
+  using acc_t = typename Mfma<T>::acc_t;
+
+  __shared__ T Asr[BM][BK + 1];
+  __shared__ T Asi[BM][BK + 1];
+  __shared__ T Bsr[BK][BN + 2];
+  __shared__ T Bsi[BK][BN + 2];
+
+  const int wg = blockIdx.x;
+  const int per_desc = mblocks * nblocks;
+  const GemmDesc d = descs[wg / per_desc];
+  const int rem = wg % per_desc;
+  const int bi = rem / nblocks, bj = rem % nblocks;
+  const int i0 = bi * BM, j0 = bj * BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, w = tid >> 6;
+  const int wrow = (w >> 1) * 32, wcol = (w & 1) * 32;
+  const int li = lane & 15, lk = lane >> 4;
+
+  const int steps_per_tile = (K + BK - 1) / BK;
+  const int total_steps = (int)d.ktiles * steps_per_tile;
+
+  acc_t accr[2][2], acci[2][2];
+#pragma unroll
+  for (int a = 0; a < 2; ++a)
+#pragma unroll
+    for (int b = 0; b < 2; ++b) {
+      accr[a][b] = {0, 0, 0, 0};
+      acci[a][b] = {0, 0, 0, 0};
+    }
+
+  T rar[LA], rai[LA], rbr[LB], rbi[LB];
+
+  auto load_step = [&](int s) {
+    const int kt = s / steps_per_tile;
+    const int k0 = (s % steps_per_tile) * BK;
+    const T* Ab = A + 2 * (d.a_off + (int64_t)kt * d.a_kstride);
+    const T* Bb = B + 2 * (d.b_off + (int64_t)kt * d.b_kstride);
+#pragma unroll
+    for (int j = 0; j < LA; ++j) {
+      const int e = j * 256 + tid;
+      int i, k;
+      if (OPA == OP_N) {
+        i = e / BK;
+        k = e % BK;
+      } else {
+        k = e / BM;
+        i = e % BM;
+      }
+      const int gi = i0 + i, gk = k0 + k;
+      T vr = T(0), vi = T(0);
+      if (gi < M && gk < K) {
+        const int64_t off =
+            (OPA == OP_N) ? ((int64_t)gi * lda + gk) : ((int64_t)gk * lda + gi);
+        vr = Ab[2 * off];
+        vi = Ab[2 * off + 1];
+      }
+      if (OPA == OP_C) vi = -vi;
+      rar[j] = vr;
+      rai[j] = vi;
+    }
+#pragma unroll
+    for (int j = 0; j < LB; ++j) {
+      const int e = j * 256 + tid;
+      int k, c;
+      if (OPB == OP_N) {
+        k = e / BN;
+        c = e % BN;
+      } else {
+        c = e / BK;
+        k = e % BK;
+      }
+      const int gk = k0 + k, gc = j0 + c;
+      T vr = T(0), vi = T(0);
+      if (gk < K && gc < N) {
+        const int64_t off =
+            (OPB == OP_N) ? ((int64_t)gk * ldb + gc) : ((int64_t)gc * ldb + gk);
+        vr = Bb[2 * off];
+        vi = Bb[2 * off + 1];
+      }
+      if (OPB == OP_C) vi = -vi;
+      rbr[j] = vr;
+      rbi[j] = vi;
+    }
+  };
+
+  load_step(0);
+
+  for (int s = 0; s < total_steps; ++s) {
+#pragma unroll
+    for (int j = 0; j < LA; ++j) {
+      const int e = j * 256 + tid;
+      int i, k;
+      if (OPA == OP_N) {
+        i = e / BK;
+        k = e % BK;
+      } else {
+        k = e / BM;
+        i = e % BM;
+      }
+      Asr[i][k] = rar[j];
+      Asi[i][k] = rai[j];
+    }
+#pragma unroll
+    for (int j = 0; j < LB; ++j) {
+      const int e = j * 256 + tid;
+      int k, c;
+      if (OPB == OP_N) {
+        k = e / BN;
+        c = e % BN;
+      } else {
+        c = e / BK;
+        k = e % BK;
+      }
+      Bsr[k][c] = rbr[j];
+      Bsi[k][c] = rbi[j];
+    }
+    __syncthreads();
+    if (s + 1 < total_steps) load_step(s + 1);  // overlaps MFMA below
+
+#pragma unroll
+    for (int ks = 0; ks < BK / 4; ++ks) {
+      T ar[2], ai[2], br[2], bi_[2];
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        ar[mi] = Asr[wrow + mi * 16 + li][ks * 4 + lk];
+        ai[mi] = Asi[wrow + mi * 16 + li][ks * 4 + lk];
+      }
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        br[ni] = Bsr[ks * 4 + lk][wcol + ni * 16 + li];
+        bi_[ni] = Bsi[ks * 4 + lk][wcol + ni * 16 + li];
+      }
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          accr[mi][ni] = Mfma<T>::mma(ar[mi], br[ni], accr[mi][ni]);
+          accr[mi][ni] = Mfma<T>::mma(-ai[mi], bi_[ni], accr[mi][ni]);
+          acci[mi][ni] = Mfma<T>::mma(ar[mi], bi_[ni], acci[mi][ni]);
+          acci[mi][ni] = Mfma<T>::mma(ai[mi], br[ni], acci[mi][ni]);
+        }
+    }
+    __syncthreads();
+  }
+
+  T* Cb = C + 2 * d.c_off;
+  const bool beta0 = (beta_re == T(0)) && (beta_im == T(0));
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const acc_t vr = accr[mi][ni];
+      const acc_t vi = acci[mi][ni];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = i0 + wrow + mi * 16 + lk * 4 + r;
+        const int col = j0 + wcol + ni * 16 + li;
+        if (row < M && col < N) {
+          const int64_t off = 2 * ((int64_t)row * ldc + col);
+          T outr = alpha_re * (T)vr[r] - alpha_im * (T)vi[r];
+          T outi = alpha_re * (T)vi[r] + alpha_im * (T)vr[r];
+          if (!beta0) {
+            const T cr = Cb[off], ci = Cb[off + 1];
+            outr += beta_re * cr - beta_im * ci;
+            outi += beta_re * ci + beta_im * cr;
+          }
+          Cb[off] = outr;
+          Cb[off + 1] = outi;
+        }
+      }
+    }
+}
+
+template <typename T>
+void launch_real(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
+                 int M, int N, int K, int lda, int ldb, int ldc, int opA,
+                 int opB, T alpha, T beta, hipStream_t stream) {
+  if (ndesc <= 0 || M <= 0 || N <= 0) return;
+  const int mblocks = (M + 127) / 128, nblocks = (N + 127) / 128;
+  const dim3 grid(ndesc * mblocks * nblocks);
+  const dim3 block(256);
+  // OP_C == OP_T for real scalars
+  const int oa = (opA == OP_C) ? OP_T : opA;
+  const int ob = (opB == OP_C) ? OP_T : opB;
+#define CASE(OA, OB)                                                        \
+  if (oa == OA && ob == OB) {                                               \
+   hipLaunchKernelGGL(( gemm_tiles_k<T, OA, OB>), dim3(grid), dim3(block), 0, stream,                     \
+        descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks,       \
+        nblocks);                                                           \
+    return;                                                                 \
+  }
+  CASE(OP_N, OP_N) CASE(OP_N, OP_T) CASE(OP_T, OP_N) CASE(OP_T, OP_T)
+#undef CASE
+}
+
+template <typename T>
+void launch_cplx(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
+                 int M, int N, int K, int lda, int ldb, int ldc, int opA,
+                 int opB, T ar, T ai, T br, T bi, hipStream_t stream) {
+  if (ndesc <= 0 || M <= 0 || N <= 0) return;
+  const int mblocks = (M + 63) / 64, nblocks = (N + 63) / 64;
+  const dim3 grid(ndesc * mblocks * nblocks);
+  const dim3 block(256);
+#define CASE(OA, OB)                                                          \
+  if (opA == OA && opB == OB) {                                               \
+   hipLaunchKernelGGL(( gemm_tiles_cplx_k<T, OA, OB>), dim3(grid), dim3(block), 0, stream,                  \
+        descs, A, B, C, M, N, K, lda, ldb, ldc, ar, ai, br, bi, mblocks,      \
+        nblocks);                                                             \
+    return;                                                                   \
+  }
+  CASE(OP_N, OP_N) CASE(OP_N, OP_T) CASE(OP_N, OP_C)
+  CASE(OP_T, OP_N) CASE(OP_T, OP_T) CASE(OP_T, OP_C)
+  CASE(OP_C, OP_N) CASE(OP_C, OP_T) CASE(OP_C, OP_C)
+#undef CASE
+}
+
+}  // namespace
+
+extern "C" {
+
+void gemm_tiles_f64(const GemmDesc* descs, int ndesc, const double* A,
+                    const double* B, double* C, int M, int N, int K, int lda,
+                    int ldb, int ldc, int opA, int opB, double alpha,
+                    double beta, hipStream_t stream) {
+  launch_real<double>(descs, ndesc, A, B, C, M, N, K, lda, ldb, ldc, opA, opB,
+                      alpha, beta, stream);
+}
+
+void gemm_tiles_f32(const GemmDesc* descs, int ndesc, const float* A,
+                    const float* B, float* C, int M, int N, int K, int lda,
+                    int ldb, int ldc, int opA, int opB, float alpha, float beta,
+                    hipStream_t stream) {
+  launch_real<float>(descs, ndesc, A, B, C, M, N, K, lda, ldb, ldc, opA, opB,
+                     alpha, beta, stream);
+}
+
+void gemm_tiles_c128(const GemmDesc* descs, int ndesc, const double* A,
+                     const double* B, double* C, int M, int N, int K, int lda,
+                     int ldb, int ldc, int opA, int opB, double alpha_re,
+                     double alpha_im, double beta_re, double beta_im,
+                     hipStream_t stream) {
+  launch_cplx<double>(descs, ndesc, A, B, C, M, N, K, lda, ldb, ldc, opA, opB,
+                      alpha_re, alpha_im, beta_re, beta_im, stream);
+}
+
+void gemm_tiles_c64(const GemmDesc* descs, int ndesc, const float* A,
+                    const float* B, float* C, int M, int N, int K, int lda,
+                    int ldb, int ldc, int opA, int opB, float alpha_re,
+                    float alpha_im, float beta_re, float beta_im,
+                    hipStream_t stream) {
+  launch_cplx<float>(descs, ndesc, A, B, C, M, N, K, lda, ldb, ldc, opA, opB,
+                     alpha_re, alpha_im, beta_re, beta_im, stream);
+}
+
+}  // extern "C"

@@ -1,0 +1,38 @@
+"""End-to-end GPU Cholesky vs torch CPU reference."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from dlaf_amd import Matrix, UpLo, cholesky_factorization  # noqa: E402
+from dlaf_amd.matrix import util as mutil  # noqa: E402
+
+
+@pytest.mark.parametrize("dtype", [torch.float64, torch.complex128, torch.float32])
+@pytest.mark.parametrize("n,nb", [(512, 128), (1024, 256), (1000, 256), (768, 512)])
+def test_cholesky_local_gpu(dtype, n, nb):
+    mat = Matrix.create(n, n, nb, nb, dtype=dtype, device="cuda")
+    mutil.set_random_hermitian_positive_definite(mat, seed=7)
+    a_ref = mat.to_global().cpu().to(torch.complex128 if dtype.is_complex else torch.float64)
+    cholesky_factorization(UpLo.Lower, mat)
+    torch.cuda.synchronize()
+    got = torch.tril(mat.to_global().cpu().to(a_ref.dtype))
+    want = torch.linalg.cholesky(a_ref)
+    err = (got - want).abs().max().item()
+    scale = want.abs().max().item()
+    tol = 1e-3 if dtype in (torch.float32,) else 1e-9
+    assert err <= tol * scale * n, f"err={err} scale={scale}"
+
+
+def test_cholesky_residual_gpu():
+    """||A - L L^H|| / ||A|| residual check, the miniapp verification style."""
+    n, nb = 2048, 512
+    mat = Matrix.create(n, n, nb, nb, dtype=torch.float64, device="cuda")
+    mutil.set_random_hermitian_positive_definite(mat, seed=1)
+    a = mat.to_global()
+    cholesky_factorization(UpLo.Lower, mat)
+    torch.cuda.synchronize()
+    L = torch.tril(mat.to_global())
+    res = (a - L @ L.mH).abs().max().item() / a.abs().max().item()
+    assert res < 1e-13 * n, f"residual={res}"

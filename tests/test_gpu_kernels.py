@@ -1,0 +1,165 @@
+"""GPU kernel numerics tests: hand-written CDNA4 kernels vs torch CPU reference.
+
+Every HIP kernel is compared against a plain PyTorch reference of the same op
+computed in fp64 (or the op's own precision when that IS fp64).
+"""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from dlaf_amd.types import Op  # noqa: E402
+from dlaf_amd.ops import tile_ops as ops  # noqa: E402
+
+DTYPES = [torch.float64, torch.float32, torch.complex128, torch.complex64]
+
+
+def _tol(dtype, k=512):
+    if dtype in (torch.float32, torch.complex64):
+        return 1e-4 * max(1, k // 64)
+    return 1e-12 * max(1, k // 64)
+
+
+def _rand(shape, dtype, device="cuda"):
+    if dtype.is_complex:
+        rd = torch.float64 if dtype == torch.complex128 else torch.float32
+        return torch.complex(
+            torch.randn(shape, dtype=rd), torch.randn(shape, dtype=rd)
+        ).to(device=device, dtype=dtype)
+    return torch.randn(shape, dtype=dtype, device=device)
+
+
+def _ref_op(x, op):
+    if op is Op.NoTrans:
+        return x
+    if op is Op.Trans:
+        return x.mT
+    return x.mH
+
+
+@pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("opA", [Op.NoTrans, Op.Trans, Op.ConjTrans])
+@pytest.mark.parametrize("opB", [Op.NoTrans, Op.Trans, Op.ConjTrans])
+def test_batch_gemm_ops(dtype, opA, opB):
+    torch.manual_seed(0)
+    M, N, K = 256, 128, 192
+    # physical shapes depend on op
+    a_shape = (M, K) if opA is Op.NoTrans else (K, M)
+    b_shape = (K, N) if opB is Op.NoTrans else (N, K)
+    A = _rand(a_shape, dtype)
+    B = _rand(b_shape, dtype)
+    C = _rand((M, N), dtype)
+    C0 = C.clone()
+    alpha, beta = (1.5 - 0.5j, 0.25 + 1j) if dtype.is_complex else (1.5, 0.25)
+    descs = ops.make_descs([0], [0], [0])
+    ops.gemm_fused(C, A, B, descs, M, N, K, a_shape[1], b_shape[1], N, opA, opB, alpha, beta)
+    torch.cuda.synchronize()
+    ref = alpha * (_ref_op(A.cpu().to(torch.promote_types(dtype, torch.complex128 if dtype.is_complex else torch.float64)), opA)
+                   @ _ref_op(B.cpu().to(torch.promote_types(dtype, torch.complex128 if dtype.is_complex else torch.float64)), opB)) \
+        + beta * C0.cpu().to(torch.promote_types(dtype, torch.complex128 if dtype.is_complex else torch.float64))
+    err = (C.cpu().to(ref.dtype) - ref).abs().max().item()
+    scale = ref.abs().max().item() + 1
+    assert err <= _tol(dtype, K) * scale, f"err={err} scale={scale}"
+
+
+@pytest.mark.parametrize("dtype", [torch.float64, torch.complex128])
+@pytest.mark.parametrize("M,N,K", [(64, 64, 64), (100, 96, 120), (33, 17, 5), (512, 512, 512), (128, 64, 1)])
+def test_batch_gemm_shapes(dtype, M, N, K):
+    torch.manual_seed(1)
+    A = _rand((M, K), dtype)
+    B = _rand((K, N), dtype)
+    C = torch.zeros((M, N), dtype=dtype, device="cuda")
+    descs = ops.make_descs([0], [0], [0])
+    ops.gemm_fused(C, A, B, descs, M, N, K, K, N, N, Op.NoTrans, Op.NoTrans, 1.0, 0.0)
+    torch.cuda.synchronize()
+    ref = A.cpu().to(torch.complex128 if dtype.is_complex else torch.float64) @ \
+        B.cpu().to(torch.complex128 if dtype.is_complex else torch.float64)
+    err = (C.cpu().to(ref.dtype) - ref).abs().max().item()
+    assert err <= _tol(dtype, K) * (ref.abs().max().item() + 1), f"err={err}"
+
+
+def test_batch_gemm_multi_desc_and_ktiles():
+    torch.manual_seed(2)
+    nb = 64
+    nt = 3
+    # A: row of k-tiles, B: column of k-tiles, C: single tile accumulated over 2 k-tiles
+    A = torch.randn(nt, nb, nb, dtype=torch.float64, device="cuda")
+    B = torch.randn(nt, nb, nb, dtype=torch.float64, device="cuda")
+    C = torch.zeros(2, nb, nb, dtype=torch.float64, device="cuda")
+    ts = nb * nb
+    descs = ops.make_descs([0, ts], [0, ts], [0, ts], ktiles=2, a_kstride=ts, b_kstride=ts)
+    ops.gemm_fused(C, A, B, descs, nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, 1.0, 0.0)
+    torch.cuda.synchronize()
+    for d in range(2):
+        ref = A[d].cpu() @ B[d].cpu() + A[d + 1].cpu() @ B[d + 1].cpu()
+        err = (C[d].cpu() - ref).abs().max().item()
+        assert err < 1e-10, f"desc {d}: err={err}"
+
+
+@pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("n", [64, 128, 100, 512, 500])
+def test_potrf_tile(dtype, n):
+    if dtype.is_complex and n == 100:
+        n = 60  # complex block size is 64
+    torch.manual_seed(3)
+    a = _rand((n, n), dtype).cpu()
+    a = a @ a.mH + n * torch.eye(n, dtype=dtype)
+    tile = a.to("cuda")
+    dinv = ops.potrf_tile(tile)
+    torch.cuda.synchronize()
+    ref = torch.linalg.cholesky(a.to(torch.complex128 if dtype.is_complex else torch.float64))
+    got = torch.tril(tile.cpu().to(ref.dtype))
+    err = (got - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err <= _tol(dtype, n) * scale * 10, f"err={err}"
+    # dinv blocks invert the diagonal blocks of the factor
+    bsz = dinv.shape[-1]
+    d0 = dinv[0].cpu().to(ref.dtype)
+    bs = min(bsz, n)
+    prod = d0[:bs, :bs] @ ref[:bs, :bs]
+    err = (prod - torch.eye(bs, dtype=ref.dtype)).abs().max().item()
+    assert err <= _tol(dtype, n) * 100, f"dinv err={err}"
+
+
+@pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("n", [64, 128, 512, 300])
+def test_trtri_lower(dtype, n):
+    torch.manual_seed(4)
+    L = torch.tril(_rand((n, n), dtype)) + 2 * n ** 0.5 * torch.eye(n, dtype=dtype, device="cuda")
+    T = torch.empty_like(L)
+    ops.trtri_tile(L, T)
+    torch.cuda.synchronize()
+    rdt = torch.complex128 if dtype.is_complex else torch.float64
+    prod = T.cpu().to(rdt) @ L.cpu().to(rdt)
+    err = (prod - torch.eye(n, dtype=rdt)).abs().max().item()
+    assert err <= _tol(dtype, n) * 100, f"err={err}"
+
+
+@pytest.mark.parametrize("dtype", [torch.float64, torch.complex128, torch.float32])
+def test_trsm_panel(dtype):
+    torch.manual_seed(5)
+    nb = 256 if not dtype.is_complex else 128
+    ntiles = 3
+    L = torch.tril(_rand((nb, nb), dtype)) + 2 * nb * torch.eye(nb, dtype=dtype, device="cuda")
+    panel = _rand((ntiles, nb, nb), dtype)
+    ref_in = panel.clone().cpu()
+    # dinv of L
+    dinv = ops.dinv_workspace(nb, dtype, "cuda")
+    bsz = dinv.shape[-1]
+    ext = ops.get_ext()
+    for d in range((nb + bsz - 1) // bsz):
+        c0 = d * bsz
+        bs = min(bsz, nb - c0)
+        ext.trtri_lower(L[c0:, c0:], dinv[d], bs, L.stride(0), bsz, False)
+    offs = [i * nb * nb for i in range(ntiles)]
+    ops.trsm_panel_right_lowerH(panel, offs, L, dinv, nb, nb, nb)
+    torch.cuda.synchronize()
+    rdt = torch.complex128 if dtype.is_complex else torch.float64
+    Lh = L.cpu().to(rdt).mH
+    for i in range(ntiles):
+        ref = torch.linalg.solve_triangular(Lh, ref_in[i].to(rdt), upper=True, left=False)
+        err = (panel[i].cpu().to(rdt) - ref).abs().max().item()
+        scale = ref.abs().max().item() + 1
+        assert err <= _tol(dtype, nb) * scale * 100, f"tile {i} err={err}"
