@@ -1,0 +1,78 @@
+#!/usr/bin/env python3
+"""Kernel microbenchmarks on one MI355X: fused GEMM TF/s, potrf/trtri latency."""
+
+import time
+
+import torch
+
+from dlaf_amd.ops import tile_ops as ops
+from dlaf_amd.types import Op
+
+
+def bench_gemm(dtype=torch.float64, nb=512, ntiles=128, iters=10, opB=Op.Trans):
+    dev = "cuda"
+    A = torch.randn(ntiles, nb, nb, dtype=dtype, device=dev) if not dtype.is_complex else (
+        torch.randn(ntiles, nb, nb, dtype=torch.float64, device=dev)
+        + 1j * torch.randn(ntiles, nb, nb, dtype=torch.float64, device=dev)).to(dtype)
+    C = torch.zeros_like(A)
+    ts = nb * nb
+    offs = [i * ts for i in range(ntiles)]
+    descs = ops.make_descs(offs, offs, offs)
+    import numpy as np
+    dt = torch.from_numpy(descs).to(dev)
+    # warmup
+    for _ in range(3):
+        ops.gemm_fused(C, A, A, dt, nb, nb, nb, nb, nb, nb, Op.NoTrans, opB, -1.0, 1.0)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        ops.gemm_fused(C, A, A, dt, nb, nb, nb, nb, nb, nb, Op.NoTrans, opB, -1.0, 1.0)
+    torch.cuda.synchronize()
+    dt_s = (time.perf_counter() - t0) / iters
+    flops = 2.0 * ntiles * nb * nb * nb * (4 if dtype.is_complex else 1)
+    print(f"gemm {dtype} nb={nb} ntiles={ntiles} opB={opB}: {dt_s*1e3:.2f} ms  "
+          f"{flops/dt_s/1e12:.2f} TFLOP/s")
+
+
+def bench_potrf(dtype=torch.float64, nb=512, iters=20):
+    a = torch.randn(nb, nb, dtype=dtype, device="cuda")
+    a = a @ a.mH + nb * torch.eye(nb, dtype=dtype, device="cuda")
+    t = a.clone()
+    dinv = ops.dinv_workspace(nb, dtype, "cuda")
+    for _ in range(3):
+        t.copy_(a)
+        ops.potrf_tile(t, dinv)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        t.copy_(a)
+        ops.potrf_tile(t, dinv)
+    torch.cuda.synchronize()
+    dt_s = (time.perf_counter() - t0) / iters
+    print(f"potrf_tile {dtype} nb={nb}: {dt_s*1e6:.0f} us")
+
+
+def bench_trtri(dtype=torch.float64, nb=512, iters=20):
+    L = torch.tril(torch.randn(nb, nb, dtype=dtype, device="cuda")) + \
+        2 * nb * torch.eye(nb, dtype=dtype, device="cuda")
+    T = torch.empty_like(L)
+    for _ in range(3):
+        ops.trtri_tile(L, T)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        ops.trtri_tile(L, T)
+    torch.cuda.synchronize()
+    dt_s = (time.perf_counter() - t0) / iters
+    print(f"trtri_tile {dtype} nb={nb}: {dt_s*1e6:.0f} us")
+
+
+if __name__ == "__main__":
+    print(torch.cuda.get_device_name(0))
+    bench_gemm(torch.float64, 512, 128)
+    bench_gemm(torch.float64, 512, 1024, iters=5)
+    bench_gemm(torch.float32, 512, 128)
+    bench_gemm(torch.complex128, 512, 64)
+    bench_potrf(torch.float64)
+    bench_potrf(torch.complex128, 512)
+    bench_trtri(torch.float64)
