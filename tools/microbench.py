@@ -1,7 +1,11 @@
 #!/usr/bin/env python3
 """Kernel microbenchmarks on one MI355X: fused GEMM TF/s, potrf/trtri latency."""
 
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 
 import torch
 
