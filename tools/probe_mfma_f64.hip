@@ -23,8 +23,8 @@ __global__ void probe(const double* A, const double* B, double* Dout) {
 int main() {
   double hA[16 * 4], hB[4 * 16], hD[16 * 16], hRaw[64 * 4];
   srand(12345);
-  for (int i = 0; i < 64; ++i) hA[i] = (double)(rand() % 100000) / 7.0;
-  for (int i = 0; i < 64; ++i) hB[i] = (double)(rand() % 100000) / 11.0;
+  for (int i = 0; i < 64; ++i) hA[i] = (double)(rand() % 997);
+  for (int i = 0; i < 64; ++i) hB[i] = (double)(rand() % 983);
   for (int i = 0; i < 16; ++i)
     for (int j = 0; j < 16; ++j) {
       double s = 0;
