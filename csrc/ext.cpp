@@ -47,7 +47,7 @@ void batch_gemm(torch::Tensor C, torch::Tensor A, torch::Tensor B,
                 torch::Tensor desc, int64_t M, int64_t N, int64_t K,
                 int64_t lda, int64_t ldb, int64_t ldc, int64_t opA, int64_t opB,
                 double alpha_re, double alpha_im, double beta_re,
-                double beta_im) {
+                double beta_im, bool inplace) {
   check_gemm_args(desc, A, B, C);
   const int nd = (int)desc.size(0);
   if (nd == 0) return;
@@ -57,12 +57,12 @@ void batch_gemm(torch::Tensor C, torch::Tensor A, torch::Tensor B,
     case at::kDouble:
       gemm_tiles_f64(descs, nd, A.data_ptr<double>(), B.data_ptr<double>(),
                      C.data_ptr<double>(), M, N, K, lda, ldb, ldc, opA, opB,
-                     alpha_re, beta_re, s);
+                     alpha_re, beta_re, s, inplace);
       break;
     case at::kFloat:
       gemm_tiles_f32(descs, nd, A.data_ptr<float>(), B.data_ptr<float>(),
                      C.data_ptr<float>(), M, N, K, lda, ldb, ldc, opA, opB,
-                     (float)alpha_re, (float)beta_re, s);
+                     (float)alpha_re, (float)beta_re, s, inplace);
       break;
     case at::kComplexDouble:
       gemm_tiles_c128(descs, nd, (const double*)A.data_ptr(),
@@ -139,34 +139,19 @@ void trtri_lower(torch::Tensor L, torch::Tensor T, int64_t n, int64_t ldl,
 // In-place Cholesky (Lower) of the leading n x n of a tile with row stride ld.
 // Also fills `dinv` ([nblocks, bsz, bsz] contiguous) with the inverses of the
 // bsz x bsz diagonal blocks of the factor — the panel TRSM then becomes GEMMs.
-void potrf_tile(torch::Tensor A, int64_t n, int64_t ld, torch::Tensor dinv) {
+// `ddesc` is the prebuilt device descriptor table ([2*nblocks, 6] int64,
+// cached by the Python layer per (n, ld, dtype)): rows 2d   = panel desc
+// (A21 offsets), rows 2d+1 = trailing desc, exactly as _potrf_descs builds.
+void potrf_tile(torch::Tensor A, int64_t n, int64_t ld, torch::Tensor dinv,
+                torch::Tensor ddesc) {
   TORCH_CHECK(A.is_cuda() && dinv.is_cuda() && dinv.is_contiguous());
   const int bsz = potrf_bsz(A.scalar_type());
   const int nblocks = (int)((n + bsz - 1) / bsz);
   TORCH_CHECK(dinv.numel() >= (int64_t)nblocks * bsz * bsz, "dinv too small");
+  TORCH_CHECK(ddesc.is_cuda() && ddesc.scalar_type() == at::kLong &&
+              ddesc.is_contiguous() && ddesc.numel() >= nblocks * 12,
+              "bad ddesc");
   auto s = cur_stream();
-  const bool cplx = A.is_complex();
-  const int opConj = cplx ? OP_C : OP_T;
-
-  // stage all descs in one upload: 2 per diagonal block (panel, trailing)
-  std::vector<int64_t> hdesc;
-  hdesc.reserve(nblocks * 12);
-  for (int d = 0; d < nblocks; ++d) {
-    // desc offsets are relative to per-launch base pointers; we use base = A
-    // for C and A operands, base = dinv block for B.
-    const int64_t c0 = (int64_t)d * bsz;
-    const int64_t panel_off = (c0 + bsz) * ld + c0;  // A21 block
-    const int64_t trail_off = (c0 + bsz) * ld + (c0 + bsz);
-    // panel: X = A21 * dinv^H (in-place; N <= BN so single column block)
-    hdesc.insert(hdesc.end(), {panel_off, panel_off, 0, 1, 0, 0});
-    // trailing: A22 -= X X^H
-    hdesc.insert(hdesc.end(), {trail_off, panel_off, panel_off, 1, 0, 0});
-  }
-  auto opts = torch::TensorOptions().dtype(at::kLong).device(A.device());
-  auto ddesc = torch::empty({(int64_t)hdesc.size()}, opts);
-  HIP_CHECK(hipMemcpyAsync(ddesc.data_ptr<int64_t>(), hdesc.data(),
-                           hdesc.size() * sizeof(int64_t), hipMemcpyHostToDevice,
-                           s));
   auto descs = reinterpret_cast<const GemmDesc*>(ddesc.data_ptr<int64_t>());
 
   for (int d = 0; d < nblocks; ++d) {
@@ -229,18 +214,18 @@ void potrf_tile(torch::Tensor A, int64_t n, int64_t ld, torch::Tensor dinv) {
         auto Ap = A.data_ptr<double>();
         auto Dp = dinv.data_ptr<double>() + doff;
         gemm_tiles_f64(dp, 1, Ap, Dp, Ap, rows_below, bs, bs, ld, bsz, ld, OP_N,
-                       OP_T, 1.0, 0.0, s);
+                       OP_T, 1.0, 0.0, s, 1);
         gemm_tiles_f64(dp + 1, 1, Ap, Ap, Ap, rows_below, rows_below, bs, ld,
-                       ld, ld, OP_N, OP_T, -1.0, 1.0, s);
+                       ld, ld, OP_N, OP_T, -1.0, 1.0, s, 0);
         break;
       }
       case at::kFloat: {
         auto Ap = A.data_ptr<float>();
         auto Dp = dinv.data_ptr<float>() + doff;
         gemm_tiles_f32(dp, 1, Ap, Dp, Ap, rows_below, bs, bs, ld, bsz, ld, OP_N,
-                       OP_T, 1.0f, 0.0f, s);
+                       OP_T, 1.0f, 0.0f, s, 1);
         gemm_tiles_f32(dp + 1, 1, Ap, Ap, Ap, rows_below, rows_below, bs, ld,
-                       ld, ld, OP_N, OP_T, -1.0f, 1.0f, s);
+                       ld, ld, OP_N, OP_T, -1.0f, 1.0f, s, 0);
         break;
       }
       case at::kComplexDouble: {

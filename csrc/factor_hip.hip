@@ -33,23 +33,23 @@ __launch_bounds__(256) __global__ void potrf_block_k(S* A, int n, int ld) {
   for (int cb = 0; cb < ncb; ++cb) {
     const int c0 = cb * 16;
     const int bsz = min(16, n - c0);
-    // 1) factor the 16x16 diagonal block
-    for (int p = 0; p < bsz; ++p) {
-      if (tid == 0) {
-        RT d = TR::real(L[c0 + p][c0 + p]);
-        L[c0 + p][c0 + p] = TR::from_real(sqrt(d));
-      }
-      __syncthreads();
-      if (tid > p && tid < bsz) {
-        RT d = TR::real(L[c0 + p][c0 + p]);
-        L[c0 + tid][c0 + p] = L[c0 + tid][c0 + p] * (RT(1) / d);
-      }
-      __syncthreads();
-      {
-        const int i = tid / 16, j = tid % 16;
+    // 1) factor the 16x16 diagonal block with DEFERRED column scaling
+    //    (LDL^T-style elimination on unscaled columns, one barrier per step:
+    //     U[i][j] -= U[i][p] conj(U[j][p]) / U[p][p], then L[:,j] = U[:,j] /
+    //     sqrt(U[j][j]) in one scale pass).
+    {
+      const int i = tid / 16, j = tid % 16;
+      for (int p = 0; p < bsz; ++p) {
+        const RT s = RT(1) / TR::real(L[c0 + p][c0 + p]);
         if (i < bsz && j < bsz && j > p && j <= i)
-          L[c0 + i][c0 + j] -= L[c0 + i][c0 + p] * TR::conj(L[c0 + j][c0 + p]);
+          L[c0 + i][c0 + j] -= (L[c0 + i][c0 + p] * TR::conj(L[c0 + j][c0 + p])) * s;
+        __syncthreads();
       }
+      // snapshot the column scale BEFORE any thread rewrites the diagonal
+      const bool act = (i < bsz && j < bsz && j <= i);
+      const RT sc = act ? RT(1) / sqrt(TR::real(L[c0 + j][c0 + j])) : RT(1);
+      __syncthreads();
+      if (act) L[c0 + i][c0 + j] = L[c0 + i][c0 + j] * sc;
       __syncthreads();
     }
     // 2) panel solve: rows below vs the diagonal block (X * D^H = A)

@@ -49,14 +49,22 @@ struct Mfma<float> {
 
 // ---------------- real kernel ----------------
 
-template <typename T, int OPA, int OPB>
+// Geometry: block = BM x BN of C, BK-deep K steps, 4 waves.
+//   BN = 64 (default): wave tile 64x32, 8 accumulator fragments (64 regs) ->
+//        fits 2 waves/SIMD (VGPR+AGPR <= 256), latency hidden by the partner.
+//   BN = 128 ("in-place-safe"): wave tile 64x64 -> a single column block spans
+//        the whole N of a panel-apply, making X = X * dinv^H safe in place
+//        (every workgroup reads all of its A rows before writing them).
+template <typename T, int OPA, int OPB, int BN>
 __launch_bounds__(256) __global__ void gemm_tiles_k(
     const GemmDesc* __restrict__ descs, const T* __restrict__ A,
     const T* __restrict__ B, T* __restrict__ C, int M, int N, int K, int lda,
     int ldb, int ldc, T alpha, T beta, int mblocks, int nblocks) {
-  constexpr int BM = 128, BN = 128, BK = 16;
+  constexpr int BM = 128, BK = 16;
   constexpr int LA = BM * BK / 256;  // elements staged per thread (A)
   constexpr int LB = BK * BN / 256;
+  constexpr int WCW = BN / 2;        // wave tile columns
+  constexpr int NFRAG = WCW / 16;
   using acc_t = typename Mfma<T>::acc_t;
 
   __shared__ T As[BM][BK + 1];
@@ -71,17 +79,17 @@ __launch_bounds__(256) __global__ void gemm_tiles_k(
 
   const int tid = threadIdx.x;
   const int lane = tid & 63, w = tid >> 6;
-  const int wrow = (w >> 1) * 64, wcol = (w & 1) * 64;
+  const int wrow = (w >> 1) * 64, wcol = (w & 1) * WCW;
   const int li = lane & 15, lk = lane >> 4;
 
   const int steps_per_tile = (K + BK - 1) / BK;
   const int total_steps = (int)d.ktiles * steps_per_tile;
 
-  acc_t acc[4][4];
+  acc_t acc[4][NFRAG];
 #pragma unroll
   for (int a = 0; a < 4; ++a)
 #pragma unroll
-    for (int b = 0; b < 4; ++b) acc[a][b] = {0, 0, 0, 0};
+    for (int b = 0; b < NFRAG; ++b) acc[a][b] = {0, 0, 0, 0};
 
   T ra[LA], rb[LB];
 
@@ -161,26 +169,26 @@ __launch_bounds__(256) __global__ void gemm_tiles_k(
 
 #pragma unroll
     for (int ks = 0; ks < BK / 4; ++ks) {
-      T af[4], bf[4];
+      T af[4], bf[NFRAG];
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) af[mi] = As[wrow + mi * 16 + li][ks * 4 + lk];
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) bf[ni] = Bs[ks * 4 + lk][wcol + ni * 16 + li];
+      for (int ni = 0; ni < NFRAG; ++ni) bf[ni] = Bs[ks * 4 + lk][wcol + ni * 16 + li];
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
+        for (int ni = 0; ni < NFRAG; ++ni)
           acc[mi][ni] = Mfma<T>::mma(af[mi], bf[ni], acc[mi][ni]);
     }
     __syncthreads();  // LDS reuse barrier
   }
 
-  // epilogue: C = alpha*acc + beta*C  (C/D map: col = lane&15, row = (lane>>4)*4+r)
+  // epilogue: C = alpha*acc + beta*C
   T* Cb = C + d.c_off;
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
+    for (int ni = 0; ni < NFRAG; ++ni) {
       const acc_t v = acc[mi][ni];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -386,9 +394,10 @@ __launch_bounds__(256) __global__ void gemm_tiles_cplx_k(
 template <typename T>
 void launch_real(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
                  int M, int N, int K, int lda, int ldb, int ldc, int opA,
-                 int opB, T alpha, T beta, hipStream_t stream) {
+                 int opB, T alpha, T beta, hipStream_t stream, int inplace) {
   if (ndesc <= 0 || M <= 0 || N <= 0) return;
-  const int mblocks = (M + 127) / 128, nblocks = (N + 127) / 128;
+  const int BN = inplace ? 128 : 64;
+  const int mblocks = (M + 127) / 128, nblocks = (N + BN - 1) / BN;
   const dim3 grid(ndesc * mblocks * nblocks);
   const dim3 block(256);
   // OP_C == OP_T for real scalars
@@ -396,9 +405,14 @@ void launch_real(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
   const int ob = (opB == OP_C) ? OP_T : opB;
 #define CASE(OA, OB)                                                        \
   if (oa == OA && ob == OB) {                                               \
-    gemm_tiles_k<T, OA, OB><<<grid, block, 0, stream>>>(                    \
-        descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks,       \
-        nblocks);                                                           \
+    if (inplace)                                                            \
+      gemm_tiles_k<T, OA, OB, 128><<<grid, block, 0, stream>>>(             \
+          descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks,     \
+          nblocks);                                                         \
+    else                                                                    \
+      gemm_tiles_k<T, OA, OB, 64><<<grid, block, 0, stream>>>(              \
+          descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks,     \
+          nblocks);                                                         \
     return;                                                                 \
   }
   CASE(OP_N, OP_N) CASE(OP_N, OP_T) CASE(OP_T, OP_N) CASE(OP_T, OP_T)
@@ -433,17 +447,17 @@ extern "C" {
 void gemm_tiles_f64(const GemmDesc* descs, int ndesc, const double* A,
                     const double* B, double* C, int M, int N, int K, int lda,
                     int ldb, int ldc, int opA, int opB, double alpha,
-                    double beta, hipStream_t stream) {
+                    double beta, hipStream_t stream, int inplace) {
   launch_real<double>(descs, ndesc, A, B, C, M, N, K, lda, ldb, ldc, opA, opB,
-                      alpha, beta, stream);
+                      alpha, beta, stream, inplace);
 }
 
 void gemm_tiles_f32(const GemmDesc* descs, int ndesc, const float* A,
                     const float* B, float* C, int M, int N, int K, int lda,
                     int ldb, int ldc, int opA, int opB, float alpha, float beta,
-                    hipStream_t stream) {
+                    hipStream_t stream, int inplace) {
   launch_real<float>(descs, ndesc, A, B, C, M, N, K, lda, ldb, ldc, opA, opB,
-                     alpha, beta, stream);
+                     alpha, beta, stream, inplace);
 }
 
 void gemm_tiles_c128(const GemmDesc* descs, int ndesc, const double* A,

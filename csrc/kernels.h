@@ -30,14 +30,16 @@ extern "C" {
 
 // ---- fused batched GEMM (MFMA f64 / f32) ----
 // C (M x N, ldc), op(A) (M x K), op(B) (K x N); K is the per-tile K extent.
+// `inplace` selects the wide-BN instantiation that makes X = X*op(B) safe in
+// place (C block == A block); required whenever C aliases A.
 void gemm_tiles_f64(const GemmDesc* descs, int ndesc, const double* A,
                     const double* B, double* C, int M, int N, int K, int lda,
                     int ldb, int ldc, int opA, int opB, double alpha,
-                    double beta, hipStream_t stream);
+                    double beta, hipStream_t stream, int inplace);
 void gemm_tiles_f32(const GemmDesc* descs, int ndesc, const float* A,
                     const float* B, float* C, int M, int N, int K, int lda,
                     int ldb, int ldc, int opA, int opB, float alpha, float beta,
-                    hipStream_t stream);
+                    hipStream_t stream, int inplace);
 // complex variants (interleaved re/im); opA/opB support OP_C (conjugate).
 void gemm_tiles_c128(const GemmDesc* descs, int ndesc, const double* A,
                      const double* B, double* C, int M, int N, int K, int lda,

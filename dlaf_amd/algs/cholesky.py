@@ -134,7 +134,7 @@ def _run_trsm_panel(table: _DescTable, key_prefix, base: torch.Tensor,
         descs = table.get((key_prefix, "trsm_apply", d))
         if descs is not None:
             ops.gemm_fused(base, base, dinv[d], descs, nb, bs, bs,
-                           nb, bsz, nb, Op.NoTrans, opc, 1.0, 0.0)
+                           nb, bsz, nb, Op.NoTrans, opc, 1.0, 0.0, inplace=True)
 
 
 # ---------------- local ----------------

@@ -67,16 +67,22 @@ def gemm_fused(
     opB: Op,
     alpha,
     beta,
+    inplace: bool = False,
 ) -> None:
-    """GPU fused batched GEMM. ``descs``: np.ndarray [n,6] or device tensor."""
+    """GPU fused batched GEMM. ``descs``: np.ndarray [n,6] or device tensor.
+
+    Set ``inplace=True`` when a desc's C block aliases its A block (panel
+    applies) — it selects the kernel geometry that reads all of A before
+    writing C.
+    """
     ext = get_ext()
     if isinstance(descs, np.ndarray):
         descs = torch.from_numpy(descs).to(C_base.device, non_blocking=True)
     ar, ai = _alpha_parts(alpha)
     br, bi = _alpha_parts(beta)
     ext.batch_gemm(
-        C_base.view(-1), A_base.view(-1), B_base.view(-1), descs,
-        M, N, K, lda, ldb, ldc, _opc(opA), _opc(opB), ar, ai, br, bi,
+        C_base.reshape(-1), A_base.reshape(-1), B_base.reshape(-1), descs,
+        M, N, K, lda, ldb, ldc, _opc(opA), _opc(opB), ar, ai, br, bi, inplace,
     )
 
 
@@ -109,6 +115,32 @@ def dinv_workspace(nb: int, dtype: torch.dtype, device) -> torch.Tensor:
     return torch.empty((nblocks, bsz, bsz), dtype=dtype, device=device)
 
 
+_POTRF_DESC_CACHE = {}
+
+
+def _potrf_descs(n: int, ld: int, dtype: torch.dtype, device) -> torch.Tensor:
+    """Device descriptor table for potrf_tile's internal panel/trailing GEMMs.
+
+    Rows 2d (panel X = A21*dinv^H) and 2d+1 (trailing A22 -= X X^H) per
+    diagonal block d; offsets relative to the tile base pointer, so one table
+    serves every tile of the same (n, ld, dtype)."""
+    key = (n, ld, dtype, str(device))
+    t = _POTRF_DESC_CACHE.get(key)
+    if t is None:
+        bsz = potrf_bsz(dtype)
+        nblocks = (n + bsz - 1) // bsz
+        rows = []
+        for d in range(nblocks):
+            c0 = d * bsz
+            panel_off = (c0 + bsz) * ld + c0
+            trail_off = (c0 + bsz) * ld + (c0 + bsz)
+            rows.append([panel_off, panel_off, 0, 1, 0, 0])
+            rows.append([trail_off, panel_off, panel_off, 1, 0, 0])
+        t = torch.tensor(rows, dtype=torch.int64).reshape(-1).to(device)
+        _POTRF_DESC_CACHE[key] = t
+    return t
+
+
 def potrf_tile(tile: torch.Tensor, dinv: Optional[torch.Tensor] = None) -> Optional[torch.Tensor]:
     """In-place lower Cholesky of a padded tile.
 
@@ -120,7 +152,8 @@ def potrf_tile(tile: torch.Tensor, dinv: Optional[torch.Tensor] = None) -> Optio
     if tile.is_cuda:
         if dinv is None:
             dinv = dinv_workspace(n, tile.dtype, tile.device)
-        get_ext().potrf_tile(tile, n, tile.stride(0), dinv)
+        ddesc = _potrf_descs(n, tile.stride(0), tile.dtype, tile.device)
+        get_ext().potrf_tile(tile, n, tile.stride(0), dinv, ddesc)
         return dinv
     L = torch.linalg.cholesky(tile)
     tile.copy_(L)
@@ -175,10 +208,10 @@ def trsm_panel_right_lowerH(
                 mb, bs, c0, ld, ld_l, ld,
                 Op.NoTrans, opc, -1.0, 1.0,
             )
-        # X[:, c0:c0+bs] @= dinv[d]^H  (in place; single column block per WG)
+        # X[:, c0:c0+bs] @= dinv[d]^H  (in place; wide-BN kernel variant)
         descs = make_descs(offs + c0, offs + c0, [0] * len(offs))
         gemm_fused(
             panel_base, panel_base, dinv[d], descs,
             mb, bs, bs, ld, bsz, ld,
-            Op.NoTrans, opc, 1.0, 0.0,
+            Op.NoTrans, opc, 1.0, 0.0, inplace=True,
         )
