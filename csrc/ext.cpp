@@ -157,47 +157,26 @@ void potrf_tile(torch::Tensor A, int64_t n, int64_t ld, torch::Tensor dinv,
   for (int d = 0; d < nblocks; ++d) {
     const int64_t c0 = (int64_t)d * bsz;
     const int bs = (int)std::min<int64_t>(bsz, n - c0);
-    // 1) factor diagonal block
-    {
-      const int64_t off = c0 * ld + c0;
-      switch (A.scalar_type()) {
-        case at::kDouble:
-          potrf_block128_f64(A.data_ptr<double>() + off, bs, ld, s);
-          break;
-        case at::kFloat:
-          potrf_block128_f32(A.data_ptr<float>() + off, bs, ld, s);
-          break;
-        case at::kComplexDouble:
-          potrf_block128_c128((double*)A.data_ptr() + 2 * off, bs, ld, s);
-          break;
-        case at::kComplexFloat:
-          potrf_block128_c64((float*)A.data_ptr() + 2 * off, bs, ld, s);
-          break;
-        default:
-          TORCH_CHECK(false);
-      }
-    }
-    // 2) inverse of the diagonal block -> dinv[d]
+    // 1+2) fused: factor diagonal block and write its inverse -> dinv[d]
     {
       const int64_t off = c0 * ld + c0;
       const int64_t doff = (int64_t)d * bsz * bsz;
       switch (A.scalar_type()) {
         case at::kDouble:
-          trtri_lower_f64(A.data_ptr<double>() + off,
-                          dinv.data_ptr<double>() + doff, bs, ld, bsz, 0, s);
+          potrf_invert_block_f64(A.data_ptr<double>() + off, bs, ld,
+                                 dinv.data_ptr<double>() + doff, 1, s);
           break;
         case at::kFloat:
-          trtri_lower_f32(A.data_ptr<float>() + off,
-                          dinv.data_ptr<float>() + doff, bs, ld, bsz, 0, s);
+          potrf_invert_block_f32(A.data_ptr<float>() + off, bs, ld,
+                                 dinv.data_ptr<float>() + doff, 1, s);
           break;
         case at::kComplexDouble:
-          trtri_lower_c128((double*)A.data_ptr() + 2 * off,
-                           (double*)dinv.data_ptr() + 2 * doff, bs, ld, bsz, 0,
-                           s);
+          potrf_invert_block_c128((double*)A.data_ptr() + 2 * off, bs, ld,
+                                  (double*)dinv.data_ptr() + 2 * doff, 1, s);
           break;
         case at::kComplexFloat:
-          trtri_lower_c64((float*)A.data_ptr() + 2 * off,
-                          (float*)dinv.data_ptr() + 2 * doff, bs, ld, bsz, 0, s);
+          potrf_invert_block_c64((float*)A.data_ptr() + 2 * off, bs, ld,
+                                 (float*)dinv.data_ptr() + 2 * doff, 1, s);
           break;
         default:
           TORCH_CHECK(false);
@@ -253,12 +232,46 @@ void potrf_tile(torch::Tensor A, int64_t n, int64_t ld, torch::Tensor dinv,
   HIP_CHECK(hipGetLastError());
 }
 
+// Factor (optional) + invert ONE bsz-block of a tile view; Tout = dinv[d].
+void factor_invert_block(torch::Tensor A, int64_t n, int64_t ld,
+                         torch::Tensor Tout, bool do_factor) {
+  TORCH_CHECK(A.is_cuda() && Tout.is_cuda());
+  auto s = cur_stream();
+  switch (A.scalar_type()) {
+    case at::kDouble:
+      TORCH_CHECK(n <= 128);
+      potrf_invert_block_f64(A.data_ptr<double>(), n, ld,
+                             Tout.data_ptr<double>(), do_factor, s);
+      break;
+    case at::kFloat:
+      TORCH_CHECK(n <= 128);
+      potrf_invert_block_f32(A.data_ptr<float>(), n, ld,
+                             Tout.data_ptr<float>(), do_factor, s);
+      break;
+    case at::kComplexDouble:
+      TORCH_CHECK(n <= 64);
+      potrf_invert_block_c128((double*)A.data_ptr(), n, ld,
+                              (double*)Tout.data_ptr(), do_factor, s);
+      break;
+    case at::kComplexFloat:
+      TORCH_CHECK(n <= 64);
+      potrf_invert_block_c64((float*)A.data_ptr(), n, ld,
+                             (float*)Tout.data_ptr(), do_factor, s);
+      break;
+    default:
+      TORCH_CHECK(false, "unsupported dtype");
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("batch_gemm", &batch_gemm,
         "fused batched tile GEMM: C[d] = alpha*op(A[d])op(B[d]) + beta*C[d]");
   m.def("potrf_block", &potrf_block, "single-workgroup Cholesky block factor");
+  m.def("factor_invert_block", &factor_invert_block,
+        "fused single-workgroup [factor+]invert of a diagonal block");
   m.def("trtri_lower", &trtri_lower, "lower-triangular block inverse");
   m.def("potrf_tile", &potrf_tile,
         "in-place tile Cholesky + diagonal-block inverses");

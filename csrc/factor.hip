@@ -122,6 +122,184 @@ __launch_bounds__(256) __global__ void trtri_block_lds_k(const S* L, S* T,
   }
 }
 
+// Fused factor + invert of one diagonal block, single workgroup.
+//
+// Real BSZ=128: LDS = L[128][128] (131 KiB) + S scratch 64x64 (32 KiB) = the
+// full 160 KiB CU budget. The inverse T = L^-1 is computed blockwise with a
+// 64-split, recycling freed LDS regions of L:
+//   T11 = inv(L11) -> S -> (global, and copied over L11's region)
+//   W   = L21*T11  -> S   (L21 still live)
+//   T22 = inv(L22) -> L21's region -> global  (L21 dead after W)
+//   T21 = -T22*W   -> global
+// Complex BSZ=64: direct column-parallel inversion into S (both fit).
+// The dinv output block is always BSZ x BSZ, identity-extended when n < BSZ.
+//
+// do_factor=0 skips the factorization phases (inverting an already-factored
+// received tile after a broadcast).
+template <typename S, int BSZ>
+__launch_bounds__(256) __global__ void potrf_invert_block_k(S* A, int n,
+                                                            int ld, S* Tout,
+                                                            int do_factor) {
+  using TR = ScalarTraits<S>;
+  using RT = typename TR::real_t;
+  constexpr int HB = BSZ / 2;
+  __shared__ S L[BSZ][BSZ];
+  __shared__ S Sc[HB][HB];  // no pad: L + Sc is exactly the 160 KiB CU budget
+  const int tid = threadIdx.x;
+
+  for (int e = tid; e < BSZ * BSZ; e += 256) {
+    const int i = e / BSZ, j = e % BSZ;
+    L[i][j] = (i < n && j < n) ? A[(int64_t)i * ld + j] : TR::zero();
+  }
+  __syncthreads();
+
+  if (do_factor) {
+    const int ncb = (n + 15) / 16;
+    const int ti = tid / 16, tj = tid % 16;
+    for (int cb = 0; cb < ncb; ++cb) {
+      const int c0 = cb * 16;
+      const int bsz16 = min(16, n - c0);
+      // deferred-scaling 16x16 diagonal factor (see potrf_block_k)
+      for (int p = 0; p < bsz16; ++p) {
+        const RT s = RT(1) / TR::real(L[c0 + p][c0 + p]);
+        if (ti < bsz16 && tj < bsz16 && tj > p && tj <= ti)
+          L[c0 + ti][c0 + tj] -= (L[c0 + ti][c0 + p] * TR::conj(L[c0 + tj][c0 + p])) * s;
+        __syncthreads();
+      }
+      const bool act = (ti < bsz16 && tj < bsz16 && tj <= ti);
+      const RT sc = act ? RT(1) / sqrt(TR::real(L[c0 + tj][c0 + tj])) : RT(1);
+      __syncthreads();
+      if (act) L[c0 + ti][c0 + tj] = L[c0 + ti][c0 + tj] * sc;
+      __syncthreads();
+      // panel solve below the 16-block
+      {
+        const int r = c0 + 16 + tid;
+        if (r < n) {
+          for (int j = 0; j < bsz16; ++j) {
+            S x = L[r][c0 + j];
+            for (int p = 0; p < j; ++p) x -= L[r][c0 + p] * TR::conj(L[c0 + j][c0 + p]);
+            L[r][c0 + j] = x * (RT(1) / TR::real(L[c0 + j][c0 + j]));
+          }
+        }
+        __syncthreads();
+      }
+      // trailing update (lower)
+      {
+        const int t0 = c0 + 16;
+        const int nt = n - t0;
+        if (nt > 0) {
+          for (int e = tid; e < nt * nt; e += 256) {
+            const int i = e / nt, j = e % nt;
+            if (j <= i) {
+              S acc = L[t0 + i][c0] * TR::conj(L[t0 + j][c0]);
+              for (int p = 1; p < bsz16; ++p)
+                acc += L[t0 + i][c0 + p] * TR::conj(L[t0 + j][c0 + p]);
+              L[t0 + i][t0 + j] -= acc;
+            }
+          }
+        }
+        __syncthreads();
+      }
+    }
+    // store the factor (lower triangle)
+    for (int e = tid; e < BSZ * BSZ; e += 256) {
+      const int i = e / BSZ, j = e % BSZ;
+      if (i < n && j < n && j <= i) A[(int64_t)i * ld + j] = L[i][j];
+    }
+    // no barrier needed: inversion only reads LDS L, which is final
+  }
+
+  // ---- inversion ----
+  // initialize Tout = identity-extended zero (overwritten below where computed)
+  for (int e = tid; e < BSZ * BSZ; e += 256) {
+    const int i = e / BSZ, j = e % BSZ;
+    Tout[i * BSZ + j] = (i == j && i >= n) ? TR::from_real(1) : TR::zero();
+  }
+  __syncthreads();  // order init writes before the computed overwrites
+
+  if constexpr (BSZ == 64) {
+    // direct column-parallel inversion (complex path; fits alongside L)
+    __shared__ S Tc[64][65];
+    const int j = tid;
+    if (j < n) {
+      for (int i = 0; i < j; ++i) Tc[i][j] = TR::zero();
+      Tc[j][j] = TR::recip(L[j][j]);
+      for (int i = j + 1; i < n; ++i) {
+        S acc = TR::zero();
+        for (int p = j; p < i; ++p) acc += L[i][p] * Tc[p][j];
+        Tc[i][j] = -(TR::recip(L[i][i]) * acc);
+      }
+    }
+    __syncthreads();
+    for (int e = tid; e < BSZ * BSZ; e += 256) {
+      const int i = e / BSZ, j2 = e % BSZ;
+      if (i < n && j2 < n) Tout[i * BSZ + j2] = Tc[i][j2];
+    }
+    return;
+  }
+
+  // BSZ == 128: 64-split scheme
+  const int h = min(HB, n);
+  const int rest = n - h;
+  // T11 = inv(L11) -> Sc
+  {
+    const int j = tid;
+    if (j < h) {
+      for (int i = 0; i < j; ++i) Sc[i][j] = TR::zero();
+      Sc[j][j] = TR::recip(L[j][j]);
+      for (int i = j + 1; i < h; ++i) {
+        S acc = TR::zero();
+        for (int p = j; p < i; ++p) acc += L[i][p] * Sc[p][j];
+        Sc[i][j] = -(TR::recip(L[i][i]) * acc);
+      }
+    }
+    __syncthreads();
+    for (int e = tid; e < h * h; e += 256) {
+      const int i = e / h, j2 = e % h;
+      Tout[i * BSZ + j2] = Sc[i][j2];
+    }
+  }
+  if (rest > 0) {
+    // W = L21 * T11 -> needs T11 (Sc) and L21 (L); write W over L11's region
+    // (L11 is dead). Then T22 = inv(L22) into Sc (Sc free after W copy? no —
+    // W lives in L11's region, Sc holds T11 still needed? T21 = -T22*W only
+    // needs W and T22. So: W -> L11 region, T22 -> Sc (overwrite T11).
+    __syncthreads();
+    for (int e = tid; e < rest * h; e += 256) {
+      const int i = e / h, j = e % h;
+      S acc = TR::zero();
+      for (int p = j; p < h; ++p) acc += L[h + i][p] * Sc[p][j];
+      L[i][j] = acc;  // W[i][j] stored in L11's region (row i < 64)
+    }
+    __syncthreads();
+    // T22 = inv(L22) -> Sc
+    {
+      const int j = tid;
+      if (j < rest) {
+        for (int i = 0; i < j; ++i) Sc[i][j] = TR::zero();
+        Sc[j][j] = TR::recip(L[h + j][h + j]);
+        for (int i = j + 1; i < rest; ++i) {
+          S acc = TR::zero();
+          for (int p = j; p < i; ++p) acc += L[h + i][h + p] * Sc[p][j];
+          Sc[i][j] = -(TR::recip(L[h + i][h + i]) * acc);
+        }
+      }
+      __syncthreads();
+      for (int e = tid; e < rest * rest; e += 256) {
+        const int i = e / rest, j2 = e % rest;
+        Tout[(h + i) * BSZ + (h + j2)] = Sc[i][j2];
+      }
+    }
+    // T21 = -T22 * W  (T22 in Sc, W in L11 region)
+    for (int e = tid; e < rest * h; e += 256) {
+      const int i = e / h, j = e % h;
+      S acc = TR::zero();
+      for (int p = 0; p <= i && p < rest; ++p) acc += Sc[i][p] * L[p][j];
+      Tout[(h + i) * BSZ + j] = -acc;
+    }
+  }
+}
+
 template <typename S>
 __global__ void trtri_lower_k(const S* L, S* T, int n, int ldl, int ldt,
                               int unit_diag) {
@@ -144,6 +322,30 @@ __global__ void trtri_lower_k(const S* L, S* T, int n, int ldl, int ldt,
 }  // namespace
 
 extern "C" {
+
+// Fused factor (optional) + block inverse. Tout is the BSZ x BSZ dinv block.
+void potrf_invert_block_f64(double* A, int n, int ld, double* Tout,
+                            int do_factor, hipStream_t stream) {
+  potrf_invert_block_k<double, 128>
+      <<<1, 256, 0, stream>>>(A, n, ld, Tout, do_factor);
+}
+void potrf_invert_block_f32(float* A, int n, int ld, float* Tout,
+                            int do_factor, hipStream_t stream) {
+  potrf_invert_block_k<float, 128>
+      <<<1, 256, 0, stream>>>(A, n, ld, Tout, do_factor);
+}
+void potrf_invert_block_c128(double* A, int n, int ld, double* Tout,
+                             int do_factor, hipStream_t stream) {
+  potrf_invert_block_k<cplx<double>, 64><<<1, 256, 0, stream>>>(
+      reinterpret_cast<cplx<double>*>(A), n, ld,
+      reinterpret_cast<cplx<double>*>(Tout), do_factor);
+}
+void potrf_invert_block_c64(float* A, int n, int ld, float* Tout,
+                            int do_factor, hipStream_t stream) {
+  potrf_invert_block_k<cplx<float>, 64><<<1, 256, 0, stream>>>(
+      reinterpret_cast<cplx<float>*>(A), n, ld,
+      reinterpret_cast<cplx<float>*>(Tout), do_factor);
+}
 
 void potrf_block128_f64(double* A, int n, int ld, hipStream_t stream) {
   potrf_block_k<double, 128><<<1, 256, 0, stream>>>(A, n, ld);

@@ -53,6 +53,18 @@ void gemm_tiles_c64(const GemmDesc* descs, int ndesc, const float* A,
                     hipStream_t stream);
 
 // ---- single-tile factorization building blocks ----
+// Fused single-workgroup [factor +] invert of one diagonal block: factors the
+// leading n x n (if do_factor) in place and writes its inverse into the
+// BSZ x BSZ block Tout (BSZ = 128 real / 64 complex; identity-extended).
+void potrf_invert_block_f64(double* A, int n, int ld, double* Tout,
+                            int do_factor, hipStream_t stream);
+void potrf_invert_block_f32(float* A, int n, int ld, float* Tout,
+                            int do_factor, hipStream_t stream);
+void potrf_invert_block_c128(double* A, int n, int ld, double* Tout,
+                             int do_factor, hipStream_t stream);
+void potrf_invert_block_c64(float* A, int n, int ld, float* Tout,
+                            int do_factor, hipStream_t stream);
+
 // In-place Cholesky (Lower) of the leading n x n (n <= 128) of a tile with row
 // stride ld; single workgroup, LDS-resident.
 void potrf_block128_f64(double* A, int n, int ld, hipStream_t stream);

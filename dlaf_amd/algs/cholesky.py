@@ -37,6 +37,7 @@ from ..matrix.panel import Panel
 from ..comm.grid import CommGrid
 from ..comm import collectives as coll
 from ..ops import tile_ops as ops
+from ..runtime import get_runtime
 
 
 def _op_conj(dtype) -> Op:
@@ -57,7 +58,7 @@ def _compute_dinv(diag: torch.Tensor, dinv: torch.Tensor) -> torch.Tensor:
     for d in range((nb + bsz - 1) // bsz):
         c0 = d * bsz
         bs = min(bsz, nb - c0)
-        ext.trtri_lower(diag[c0:, c0:], dinv[d], bs, diag.stride(0), bsz, False)
+        ext.factor_invert_block(diag[c0:, c0:], bs, diag.stride(0), dinv[d], False)
     return dinv
 
 
@@ -154,13 +155,25 @@ def _local_plan(mat: Matrix) -> _DescTable:
         offs = np.array([mat.tile_offset((int(i), k)) for i in rows], dtype=np.int64)
         if len(offs):
             _trsm_plan_rows(table, ("k", k), offs, nb, nb, bsz)
-            # trailing: (i, j) for j in k+1..nt, i in j..nt
-            js = np.repeat(rows, nt - rows)
-            is_ = np.concatenate([np.arange(j, nt) for j in rows]) if len(rows) else np.empty(0, np.int64)
-            c = np.array([mat.tile_offset((int(i), int(j))) for i, j in zip(is_, js)], dtype=np.int64)
-            a = np.array([mat.tile_offset((int(i), k)) for i in is_], dtype=np.int64)
-            b = np.array([mat.tile_offset((int(j), k)) for j in js], dtype=np.int64)
-            table.add(("k", k, "trail"), _rows6(c, a, b))
+            # trailing split for lookahead:
+            #   head = next panel column (j == k+1), factored eagerly on the
+            #          high-priority stream so potrf(k+1) is not blocked;
+            #   tail = the rest (j >= k+2), the bulk, on the update stream.
+            def _descs_for(js_range):
+                c, a, b = [], [], []
+                for j in js_range:
+                    for i in range(j, nt):
+                        c.append(mat.tile_offset((i, j)))
+                        a.append(mat.tile_offset((i, k)))
+                        b.append(mat.tile_offset((j, k)))
+                return _rows6(np.array(c, dtype=np.int64), np.array(a, dtype=np.int64),
+                              np.array(b, dtype=np.int64)) if c else None
+            h = _descs_for([k + 1])
+            if h is not None:
+                table.add(("k", k, "head"), h)
+            t = _descs_for(range(k + 2, nt))
+            if t is not None:
+                table.add(("k", k, "tail"), t)
     table.upload(mat.device)
     _PLAN_CACHE[key] = table
     return table
@@ -173,17 +186,45 @@ def _cholesky_local(mat: Matrix) -> None:
     gpu = mat.device.type == "cuda"
     opc = _op_conj(mat.dtype)
     if gpu:
+        # Lookahead schedule on two streams (see plan docstring):
+        #   Sp (high prio): potrf(k) + panel TRSM(k) + head update (col k+1)
+        #   Su:             tail update (cols >= k+2) — the bulk of the flops
+        # P(k) waits tail(k-2); head(k) waits tail(k-1); tail(k) waits P(k).
         table = _local_plan(mat)
         dinv = ops.dinv_workspace(nb, mat.dtype, mat.device)
         st = mat.storage
+        rt = get_runtime(mat.device)
+        sp, su = rt.hp_streams[0], rt.np_streams[0]
+        cur = torch.cuda.current_stream(mat.device)
+        sp.wait_stream(cur)
+        su.wait_stream(cur)
+        ev_tail = [None] * nt
         for k in range(nt):
-            diag = mat.tile((k, k))
-            ops.potrf_tile(diag, dinv)
-            _run_trsm_panel(table, ("k", k), st, diag, dinv, nb, opc)
-            descs = table.get(("k", k, "trail"))
-            if descs is not None:
-                ops.gemm_fused(st, st, st, descs, nb, nb, nb, nb, nb, nb,
-                               Op.NoTrans, opc, -1.0, 1.0)
+            with torch.cuda.stream(sp):
+                if k >= 2 and ev_tail[k - 2] is not None:
+                    sp.wait_event(ev_tail[k - 2])
+                diag = mat.tile((k, k))
+                ops.potrf_tile(diag, dinv)
+                _run_trsm_panel(table, ("k", k), st, diag, dinv, nb, opc)
+                ev_p = torch.cuda.Event()
+                ev_p.record(sp)
+                head = table.get(("k", k, "head"))
+                if head is not None:
+                    if k >= 1 and ev_tail[k - 1] is not None:
+                        sp.wait_event(ev_tail[k - 1])
+                    ops.gemm_fused(st, st, st, head, nb, nb, nb, nb, nb, nb,
+                                   Op.NoTrans, opc, -1.0, 1.0)
+            tail = table.get(("k", k, "tail"))
+            if tail is not None:
+                with torch.cuda.stream(su):
+                    su.wait_event(ev_p)
+                    ops.gemm_fused(st, st, st, tail, nb, nb, nb, nb, nb, nb,
+                                   Op.NoTrans, opc, -1.0, 1.0)
+                    ev = torch.cuda.Event()
+                    ev.record(su)
+                    ev_tail[k] = ev
+        cur.wait_stream(sp)
+        cur.wait_stream(su)
     else:
         for k in range(nt):
             diag = mat.tile((k, k))
