@@ -28,7 +28,8 @@ hipStream_t cur_stream() {
 }
 
 int potrf_bsz(at::ScalarType t) {
-  return (t == at::kComplexDouble || t == at::kComplexFloat) ? 64 : 128;
+  (void)t;
+  return 64;  // see potrf_invert_block launchers: 64 keeps the kernel co-residable
 }
 
 void check_gemm_args(const torch::Tensor& desc, const torch::Tensor& A,
@@ -239,12 +240,12 @@ void factor_invert_block(torch::Tensor A, int64_t n, int64_t ld,
   auto s = cur_stream();
   switch (A.scalar_type()) {
     case at::kDouble:
-      TORCH_CHECK(n <= 128);
+      TORCH_CHECK(n <= 64);
       potrf_invert_block_f64(A.data_ptr<double>(), n, ld,
                              Tout.data_ptr<double>(), do_factor, s);
       break;
     case at::kFloat:
-      TORCH_CHECK(n <= 128);
+      TORCH_CHECK(n <= 64);
       potrf_invert_block_f32(A.data_ptr<float>(), n, ld,
                              Tout.data_ptr<float>(), do_factor, s);
       break;
@@ -275,6 +276,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("trtri_lower", &trtri_lower, "lower-triangular block inverse");
   m.def("potrf_tile", &potrf_tile,
         "in-place tile Cholesky + diagonal-block inverses");
-  m.attr("POTRF_BSZ_REAL") = 128;
+  m.attr("POTRF_BSZ_REAL") = 64;
   m.attr("POTRF_BSZ_CPLX") = 64;
 }

@@ -324,14 +324,17 @@ __global__ void trtri_lower_k(const S* L, S* T, int n, int ldl, int ldt,
 extern "C" {
 
 // Fused factor (optional) + block inverse. Tout is the BSZ x BSZ dinv block.
+// BSZ = 64 for ALL dtypes: the 64-block kernel needs ~73 KiB LDS, so it can
+// co-reside with trailing-update GEMM blocks on a CU — a 128-block (160 KiB)
+// kernel can never be scheduled next to them and serializes the lookahead.
 void potrf_invert_block_f64(double* A, int n, int ld, double* Tout,
                             int do_factor, hipStream_t stream) {
-  potrf_invert_block_k<double, 128>
+  potrf_invert_block_k<double, 64>
       <<<1, 256, 0, stream>>>(A, n, ld, Tout, do_factor);
 }
 void potrf_invert_block_f32(float* A, int n, int ld, float* Tout,
                             int do_factor, hipStream_t stream) {
-  potrf_invert_block_k<float, 128>
+  potrf_invert_block_k<float, 64>
       <<<1, 256, 0, stream>>>(A, n, ld, Tout, do_factor);
 }
 void potrf_invert_block_c128(double* A, int n, int ld, double* Tout,

@@ -396,7 +396,8 @@ void launch_real(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
                  int M, int N, int K, int lda, int ldb, int ldc, int opA,
                  int opB, T alpha, T beta, hipStream_t stream, int inplace) {
   if (ndesc <= 0 || M <= 0 || N <= 0) return;
-  const int BN = inplace ? 128 : 64;
+  // N <= 64 is a single column block even at BN=64, hence in-place safe.
+  const int BN = (inplace && N > 64) ? 128 : 64;
   const int mblocks = (M + 127) / 128, nblocks = (N + BN - 1) / BN;
   const dim3 grid(ndesc * mblocks * nblocks);
   const dim3 block(256);

@@ -106,7 +106,9 @@ def gemm_tile(C: torch.Tensor, A: torch.Tensor, B: torch.Tensor, opA: Op, opB: O
 # ---------------- factorization tile ops ----------------
 
 def potrf_bsz(dtype: torch.dtype) -> int:
-    return 64 if is_complex(dtype) else 128
+    # 64 for every dtype: the fused factor+invert kernel then fits next to
+    # trailing-GEMM blocks on a CU (LDS budget), so lookahead can overlap it.
+    return 64
 
 
 def dinv_workspace(nb: int, dtype: torch.dtype, device) -> torch.Tensor:
