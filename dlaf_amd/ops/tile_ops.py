@@ -176,6 +176,40 @@ def trtri_tile(L: torch.Tensor, out: torch.Tensor, unit_diag: bool = False) -> N
     out.copy_(torch.tril(T))
 
 
+def tri_inverse_full(A: torch.Tensor, lower: bool, unit: bool = False) -> torch.Tensor:
+    """Full-tile inverse of a triangular tile (other triangle zeroed).
+
+    GPU: native column-parallel ``trtri_lower`` kernel (upper handled via
+    T = (trtri_lower(A^H))^H). CPU: solve_triangular against the identity.
+    Used by the TRSM/TRMM/inverse algorithms to turn per-tile triangular
+    solves into fused GEMMs (TRSM-as-GEMM, see csrc/factor.hip).
+    """
+    n = A.shape[0]
+    if A.is_cuda:
+        out = torch.empty_like(A)
+        if lower:
+            get_ext().trtri_lower(A, out, n, A.stride(0), out.stride(0), unit)
+            return out
+        tmp = A.mH.contiguous()
+        low = torch.empty_like(tmp)
+        get_ext().trtri_lower(tmp, low, n, tmp.stride(0), low.stride(0), unit)
+        return low.mH.contiguous()
+    eye = torch.eye(n, dtype=A.dtype, device=A.device)
+    if unit:
+        tri = torch.tril(A, -1) + eye if lower else torch.triu(A, 1) + eye
+    else:
+        tri = torch.tril(A) if lower else torch.triu(A)
+    return torch.linalg.solve_triangular(tri, eye, upper=not lower, unitriangular=False)
+
+
+def tri_mask(A: torch.Tensor, lower: bool, unit: bool = False) -> torch.Tensor:
+    """Masked copy of a triangular tile (for TRMM diag-block multiplies)."""
+    if unit:
+        eye = torch.eye(A.shape[0], dtype=A.dtype, device=A.device)
+        return (torch.tril(A, -1) if lower else torch.triu(A, 1)) + eye
+    return torch.tril(A) if lower else torch.triu(A)
+
+
 def trsm_panel_right_lowerH(
     panel_base: torch.Tensor,
     tile_offs: Sequence[int],
