@@ -1,13 +1,155 @@
 """Triangular inverse (TRTRI) and inverse from Cholesky factor (POTRI).
 
-Counterpart of ``inverse/triangular/impl.h`` and ``inverse/cholesky/impl.h``.
+Counterpart of ``inverse/triangular/impl.h:183-549`` and
+``inverse/cholesky/impl.h:180-540``.
+
+TRTRI (Lower, in place, X = L^-1), processed by tile-column k DESCENDING
+(LAPACK xTRTRI lower order): the trailing inverse X22 (already computed, in
+place) is applied to the original column L[>k, k]
+
+    acc[i]  = sum_{k < j <= i} X[i,j] L[j,k]      (fused GEMMs + row-reduce)
+    X[i,k]  = -acc[i] @ inv(L[k,k])
+    X[k,k]  = inv(L[k,k])
+
+LAUUM-stage of POTRI (A^-1 = X^H X for X = L^-1, lower triangle, in place),
+processed by tile-row k ASCENDING: row k of X is broadcast (direct row panel +
+two-hop transposed panel), then contributes the rank-nb update
+B[i,j] += X[k,i]^H X[k,j] for j <= i <= k; the first contribution to row i
+arrives exactly at step k = i and overwrites X there (beta = 0).
+
+Comm per step: contiguous panel broadcasts + one row-direction reduce
+(TRTRI); all local tile products are ONE fused-GEMM launch on GPU.
 """
+
 from __future__ import annotations
 
+from typing import Optional
 
-def triangular_inverse(*args, **kwargs):
-    raise NotImplementedError("triangular_inverse: in progress")
+import torch
+
+from ..types import UpLo, Diag, Op, is_complex
+from ..matrix.matrix import Matrix
+from ..matrix.panel import Panel
+from ..comm.grid import CommGrid
+from ..comm import collectives as coll
+from ..ops import tile_ops as ops
+from . import _panels as pan
+from .triangular import _trivial_grid
 
 
-def inverse_from_cholesky_factor(*args, **kwargs):
-    raise NotImplementedError("inverse_from_cholesky_factor: in progress")
+def _opc(dtype) -> Op:
+    return Op.ConjTrans if is_complex(dtype) else Op.Trans
+
+
+def triangular_inverse(uplo: UpLo, diag: Diag, mat: Matrix,
+                       grid: Optional[CommGrid] = None) -> None:
+    """In-place inverse of a (lower) triangular tiled matrix.
+
+    Reference: ``inverse/triangular/impl.h`` (Lower; Upper by symmetry is not
+    provided by the reference miniapps either).
+    """
+    assert uplo == UpLo.Lower, "only Lower implemented (as the reference's miniapps)"
+    d = mat.dist
+    assert d.m == d.n and d.mb == d.nb
+    g = _trivial_grid(grid if grid is not None else mat.grid)
+    unit = diag == Diag.Unit
+    nt = d.nr_tiles[0]
+    nb = d.nb
+    dev, dt = mat.device, mat.dtype
+    lr, lc = d.local_nr_tiles
+
+    colpL = Panel(Panel.COL, d, dt, dev)    # original column k, keyed by rows
+    rowpL = Panel(Panel.ROW, d, dt, dev)    # same tiles keyed by cols (transposed)
+    acc = Panel(Panel.COL, d, dt, dev)      # partial sums of X22 @ L[>k,k]
+    diag_ws = torch.empty((nb, nb), dtype=dt, device=dev)
+
+    # Unit diag: the diagonal is implicit (ones) and must not be modified; the
+    # algorithm stores explicit unit-inverses (ones on the diagonal) so the
+    # trailing products are plain GEMMs, and the caller's diagonal entries are
+    # restored afterwards.
+    saved_diag = {}
+    if unit:
+        for k in range(nt):
+            if d.is_tile_local((k, k)):
+                saved_diag[k] = mat.tile((k, k)).diagonal().clone()
+
+    for k in range(nt - 1, -1, -1):
+        kc = d.rank_of_tile_col(k)
+        li0 = d.next_local_tile_row(k + 1)
+        lj0 = d.next_local_tile_col(k + 1)
+        # 1) acc[i] = sum_{k<j<=i} X[i,j] L[j,k]  (X = already-inverted trailing).
+        # NOTE: the panel broadcasts must run on every rank of their process
+        # group — their internal ranges are group-consistent, the local row
+        # range li0:lr is NOT (it differs across the rows of a column group).
+        pan.bcast_col_panel(mat, g, k, li0, lr, colpL)
+        pan.transpose_col_to_row(d, g, colpL, rowpL, lj0, lc)
+        if li0 < lr:
+            acc.storage[li0:lr].zero_()
+            items = []
+            for li in range(li0, lr):
+                i = d.global_tile_of_local((li, 0))[0]
+                for lj in range(lj0, d.next_local_tile_col(i + 1)):
+                    items.append((acc.offset(li), mat.local_tile_offset(li, lj), rowpL.offset(lj)))
+            ops.gemm_items(acc.storage, mat.storage, rowpL.storage, items, nb,
+                           Op.NoTrans, Op.NoTrans, 1.0, 1.0)
+            if g.row_group is not None:
+                coll.reduce_sum(acc.range_view(li0, lr), g.global_rank_of_row_member(kc),
+                                g.row_group)
+        # 2) invert diagonal block on the owning column, write column k
+        dtile = pan.bcast_diag_to_col(mat, g, k, diag_ws)
+        if d.rank_col == kc:
+            inv = ops.tri_inverse_full(dtile, lower=True, unit=unit)
+            if li0 < lr:
+                # X[i,k] = -acc[i] @ inv  (write into matrix tiles)
+                c = [mat.tile_offset((d.global_tile_of_local((li, 0))[0], k))
+                     for li in range(li0, lr)]
+                items = list(zip(c, [acc.offset(li) for li in range(li0, lr)], [0] * len(c)))
+                ops.gemm_items(mat.storage, acc.storage, inv, items, nb,
+                               Op.NoTrans, Op.NoTrans, -1.0, 0.0)
+            kr = d.rank_of_tile_row(k)
+            if d.rank_row == kr:
+                mat.tile((k, k)).copy_(inv)
+
+    if unit:
+        for k, dg in saved_diag.items():
+            mat.tile((k, k)).diagonal().copy_(dg)
+
+
+def inverse_from_cholesky_factor(uplo: UpLo, mat: Matrix,
+                                 grid: Optional[CommGrid] = None) -> None:
+    """A^-1 = L^-H L^-1 from the Cholesky factor L, lower triangle in place.
+
+    Reference: ``inverse/cholesky/impl.h:180-540`` (TRTRI then LAUUM-style
+    assembly).
+    """
+    assert uplo == UpLo.Lower, "only Lower implemented (as the reference miniapps)"
+    d = mat.dist
+    assert d.m == d.n and d.mb == d.nb
+    g = _trivial_grid(grid if grid is not None else mat.grid)
+    triangular_inverse(UpLo.Lower, Diag.NonUnit, mat, grid)
+
+    nt = d.nr_tiles[0]
+    nb = d.nb
+    dev, dt = mat.device, mat.dtype
+    lr, lc = d.local_nr_tiles
+    opc = _opc(dt)
+    rowpX = Panel(Panel.ROW, d, dt, dev)   # X[k, j] keyed by local cols
+    colpX = Panel(Panel.COL, d, dt, dev)   # X[k, i] keyed by local rows
+
+    for k in range(nt):
+        # broadcast row k of X (cols <= k), direct + transposed
+        lj1 = d.next_local_tile_col(k + 1)
+        pan.bcast_row_panel(mat, g, k, 0, lj1, rowpX)
+        li1 = d.next_local_tile_row(k + 1)
+        pan.transpose_row_to_col(d, g, rowpX, colpX, 0, li1)
+        # B[i,j] (+)= X[k,i]^H X[k,j] for j <= i <= k; first write at i == k
+        first, accum = [], []
+        for li in range(0, li1):
+            i = d.global_tile_of_local((li, 0))[0]
+            for lj in range(0, d.next_local_tile_col(i + 1)):
+                trip = (mat.local_tile_offset(li, lj), colpX.offset(li), rowpX.offset(lj))
+                (first if i == k else accum).append(trip)
+        ops.gemm_items(mat.storage, colpX.storage, rowpX.storage, first, nb,
+                       opc, Op.NoTrans, 1.0, 0.0)
+        ops.gemm_items(mat.storage, colpX.storage, rowpX.storage, accum, nb,
+                       opc, Op.NoTrans, 1.0, 1.0)

@@ -176,6 +176,37 @@ def trtri_tile(L: torch.Tensor, out: torch.Tensor, unit_diag: bool = False) -> N
     out.copy_(torch.tril(T))
 
 
+def gemm_items(
+    C_base: torch.Tensor,
+    A_base: torch.Tensor,
+    B_base: torch.Tensor,
+    items: Sequence[Tuple[int, int, int]],
+    nb: int,
+    opA: Op,
+    opB: Op,
+    alpha,
+    beta,
+    inplace: bool = False,
+) -> None:
+    """Tile-triple GEMM over element offsets into flat bases, CPU or GPU.
+
+    ``items``: sequence of (c_off, a_off, b_off). GPU: ONE fused kernel launch.
+    CPU: loop over reshaped nb x nb views (the reference MC backend analog).
+    """
+    if not len(items):
+        return
+    if C_base.is_cuda:
+        c, a, b = zip(*items)
+        gemm_fused(C_base, A_base, B_base, make_descs(c, a, b),
+                   nb, nb, nb, nb, nb, nb, opA, opB, alpha, beta, inplace=inplace)
+        return
+    ts = nb * nb
+    cv, av, bv = C_base.reshape(-1), A_base.reshape(-1), B_base.reshape(-1)
+    for c, a, b in items:
+        gemm_tile(cv[c:c + ts].view(nb, nb), av[a:a + ts].view(nb, nb),
+                  bv[b:b + ts].view(nb, nb), opA, opB, alpha, beta)
+
+
 def tri_inverse_full(A: torch.Tensor, lower: bool, unit: bool = False) -> torch.Tensor:
     """Full-tile inverse of a triangular tile (other triangle zeroed).
 
