@@ -1,0 +1,249 @@
+// Band -> tridiagonal bulge chasing (CPU), with compact reflector recording.
+//
+// Counterpart of the reference's eigensolver/band_to_tridiag/mc.h
+// (SweepWorker::start_sweep/do_step, mc.h:477-631): a Hermitian band matrix of
+// bandwidth b (compact lower-band storage, column major, ld rows >= 2b so the
+// transient bulge stays in storage) is reduced to real tridiagonal form by
+// n-2 sweeps. Sweep s eliminates column s below the first subdiagonal with a
+// length-<=b Householder reflector and chases the resulting bulge down with
+// one reflector per b rows:
+//
+//   step t (j = 1 + s + t*b, n = min(b, size-j), m = min(b, size-b-j)):
+//     two-sided H^H * A[j:j+n, j:j+n] * H          (Hermitian diag block)
+//     right     A[j+n:j+n+m, j:j+n] * H            (off-diag block, fills it)
+//     reflector from A[j+n:j+n+m, j]               (eliminates the spilled col)
+//     left      H'^H * A[j+n:j+n+m, j+1:j+n]       (mix rows; bulge remains in
+//                                                   cols j+1.. for LATER sweeps)
+//
+// Reflectors are stored compactly: slot (s, t) -> [tau, v0=1 implied, v1..],
+// stride (b+1), per-sweep offsets; the back-transform applies them per sweep
+// as disjoint row-block rank-1 updates (see dlaf_amd/algs/band2tridiag.py).
+//
+// Sequential v1 (the reference's MC backend is CPU too); the wavefront
+// parallelization over sweeps (semaphore scheme, mc.h:666-693) can be added
+// without changing the storage contract.
+
+#include <torch/extension.h>
+
+#include <cmath>
+#include <complex>
+#include <vector>
+
+namespace {
+
+template <class T>
+struct real_of {
+  using type = T;
+};
+template <class R>
+struct real_of<std::complex<R>> {
+  using type = R;
+};
+
+template <class T>
+inline typename real_of<T>::type re(const T& x) {
+  if constexpr (std::is_same_v<T, std::complex<float>> ||
+                std::is_same_v<T, std::complex<double>>)
+    return x.real();
+  else
+    return x;
+}
+
+template <class T>
+inline T conj_(const T& x) {
+  if constexpr (std::is_same_v<T, std::complex<float>> ||
+                std::is_same_v<T, std::complex<double>>)
+    return std::conj(x);
+  else
+    return x;
+}
+
+template <class T>
+inline typename real_of<T>::type abs2(const T& x) {
+  if constexpr (std::is_same_v<T, std::complex<float>> ||
+                std::is_same_v<T, std::complex<double>>)
+    return x.real() * x.real() + x.imag() * x.imag();
+  else
+    return x * x;
+}
+
+// LAPACK-style larfg on x[0..n): returns tau; x[0] <- beta (real),
+// x[1..] <- v tail (v0 = 1 implicit).
+template <class T>
+T hh_reflector(int64_t n, T* x) {
+  using R = typename real_of<T>::type;
+  if (n <= 1) {
+    // still normalize a complex x[0]? nothing to eliminate
+    return T(0);
+  }
+  R xnorm2 = 0;
+  for (int64_t i = 1; i < n; ++i) xnorm2 += abs2(x[i]);
+  T alpha = x[0];
+  R a_re = re(alpha);
+  R a_im = 0;
+  if constexpr (std::is_same_v<T, std::complex<float>> ||
+                std::is_same_v<T, std::complex<double>>)
+    a_im = alpha.imag();
+  if (xnorm2 == 0 && a_im == 0) return T(0);
+  R beta = -std::sqrt(a_re * a_re + a_im * a_im + xnorm2);
+  if (a_re < 0) beta = -beta;
+  T tau;
+  if constexpr (std::is_same_v<T, std::complex<float>> ||
+                std::is_same_v<T, std::complex<double>>)
+    tau = T((beta - a_re) / beta, -a_im / beta);
+  else
+    tau = (beta - a_re) / beta;
+  T scale = T(1) / (alpha - T(beta));
+  for (int64_t i = 1; i < n; ++i) x[i] *= scale;
+  x[0] = T(beta);
+  return tau;
+}
+
+// Two-sided Hermitian update of the n x n lower-band-stored block at column j:
+// block(p, q) = a[(p - q) + (j + q) * ld], p >= q. w: scratch length n.
+template <class T>
+void apply_two_sided(int64_t nn, T tau, const T* v, T* a, int64_t ld, T* w) {
+  if (tau == T(0) || nn <= 0) return;
+  auto blk = [&](int64_t p, int64_t q) -> T& { return a[(p - q) + q * ld]; };
+  // u = B v (Hermitian, lower stored)
+  for (int64_t p = 0; p < nn; ++p) w[p] = T(0);
+  for (int64_t q = 0; q < nn; ++q) {
+    w[q] += T(re(blk(q, q))) * v[q];
+    for (int64_t p = q + 1; p < nn; ++p) {
+      w[p] += blk(p, q) * v[q];
+      w[q] += conj_(blk(p, q)) * v[p];
+    }
+  }
+  // w' = tau u - 1/2 |tau|^2 (v^H u) v
+  T vhu = T(0);
+  for (int64_t p = 0; p < nn; ++p) vhu += conj_(v[p]) * w[p];
+  T half = T(abs2(tau) / typename real_of<T>::type(2)) * vhu;
+  for (int64_t p = 0; p < nn; ++p) w[p] = tau * w[p] - half * v[p];
+  // B -= v w'^H + w' v^H
+  for (int64_t q = 0; q < nn; ++q)
+    for (int64_t p = q; p < nn; ++p)
+      blk(p, q) -= v[p] * conj_(w[q]) + w[p] * conj_(v[q]);
+}
+
+// Right-apply C <- C (I - tau v v^H) to the m x n block whose (p, q) entry is
+// a[(d0 + p - q) + (j + q) * ld] (d0 = n for the off-diag block).
+template <class T>
+void apply_right(int64_t m, int64_t nn, T tau, const T* v, T* a, int64_t ld,
+                 int64_t d0, T* w) {
+  if (tau == T(0) || m <= 0 || nn <= 0) return;
+  auto blk = [&](int64_t p, int64_t q) -> T& { return a[(d0 + p - q) + q * ld]; };
+  for (int64_t p = 0; p < m; ++p) {
+    T s = T(0);
+    for (int64_t q = 0; q < nn; ++q) s += blk(p, q) * v[q];
+    s *= tau;
+    for (int64_t q = 0; q < nn; ++q) blk(p, q) -= s * conj_(v[q]);
+  }
+}
+
+// Left-apply C <- (I - conj(tau) v v^H) C: v indexes ROWS.
+template <class T>
+void apply_left(int64_t m, int64_t nn, T tau, const T* v, T* a, int64_t ld,
+                int64_t d0, T* w) {
+  if (tau == T(0) || m <= 0 || nn <= 0) return;
+  auto blk = [&](int64_t p, int64_t q) -> T& { return a[(d0 + p - q) + q * ld]; };
+  T ct = conj_(tau);
+  for (int64_t q = 0; q < nn; ++q) {
+    T s = T(0);
+    for (int64_t p = 0; p < m; ++p) s += conj_(v[p]) * blk(p, q);
+    s *= ct;
+    for (int64_t p = 0; p < m; ++p) blk(p, q) -= s * v[p];
+  }
+}
+
+// Sequential chase of the whole band matrix.
+// band: [ld, size] column-major torch tensor (band.stride(1) == ld... we use
+//   contiguous [size, ld] row-major = column-major [ld, size]).
+// vstore: flat reflector storage; offsets[s] = slot index of sweep s's step 0;
+//   slot k of sweep s lives at (offsets[s] + k) * (b + 1).
+template <class T>
+void chase_impl(T* a, int64_t ld, int64_t size, int64_t b, T* vstore,
+                const int64_t* offsets) {
+  std::vector<T> v(b + 1), w(b);
+  int64_t vstride = b + 1;
+  // sweeps eliminate columns 0 .. size-3 (column size-2 is already tridiagonal)
+  for (int64_t s = 0; s + 3 <= size; ++s) {
+    // start sweep: eliminate column s below the first subdiagonal
+    int64_t n0 = std::min(size - s - 1, b);
+    T tau = hh_reflector(n0, a + 1 + s * ld);
+    // copy out reflector (v0 = 1 implicit in storage too: slot = tau, 1, tail)
+    {
+      T* slot = vstore + offsets[s] * vstride;
+      slot[0] = tau;
+      slot[1] = T(1);
+      for (int64_t i = 1; i < n0; ++i) slot[1 + i] = a[1 + i + s * ld];
+      for (int64_t i = n0; i < b; ++i) slot[1 + i] = T(0);
+    }
+    v[0] = T(1);
+    for (int64_t i = 1; i < n0; ++i) v[i] = a[1 + i + s * ld];
+    // zero the eliminated entries in storage (they are now implicit)
+    for (int64_t i = 1; i < n0; ++i) a[1 + i + s * ld] = T(0);
+
+    int64_t step = 0;
+    int64_t nref = n0;
+    while (true) {
+      int64_t j = 1 + s + step * b;
+      int64_t nn = std::min(b, size - j);
+      int64_t m = std::min(b, size - b - j);
+      // two-sided on the diag block with the current reflector (length nref)
+      apply_two_sided(nn, tau, v.data(), a + j * ld, ld, w.data());
+      if (m > 0) apply_right(m, nn, tau, v.data(), a + j * ld, ld, nn, w.data());
+      if (m <= 1) break;
+      // new reflector from the spilled column j (rows j+nn .. j+nn+m)
+      tau = hh_reflector(m, a + nn + j * ld);
+      ++step;
+      {
+        T* slot = vstore + (offsets[s] + step) * vstride;
+        slot[0] = tau;
+        slot[1] = T(1);
+        for (int64_t i = 1; i < m; ++i) slot[1 + i] = a[nn + i + j * ld];
+        for (int64_t i = m; i < b; ++i) slot[1 + i] = T(0);
+      }
+      v[0] = T(1);
+      for (int64_t i = 1; i < m; ++i) v[i] = a[nn + i + j * ld];
+      for (int64_t i = 1; i < m; ++i) a[nn + i + j * ld] = T(0);
+      nref = m;
+      apply_left(m, nn - 1, tau, v.data(), a + (nn - 1) + (j + 1) * ld, ld, 0,
+                 w.data());
+    }
+  }
+}
+
+}  // namespace
+
+// band: [size, ld] row-major (= column-major [ld, size]) with
+//   band[j][d] = A[j+d, j], d = 0..ld-1, ld >= 2b (bulge headroom).
+// vstore: [total_slots, b+1]; offsets: [size] int64 slot offsets per sweep.
+void band_chase(torch::Tensor band, int64_t b, torch::Tensor vstore,
+                torch::Tensor offsets) {
+  TORCH_CHECK(!band.is_cuda(), "band_chase is a CPU stage (as the reference)");
+  TORCH_CHECK(band.is_contiguous() && vstore.is_contiguous());
+  int64_t size = band.size(0);
+  int64_t ld = band.size(1);
+  TORCH_CHECK(ld >= 2 * b, "band storage needs >= 2b rows for the bulge");
+  TORCH_CHECK(vstore.size(1) == b + 1);
+  TORCH_CHECK(offsets.scalar_type() == torch::kInt64);
+  auto offs = offsets.data_ptr<int64_t>();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::ComplexFloat, at::ScalarType::ComplexDouble,
+      band.scalar_type(), "band_chase", [&] {
+        using T = scalar_t;
+        if constexpr (std::is_same_v<T, c10::complex<float>>) {
+          chase_impl(reinterpret_cast<std::complex<float>*>(band.data_ptr<T>()),
+                     ld, size, b,
+                     reinterpret_cast<std::complex<float>*>(vstore.data_ptr<T>()),
+                     offs);
+        } else if constexpr (std::is_same_v<T, c10::complex<double>>) {
+          chase_impl(reinterpret_cast<std::complex<double>*>(band.data_ptr<T>()),
+                     ld, size, b,
+                     reinterpret_cast<std::complex<double>*>(vstore.data_ptr<T>()),
+                     offs);
+        } else {
+          chase_impl(band.data_ptr<T>(), ld, size, b, vstore.data_ptr<T>(), offs);
+        }
+      });
+}

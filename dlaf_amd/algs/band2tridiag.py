@@ -1,0 +1,131 @@
+"""Band -> tridiagonal reduction (bulge chasing) and its back-transform.
+
+Counterpart of ``eigensolver/band_to_tridiag/mc.h`` (local 663-989; CPU-only
+compute, as the reference: the band is copied D2H, chased on the host, and
+the reflectors are applied to the eigenvectors on the GPU afterwards) and of
+``eigensolver/bt_band_to_tridiag/impl.h``.
+
+The chase itself is the native C++ engine (``csrc/band_chase.cpp``). The
+back-transform exploits that the reflectors of ONE sweep act on DISJOINT
+row blocks of length ``band``: each sweep becomes two batched GEMV-shaped
+device ops over an [nblocks, band, nE] view of E (sweeps applied in reverse
+chronological order).
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+
+from ..types import UpLo
+from ..matrix.matrix import Matrix
+from ..ops._ext import get_ext
+
+
+@dataclass
+class TridiagResult:
+    """d, e: real tridiagonal; vstore/offsets: compact chase reflectors."""
+    d: torch.Tensor            # [n] real
+    e: torch.Tensor            # [n-1] real
+    band: int
+    n: int
+    vstore: torch.Tensor       # [slots, band+1]: tau, v0=1, v-tail (CPU)
+    offsets: torch.Tensor      # [n] int64 slot offset per sweep
+    nslots: torch.Tensor       # [n] int64 slots per sweep
+    phases: Optional[torch.Tensor] = None   # [n] unitary diag (complex input)
+
+
+def _slot_counts(n: int, b: int) -> torch.Tensor:
+    counts = torch.zeros(n, dtype=torch.int64)
+    for s in range(max(n - 2, 0)):
+        k = 1
+        t = 0
+        while True:
+            j = 1 + s + t * b
+            m = min(b, n - b - j)
+            if m <= 1:
+                break
+            k += 1
+            t += 1
+        counts[s] = k
+    return counts
+
+
+def band_to_tridiagonal(uplo: UpLo, band: int, mat: Matrix) -> TridiagResult:
+    """Reduce the band part of ``mat`` (lower, bandwidth ``band``) to real
+    tridiagonal form; returns the tridiagonal and the chase reflectors."""
+    assert uplo == UpLo.Lower
+    d = mat.dist
+    n = d.m
+    assert d.m == d.n
+    g = mat.grid
+    assert g is None or not g.distributed, \
+        "distributed band_to_tridiagonal lands with the distributed eigensolver"
+    A = mat.to_global().cpu()
+    b = band
+    ld = 2 * b
+    store = torch.zeros((n, ld), dtype=A.dtype)
+    for dd in range(min(b, n - 1) + 1):
+        store[: n - dd, dd] = torch.diagonal(A, -dd)
+
+    counts = _slot_counts(n, b)
+    offsets = torch.zeros(n, dtype=torch.int64)
+    if n > 1:
+        offsets[1:] = torch.cumsum(counts, 0)[:-1]
+    total = int(counts.sum().item())
+    vstore = torch.zeros((max(total, 1), b + 1), dtype=A.dtype)
+    if n > 2:
+        get_ext().band_chase(store, b, vstore, offsets)
+
+    dvec = store[:, 0]
+    evec = store[: n - 1, 1] if n > 1 else store[:0, 1]
+    phases = None
+    if A.is_complex():
+        ph = torch.ones(n, dtype=A.dtype)
+        e_abs = evec.abs()
+        for j in range(n - 1):
+            aj = e_abs[j]
+            ph[j + 1] = ph[j] * (evec[j] / aj if aj > 0 else 1.0)
+        phases = ph
+        d_real = dvec.real.clone()
+        e_real = e_abs.to(d_real.dtype)
+    else:
+        d_real = dvec.clone()
+        e_real = evec.clone()
+    return TridiagResult(d=d_real, e=e_real, band=b, n=n, vstore=vstore,
+                         offsets=offsets, nslots=counts, phases=phases)
+
+
+def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult) -> None:
+    """Back-transform E <- Q (D E) where A_band = Q (D T_real D^H) Q^H.
+
+    E: [n, nE] device tensor of tridiagonal eigenvectors, updated in place.
+    Reference: ``eigensolver/bt_band_to_tridiag/impl.h:59-1031``.
+    """
+    n, nE = E.shape
+    b = tri.band
+    dev = E.device
+    if tri.phases is not None:
+        E *= tri.phases.to(dev).unsqueeze(1)
+    if n <= 2 or tri.vstore.numel() == 0:
+        return
+    V = tri.vstore.to(dev)
+    offsets = tri.offsets
+    counts = tri.nslots
+    # pad so every sweep's blocks are uniform [K, b]
+    Epad = torch.zeros((n + b, nE), dtype=E.dtype, device=dev)
+    Epad[:n] = E
+    for s in range(n - 3, -1, -1):
+        K = int(counts[s])
+        if K == 0:
+            continue
+        o = int(offsets[s])
+        taus = V[o:o + K, 0]
+        Vs = V[o:o + K, 1:]                       # [K, b] (v0 = 1 stored)
+        r0 = 1 + s
+        seg = Epad[r0:r0 + K * b].view(K, b, nE)
+        w = Vs.conj().unsqueeze(1) @ seg          # [K, 1, nE]
+        seg -= taus.view(K, 1, 1) * (Vs.unsqueeze(2) @ w)
+    E.copy_(Epad[:n])

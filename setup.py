@@ -18,6 +18,7 @@ setup(
             name="dlaf_amd._hip",
             sources=[
                 "csrc/ext.cpp",
+                "csrc/band_chase.cpp",
                 "csrc/gemm_tiles.hip",
                 "csrc/factor.hip",
             ],
