@@ -1,14 +1,123 @@
 """Symmetric/Hermitian (generalized) eigensolver pipeline.
 
-Counterpart of ``eigensolver/eigensolver/impl.h`` (red2band -> band2tridiag ->
-tridiag D&C -> back-transforms) and ``eigensolver/gen_eigensolver/impl.h``.
+Counterpart of ``eigensolver/eigensolver/impl.h:37-105`` (HEEV: red2band ->
+band2tridiag -> tridiag D&C -> bt_band2tridiag -> bt_red2band, with partial
+spectrum) and ``eigensolver/gen_eigensolver/impl.h:30-104`` (HEGV: chol(B) ->
+hegst -> heev -> triangular back-substitution).
 """
+
 from __future__ import annotations
 
+from typing import Optional, Tuple
 
-def hermitian_eigensolver(*args, **kwargs):
-    raise NotImplementedError("hermitian_eigensolver: in progress")
+import torch
+
+from ..types import UpLo, Op, Side, Diag, is_complex, real_dtype
+from ..matrix.matrix import Matrix
+from ..comm.grid import CommGrid
+from .red2band import reduction_to_band, bt_reduction_to_band
+from .band2tridiag import band_to_tridiagonal, bt_band_to_tridiagonal
+from .tridiag_dc import tridiagonal_eigensolver
+from .cholesky import cholesky_factorization
+from .gen_to_std import generalized_to_standard
+from .triangular import triangular_solver
 
 
-def hermitian_generalized_eigensolver(*args, **kwargs):
-    raise NotImplementedError("hermitian_generalized_eigensolver: in progress")
+def get_band_size(nb: int) -> int:
+    """Band size for the two-stage reduction (reference
+    ``eigensolver/internal/get_band_size.h:9-20``: nb/divisor >= min_band=100)."""
+    band = nb
+    d = 2
+    while band % d == 0 and band // d >= 100:
+        band //= d
+    if band == nb and nb > 128:
+        # non-power-of-two nb: fall back to the largest divisor-ish cut
+        for cand in (128, 96, 112, 100):
+            if nb % cand == 0:
+                return cand
+    return band
+
+
+def hermitian_eigensolver(
+    uplo: UpLo,
+    mat: Matrix,
+    grid: Optional[CommGrid] = None,
+    band: Optional[int] = None,
+    eigenvalues_index_begin: int = 0,
+    eigenvalues_index_end: Optional[int] = None,
+) -> Tuple[torch.Tensor, Matrix]:
+    """Eigendecomposition A = E diag(w) E^H of a Hermitian tiled matrix.
+
+    ``mat`` is overwritten (band + reflectors). Returns (w [real tensor],
+    E [Matrix, same dtype/device/grid]); the partial-spectrum indices select
+    eigenvector columns (back-transforms applied only to the slice, the
+    reference's MatrixRef mechanism).
+    """
+    assert uplo == UpLo.Lower, "only Lower implemented (as the reference miniapps)"
+    d = mat.dist
+    n = d.m
+    g = grid if grid is not None else mat.grid
+    assert g is None or not g.distributed, \
+        "distributed hermitian_eigensolver lands with the distributed pipeline"
+    if band is None:
+        band = get_band_size(d.nb)
+    band = max(1, min(band, max(n - 1, 1)))
+    ib = eigenvalues_index_begin
+    ie = n if eigenvalues_index_end is None else eigenvalues_index_end
+
+    refl = reduction_to_band(mat, band)
+    tri = band_to_tridiagonal(UpLo.Lower, band, mat)
+    w, E_real = tridiagonal_eigensolver(tri.d, tri.e, device=mat.device)
+    w = w[ib:ie].clone()
+    E = E_real[:, ib:ie].to(mat.dtype).contiguous()
+    bt_band_to_tridiagonal(E, tri)
+    bt_reduction_to_band(E, mat, refl)
+
+    nE = E.shape[1]
+    evecs = Matrix.create(n, max(nE, 1), d.mb, d.nb, dtype=mat.dtype,
+                          device=mat.device, grid=mat.grid)
+    if nE:
+        _set_cols(evecs, E)
+    return w, evecs
+
+
+def _set_cols(evecs: Matrix, E: torch.Tensor) -> None:
+    de = evecs.dist
+    if (de.m, de.n) == tuple(E.shape):
+        evecs.set_from_global(E)
+        return
+    # partial spectrum: E has fewer columns than the (n x nE) matrix shape
+    full = torch.zeros((de.m, de.n), dtype=E.dtype, device=E.device)
+    full[:, : E.shape[1]] = E
+    evecs.set_from_global(full)
+
+
+def hermitian_generalized_eigensolver(
+    uplo: UpLo,
+    mat_a: Matrix,
+    mat_b: Matrix,
+    grid: Optional[CommGrid] = None,
+    factorized: bool = False,
+    eigenvalues_index_begin: int = 0,
+    eigenvalues_index_end: Optional[int] = None,
+) -> Tuple[torch.Tensor, Matrix]:
+    """Generalized problem A x = lambda B x (B HPD): returns (w, E).
+
+    ``mat_b`` is overwritten with its Cholesky factor (or is already the
+    factor when ``factorized``); ``mat_a`` is overwritten. Reference:
+    ``eigensolver/gen_eigensolver/impl.h:30-104``.
+    """
+    assert uplo == UpLo.Lower
+    g = grid if grid is not None else mat_a.grid
+    if not factorized:
+        cholesky_factorization(UpLo.Lower, mat_b, g)
+    generalized_to_standard(UpLo.Lower, mat_a, mat_b, g)
+    w, evecs = hermitian_eigensolver(
+        UpLo.Lower, mat_a, g,
+        eigenvalues_index_begin=eigenvalues_index_begin,
+        eigenvalues_index_end=eigenvalues_index_end,
+    )
+    # back-substitute: x = L^-H y
+    triangular_solver(Side.Left, UpLo.Lower, Op.ConjTrans, Diag.NonUnit, 1.0,
+                      mat_b, evecs, g)
+    return w, evecs

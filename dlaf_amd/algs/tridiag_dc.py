@@ -1,0 +1,223 @@
+"""Divide & conquer tridiagonal eigensolver (Cuppen).
+
+Counterpart of ``eigensolver/tridiag_solver/impl.h`` + ``merge.h`` (the
+reference's flagship algorithm): recursive binary split, small dense leaves,
+and rank-1 merges
+
+    T = Qb (D + rho z z^T) Qb^H,  z = [Q1 last row, Q2 first row]/sqrt(2)
+
+with (reference ``merge.h:1078-1214`` step structure):
+  1. deflation (tiny z_i; близкие d pairs via Givens rotations on Q columns)
+  2. secular-equation roots  1 + rho sum z_i^2/(d_i - λ) = 0 — here a
+     VECTORIZED bracketed-Newton iteration over all roots simultaneously
+     (device-resident), replacing the reference's per-root ``laed4`` host loop
+  3. Gu/Eisenstat z-hat recomputation (orthogonality independent of root error)
+  4. rank-1 eigenvectors + the big Qb @ U GEMM (device)
+
+All O(k^2)+ work (secular iterations, U assembly, GEMMs) runs on the compute
+device; only the O(k) deflation bookkeeping is host-side — the MI355X-native
+replacement for the reference's bulk host thread teams.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Tuple
+
+import numpy as np
+import torch
+
+_EPS = np.finfo(np.float64).eps
+
+
+def _leaf(d: np.ndarray, e: np.ndarray) -> Tuple[np.ndarray, np.ndarray]:
+    n = len(d)
+    T = np.diag(d)
+    if n > 1:
+        T += np.diag(e, -1) + np.diag(e, 1)
+    w, v = np.linalg.eigh(T)
+    return w, v
+
+
+def _secular_roots(d: torch.Tensor, z: torch.Tensor, rho: float,
+                   iters: int = 60) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Roots of 1 + rho sum z_i^2/(d_i - lam), rho > 0, d ascending, z != 0.
+
+    Returns (shift_idx [k] int64, mu [k]): lam_j = d[shift_idx_j] + mu_j; the
+    (pole, offset) representation keeps d_i - lam_j accurate near poles.
+    """
+    k = d.shape[0]
+    dev = d.device
+    z2 = z * z
+    if k == 1:
+        return torch.zeros(1, dtype=torch.int64, device=dev), rho * z2
+    # interval per root j: (d_j, d_{j+1}), last: (d_{k-1}, d_{k-1} + rho)
+    d_lo = d
+    d_hi = torch.cat([d[1:], (d[-1] + rho).reshape(1)])
+    mid = 0.5 * (d_lo + d_hi)
+    # f(mid): [k] evaluations against all poles
+    fm = 1.0 + rho * (z2.unsqueeze(1) / (d.unsqueeze(1) - mid.unsqueeze(0))).sum(0)
+    # choose shift: left pole if f(mid) >= 0 (root left of mid), else right pole
+    left = fm >= 0
+    left = torch.cat([left[:-1], torch.ones(1, dtype=torch.bool, device=dev)])  # last root: left pole
+    sidx = torch.where(left, torch.arange(k, device=dev),
+                       torch.arange(1, k + 1, device=dev).clamp(max=k - 1))
+    shift = d[sidx]
+    # bracket for mu = lam - shift
+    lo = torch.where(left, torch.zeros_like(d), mid - shift)
+    hi = torch.where(left, mid - shift, torch.zeros_like(d))
+    hi = torch.where(torch.arange(k, device=dev) == k - 1,
+                     torch.full_like(d, rho), hi)
+    delta0 = d.unsqueeze(1) - shift.unsqueeze(0)     # [k poles, k roots], exact
+    mu = 0.5 * (lo + hi)
+    for _ in range(iters):
+        diff = delta0 - mu.unsqueeze(0)              # d_i - lam_j
+        t = z2.unsqueeze(1) / diff
+        g = 1.0 + rho * t.sum(0)
+        gp = rho * (t / diff).sum(0)                 # g' in lam; dg/dmu = +g'... careful
+        # g(mu) decreasing? f(lam) = 1 + rho sum z^2/(d - lam); f' = rho sum z^2/(d-lam)^2 > 0
+        step = g / gp.clamp_min(1e-300)
+        mu_n = mu - step
+        # update bracket by sign of g (f increasing in lam/mu)
+        lo = torch.where(g < 0, mu, lo)
+        hi = torch.where(g > 0, mu, hi)
+        bad = ~torch.isfinite(mu_n) | (mu_n <= lo) | (mu_n >= hi)
+        mu = torch.where(bad, 0.5 * (lo + hi), mu_n)
+    return sidx, mu
+
+
+def _merge(w1, Q1, w2, Q2, rho, device):
+    """One rank-1 merge; inputs torch (device), returns (w, Q) device."""
+    n1 = w1.shape[0]
+    d = torch.cat([w1, w2])
+    z = torch.cat([Q1[-1, :].conj(), Q2[0, :].conj()])
+    n = d.shape[0]
+    # normalize: T = D + rho z z^T, ||z||^2 = 2 -> z/=||z||, rho*=||z||^2
+    znorm2 = float(z @ z)
+    if znorm2 == 0 or rho == 0:
+        Q = torch.zeros((n, n), dtype=Q1.dtype, device=device)
+        Q[:n1, :n1] = Q1
+        Q[n1:, n1:] = Q2
+        w, idx = torch.sort(d)
+        return w, Q[:, idx]
+    rho_eff = rho * znorm2
+    z = z / math.sqrt(znorm2)
+    # reduce to rho > 0 by negation symmetry
+    negate = rho_eff < 0
+    if negate:
+        d = -d
+        rho_eff = -rho_eff
+    # sort d ascending
+    d_s, perm = torch.sort(d)
+    z_s = z[perm]
+
+    # ---- deflation (host bookkeeping, O(k)) ----
+    dn = d_s.cpu().numpy().copy()
+    zn = z_s.cpu().numpy().copy()
+    k = n
+    dmax = max(np.abs(dn).max(), rho_eff) if k else 1.0
+    tol = 8.0 * _EPS * max(dmax, 1e-300)
+    deflated = np.zeros(k, dtype=bool)
+    rots = []  # (i, j, c, s)
+    deflated |= np.abs(rho_eff * zn) <= tol
+    last = -1
+    for i in range(k):
+        if deflated[i]:
+            continue
+        if last >= 0 and (dn[i] - dn[last]) <= tol:
+            zi, zj = zn[last], zn[i]
+            r = math.hypot(zi, zj)
+            c, s = zj / r, -zi / r
+            # zero z[last], keep z[i] = r
+            zn[i] = r
+            zn[last] = 0.0
+            di, dj = dn[last], dn[i]
+            dn[last] = di * c * c + dj * s * s
+            dn[i] = di * s * s + dj * c * c
+            rots.append((last, i, c, s))
+            deflated[last] = True
+        last = i
+
+    nd_idx = np.nonzero(~deflated)[0]
+    df_idx = np.nonzero(deflated)[0]
+    k1 = len(nd_idx)
+
+    # ---- build permuted Q (and apply rotations) on device ----
+    Qb = torch.zeros((n, n), dtype=Q1.dtype, device=device)
+    Qb[:n1, :n1] = Q1
+    Qb[n1:, n1:] = Q2
+    Qb = Qb[:, perm]
+    for (i, j, c, s) in rots:
+        gi = Qb[:, i].clone()
+        gj = Qb[:, j].clone()
+        # Qb <- Qb G with G = [[c, -s], [s, c]]; z' = G^T z zeroes component i
+        Qb[:, i] = c * gi + s * gj
+        Qb[:, j] = -s * gi + c * gj
+
+    w_out = torch.empty(n, dtype=d_s.dtype, device=device)
+    Q_out = torch.empty_like(Qb)
+
+    if k1 > 0:
+        dk = torch.from_numpy(dn[nd_idx]).to(device)
+        zk = torch.from_numpy(zn[nd_idx]).to(device)
+        sidx, mu = _secular_roots(dk, zk, rho_eff)
+        lam = dk[sidx] + mu
+        # delta[i, j] = d_i - lam_j, via exact pole differences
+        delta = (dk.unsqueeze(1) - dk[sidx].unsqueeze(0)) - mu.unsqueeze(0)
+        # Gu/Eisenstat z-hat: |zh_i|^2 = prod_j (lam_j - d_i) / prod_{j!=i} (d_j - d_i)
+        dd = dk.unsqueeze(1) - dk.unsqueeze(0)       # d_j - d_i at [i, j]... sign care
+        num = (-delta).abs().clamp_min(1e-300).log().sum(1)
+        den_m = dd.abs().clamp_min(1e-300).log()
+        den = den_m.sum(1) - torch.diagonal(den_m)
+        zh = torch.exp(0.5 * (num - den))
+        zh = torch.where(zk < 0, -zh, zh)
+        # eigenvectors of the rank-1 system
+        U = zh.unsqueeze(1) / delta                  # [k1, k1]
+        U = U / torch.linalg.vector_norm(U, dim=0, keepdim=True)
+        lam_out = lam
+        V_nd = Qb[:, torch.from_numpy(nd_idx).to(device)] @ U.to(Qb.dtype)
+    else:
+        lam_out = torch.empty(0, dtype=d_s.dtype, device=device)
+        V_nd = torch.empty((n, 0), dtype=Qb.dtype, device=device)
+
+    all_vals = torch.cat([lam_out, torch.from_numpy(dn[df_idx]).to(device)])
+    if negate:
+        all_vals = -all_vals
+    order = torch.argsort(all_vals)
+    w_out = all_vals[order]
+    V = torch.cat([V_nd, Qb[:, torch.from_numpy(df_idx).to(device)]], dim=1)
+    Q_out = V[:, order]
+    return w_out, Q_out
+
+
+def tridiagonal_eigensolver(d: torch.Tensor, e: torch.Tensor,
+                            device=None, leaf: int = 64):
+    """Eigendecomposition of a real symmetric tridiagonal matrix.
+
+    Returns (evals [n] fp, evecs [n, n]) on ``device``. Reference:
+    ``eigensolver/tridiag_solver/impl.h:198-278`` (local).
+    """
+    if device is None:
+        device = d.device
+    dn = d.detach().cpu().numpy().astype(np.float64).copy()
+    en = e.detach().cpu().numpy().astype(np.float64).copy()
+    n = len(dn)
+    if n == 0:
+        z = torch.zeros(0, dtype=torch.float64, device=device)
+        return z, torch.zeros((0, 0), dtype=torch.float64, device=device)
+
+    def solve(lo: int, hi: int):
+        m = hi - lo
+        if m <= leaf:
+            w, v = _leaf(dn[lo:hi].copy(), en[lo:hi - 1].copy())
+            return (torch.from_numpy(w).to(device), torch.from_numpy(v).to(device))
+        mid = lo + m // 2
+        rho = float(en[mid - 1])
+        # Cuppen: subtract the coupling from the two touching diagonals
+        dn[mid - 1] -= rho
+        dn[mid] -= rho
+        w1, Q1 = solve(lo, mid)
+        w2, Q2 = solve(mid, hi)
+        return _merge(w1, Q1, w2, Q2, rho, device)
+
+    return solve(0, n)
