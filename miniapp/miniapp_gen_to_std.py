@@ -1,0 +1,29 @@
+#!/usr/bin/env python3
+"""HEGST miniapp (reference ``miniapp/miniapp_gen_to_std.cpp``)."""
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
+import torch
+from _harness import run_miniapp, random_spd, random_herm
+from dlaf_amd import UpLo, generalized_to_standard, cholesky_factorization
+from dlaf_amd.types import total_ops
+
+
+def setup(ctx):
+    a = random_herm(ctx)
+    l = random_spd(ctx)
+    cholesky_factorization(UpLo.Lower, l, ctx.comm_grid)
+    return {"a": a, "l": l}
+
+
+def run(ctx, st):
+    generalized_to_standard(UpLo.Lower, st["a"], st["l"], ctx.comm_grid)
+
+
+def flops(ctx):
+    n = float(ctx.opts.matrix_size)
+    return total_ops(ctx.dtype, n**3 / 2, n**3 / 2)
+
+
+if __name__ == "__main__":
+    run_miniapp("miniapp_gen_to_std", setup, run, flops)
