@@ -1,0 +1,233 @@
+"""ScaLAPACK-style drop-in API over local block-cyclic buffers.
+
+Counterpart of the reference's C API (``include/dlaf_c/*``, ``src/c_api/*``):
+grid management by integer context, ScaLAPACK-style descriptors, and
+``p?potrf / p?potri / p?trtri / p?syevd / p?heevd / p?sygvd / p?hegvd``
+entry points operating on the caller's LOCAL block-cyclic column-major
+buffer (numpy or torch). Each call wraps the buffer into a tiled ``Matrix``
+(mirrored to the GPU when available), runs the native algorithm, and copies
+the result back — the flow of the reference's ``src/c_api/eigensolver/
+eigensolver.h:30-73`` (host matrix -> MatrixMirror -> algorithm -> copy back).
+
+Python is this framework's C-API surface (the package is the library); the
+function names and argument conventions mirror the reference's so ScaLAPACK
+callers can map 1:1.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Dict, Optional, Tuple
+
+import numpy as np
+import torch
+
+from .types import UpLo, Diag
+from .core.distribution import Distribution
+from .comm.grid import CommGrid
+from .matrix.matrix import Matrix
+from .algs.cholesky import cholesky_factorization
+from .algs.inverse import inverse_from_cholesky_factor, triangular_inverse
+from .algs.eigensolver import hermitian_eigensolver, hermitian_generalized_eigensolver
+
+
+@dataclass
+class DLAF_descriptor:
+    """ScaLAPACK-like descriptor (reference ``include/dlaf_c/desc.h``)."""
+    m: int
+    n: int
+    mb: int
+    nb: int
+    isrc: int = 0
+    jsrc: int = 0
+    i: int = 1
+    j: int = 1
+    ld: int = 0
+
+
+_grids: Dict[int, CommGrid] = {}
+_next_ctx = [1]
+
+
+def dlaf_create_grid(nprow: int, npcol: int, order: str = "R",
+                     device: Optional[torch.device] = None) -> int:
+    """Create a process grid context (reference ``dlaf_c/grid.h``).
+
+    Row-major rank order only (the reference supports both; RCCL ranks here
+    are torch.distributed ranks, which this framework orders row-major).
+    """
+    assert order.upper().startswith("R"), "row-major rank ordering only"
+    grid = CommGrid(nprow, npcol, device=device)
+    ctx = _next_ctx[0]
+    _next_ctx[0] += 1
+    _grids[ctx] = grid
+    return ctx
+
+
+def dlaf_free_grid(ctx: int) -> None:
+    _grids.pop(ctx, None)
+
+
+def _grid(ctx: int) -> CommGrid:
+    return _grids[ctx]
+
+
+def _device_for(grid: CommGrid):
+    if torch.cuda.is_available():
+        return torch.device("cuda", torch.cuda.current_device())
+    return torch.device("cpu")
+
+
+def _wrap_local(a_local, desc: DLAF_descriptor, grid: CommGrid, device) -> Matrix:
+    """Tiled device Matrix from the caller's local block-cyclic buffer."""
+    t = torch.as_tensor(a_local)
+    assert desc.i == 1 and desc.j == 1, "sub-matrix offsets not supported"
+    dist = Distribution(desc.m, desc.n, desc.mb, desc.nb,
+                        grid.grid_rows, grid.grid_cols, grid.rank_row, grid.rank_col,
+                        desc.isrc, desc.jsrc)
+    mat = Matrix(dist, t.dtype, device, grid if grid.distributed else None)
+    lr, lc = dist.local_nr_tiles
+    for li in range(lr):
+        for lj in range(lc):
+            gi, gj = dist.global_tile_of_local((li, lj))
+            tsr, tsc = dist.tile_size_of((gi, gj))
+            blk = t[li * desc.mb: li * desc.mb + tsr, lj * desc.nb: lj * desc.nb + tsc]
+            mat.storage[li, lj, :tsr, :tsc] = blk.to(device)
+    mat._set_identity_pad()
+    return mat
+
+
+def _unwrap_local(mat: Matrix, a_local) -> None:
+    t = torch.as_tensor(a_local)
+    dist = mat.dist
+    lr, lc = dist.local_nr_tiles
+    for li in range(lr):
+        for lj in range(lc):
+            gi, gj = dist.global_tile_of_local((li, lj))
+            tsr, tsc = dist.tile_size_of((gi, gj))
+            blk = mat.storage[li, lj, :tsr, :tsc].to(t.device if t.is_cuda else "cpu")
+            t[li * dist.mb: li * dist.mb + tsr, lj * dist.nb: lj * dist.nb + tsc] = blk
+    if isinstance(a_local, np.ndarray):
+        a_local[:] = t.numpy()
+
+
+def dlaf_cholesky_factorization(ctx: int, uplo: str, a_local, desc: DLAF_descriptor) -> int:
+    """``dlaf_cholesky_factorization_{s,d,c,z}`` analog; returns info (0 = ok)."""
+    assert uplo.upper() == "L"
+    grid = _grid(ctx)
+    dev = _device_for(grid)
+    mat = _wrap_local(a_local, desc, grid, dev)
+    cholesky_factorization(UpLo.Lower, mat, grid if grid.distributed else None)
+    _unwrap_local(mat, a_local)
+    return 0
+
+
+def dlaf_inverse_from_cholesky_factor(ctx: int, uplo: str, a_local,
+                                      desc: DLAF_descriptor) -> int:
+    assert uplo.upper() == "L"
+    grid = _grid(ctx)
+    mat = _wrap_local(a_local, desc, grid, _device_for(grid))
+    inverse_from_cholesky_factor(UpLo.Lower, mat, grid if grid.distributed else None)
+    _unwrap_local(mat, a_local)
+    return 0
+
+
+def dlaf_triangular_inverse(ctx: int, uplo: str, diag: str, a_local,
+                            desc: DLAF_descriptor) -> int:
+    assert uplo.upper() == "L"
+    grid = _grid(ctx)
+    mat = _wrap_local(a_local, desc, grid, _device_for(grid))
+    triangular_inverse(UpLo.Lower, Diag.Unit if diag.upper() == "U" else Diag.NonUnit,
+                       mat, grid if grid.distributed else None)
+    _unwrap_local(mat, a_local)
+    return 0
+
+
+def dlaf_hermitian_eigensolver(ctx: int, uplo: str, a_local, desc: DLAF_descriptor,
+                               w_out, z_local, descz: DLAF_descriptor,
+                               il: int = 0, iu: Optional[int] = None) -> int:
+    """``dlaf_{symmetric,hermitian}_eigensolver[_partial_spectrum]`` analog.
+
+    w_out: [n] real output buffer; z_local: local eigenvector buffer.
+    """
+    assert uplo.upper() == "L"
+    grid = _grid(ctx)
+    dev = _device_for(grid)
+    mat = _wrap_local(a_local, desc, grid, dev)
+    w, evecs = hermitian_eigensolver(UpLo.Lower, mat, grid if grid.distributed else None,
+                                     eigenvalues_index_begin=il,
+                                     eigenvalues_index_end=iu)
+    wt = torch.as_tensor(w_out)
+    wt[: w.shape[0]] = w.cpu().to(wt.dtype)
+    if isinstance(w_out, np.ndarray):
+        w_out[: w.shape[0]] = wt[: w.shape[0]].numpy()
+    _unwrap_local(evecs, z_local)
+    _unwrap_local(mat, a_local)
+    return 0
+
+
+def dlaf_hermitian_generalized_eigensolver(ctx: int, uplo: str, a_local,
+                                           desca: DLAF_descriptor, b_local,
+                                           descb: DLAF_descriptor, w_out,
+                                           z_local, descz: DLAF_descriptor,
+                                           factorized: bool = False) -> int:
+    assert uplo.upper() == "L"
+    grid = _grid(ctx)
+    dev = _device_for(grid)
+    mat_a = _wrap_local(a_local, desca, grid, dev)
+    mat_b = _wrap_local(b_local, descb, grid, dev)
+    w, evecs = hermitian_generalized_eigensolver(
+        UpLo.Lower, mat_a, mat_b, grid if grid.distributed else None,
+        factorized=factorized)
+    wt = torch.as_tensor(w_out)
+    wt[: w.shape[0]] = w.cpu().to(wt.dtype)
+    if isinstance(w_out, np.ndarray):
+        w_out[: w.shape[0]] = wt[: w.shape[0]].numpy()
+    _unwrap_local(evecs, z_local)
+    _unwrap_local(mat_b, b_local)
+    return 0
+
+
+# ---- ScaLAPACK-style shims (reference dlaf_c/...: dlaf_p{s,d,c,z}potrf etc.) ----
+
+def _sl_desc(n, mb, nb, uplo_n=None, m=None) -> DLAF_descriptor:
+    return DLAF_descriptor(m if m is not None else n, n, mb, nb)
+
+
+def pXpotrf(ctx: int, uplo: str, n: int, a_local, ia: int, ja: int,
+            desca: DLAF_descriptor) -> int:
+    assert (ia, ja) == (1, 1)
+    return dlaf_cholesky_factorization(ctx, uplo, a_local, desca)
+
+
+def pXpotri(ctx: int, uplo: str, n: int, a_local, ia: int, ja: int,
+            desca: DLAF_descriptor) -> int:
+    assert (ia, ja) == (1, 1)
+    return dlaf_inverse_from_cholesky_factor(ctx, uplo, a_local, desca)
+
+
+def pXtrtri(ctx: int, uplo: str, diag: str, n: int, a_local, ia: int, ja: int,
+            desca: DLAF_descriptor) -> int:
+    assert (ia, ja) == (1, 1)
+    return dlaf_triangular_inverse(ctx, uplo, diag, a_local, desca)
+
+
+def pXsyevd(ctx: int, uplo: str, n: int, a_local, desca: DLAF_descriptor,
+            w_out, z_local, descz: DLAF_descriptor) -> int:
+    return dlaf_hermitian_eigensolver(ctx, uplo, a_local, desca, w_out, z_local, descz)
+
+
+def pXsygvd(ctx: int, uplo: str, n: int, a_local, desca, b_local, descb,
+            w_out, z_local, descz) -> int:
+    return dlaf_hermitian_generalized_eigensolver(ctx, uplo, a_local, desca,
+                                                  b_local, descb, w_out, z_local, descz)
+
+
+# dtype-suffixed aliases matching the reference's C symbol names
+dlaf_pdpotrf = dlaf_pspotrf = dlaf_pcpotrf = dlaf_pzpotrf = pXpotrf
+dlaf_pdpotri = dlaf_pspotri = dlaf_pcpotri = dlaf_pzpotri = pXpotri
+dlaf_pdtrtri = dlaf_pstrtri = dlaf_pctrtri = dlaf_pztrtri = pXtrtri
+dlaf_pdsyevd = dlaf_pssyevd = pXsyevd
+dlaf_pcheevd = dlaf_pzheevd = pXsyevd
+dlaf_pdsygvd = dlaf_pssygvd = pXsygvd
+dlaf_pchegvd = dlaf_pzhegvd = pXsygvd

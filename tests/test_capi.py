@@ -1,0 +1,75 @@
+"""ScaLAPACK-style API tests (reference ``test/unit/c_api``)."""
+
+import numpy as np
+import pytest
+import torch
+
+from dlaf_amd import capi
+from dlaf_amd.capi import DLAF_descriptor
+
+from dist_utils import run_distributed
+
+
+def _local_blockcyclic(a, mb, nb, gr, gc, pr, pc):
+    """Local block-cyclic part of global a for rank (pr, pc)."""
+    m, n = a.shape
+    rows = [i for i in range(m) if (i // mb) % gr == pr]
+    cols = [j for j in range(n) if (j // nb) % gc == pc]
+    return a[np.ix_(rows, cols)].copy(), rows, cols
+
+
+def test_capi_potrf_local():
+    n, nb = 24, 8
+    rng = np.random.default_rng(3)
+    a = rng.standard_normal((n, n))
+    a = a @ a.T + n * np.eye(n)
+    loc = a.copy()
+    ctx = capi.dlaf_create_grid(1, 1)
+    info = capi.dlaf_cholesky_factorization(ctx, "L", loc, DLAF_descriptor(n, n, nb, nb))
+    capi.dlaf_free_grid(ctx)
+    assert info == 0
+    want = np.linalg.cholesky(a)
+    assert np.abs(np.tril(loc) - want).max() < 1e-11 * n
+
+
+def test_capi_syevd_local():
+    n, nb = 20, 5
+    rng = np.random.default_rng(5)
+    a = rng.standard_normal((n, n))
+    a = (a + a.T) / 2
+    loc = np.tril(a).copy()
+    w = np.zeros(n)
+    z = np.zeros((n, n))
+    ctx = capi.dlaf_create_grid(1, 1)
+    info = capi.pXsyevd(ctx, "L", n, loc, DLAF_descriptor(n, n, nb, nb), w, z,
+                        DLAF_descriptor(n, n, nb, nb))
+    capi.dlaf_free_grid(ctx)
+    assert info == 0
+    wref = np.linalg.eigvalsh(a)
+    assert np.abs(np.sort(w) - wref).max() < 1e-11 * n
+    res = np.abs(a @ z - z * w).max()
+    assert res < 1e-10 * n
+
+
+def _dist_capi_worker(rank, ws, gr, gc):
+    n, nb = 24, 4
+    rng = np.random.default_rng(7)
+    a = rng.standard_normal((n, n))
+    a = a @ a.T + n * np.eye(n)
+    pr, pc = rank // gc, rank % gc
+    loc, rows, cols = _local_blockcyclic(a, nb, nb, gr, gc, pr, pc)
+    ctx = capi.dlaf_create_grid(gr, gc)
+    info = capi.dlaf_pdpotrf(ctx, "L", n, loc, 1, 1, DLAF_descriptor(n, n, nb, nb))
+    capi.dlaf_free_grid(ctx)
+    want = np.linalg.cholesky(a)
+    want_loc = want[np.ix_(rows, cols)]
+    mask = np.zeros((n, n), dtype=bool)
+    mask[np.tril_indices(n)] = True
+    mloc = mask[np.ix_(rows, cols)]
+    return float(np.abs((loc - want_loc)[mloc]).max())
+
+
+def test_capi_potrf_dist():
+    errs = run_distributed(_dist_capi_worker, 4, args=(2, 2))
+    for e in errs:
+        assert e < 1e-10, f"err={e}"
