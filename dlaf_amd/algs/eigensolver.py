@@ -57,11 +57,15 @@ def hermitian_eigensolver(
     d = mat.dist
     n = d.m
     g = grid if grid is not None else mat.grid
-    assert g is None or not g.distributed, \
-        "distributed hermitian_eigensolver lands with the distributed pipeline"
     if band is None:
         band = get_band_size(d.nb)
     band = max(1, min(band, max(n - 1, 1)))
+    if g is not None and g.distributed:
+        from .eigensolver_dist import hermitian_eigensolver_dist
+        return hermitian_eigensolver_dist(
+            uplo, mat, g, band,
+            eigenvalues_index_begin=eigenvalues_index_begin,
+            eigenvalues_index_end=eigenvalues_index_end)
     ib = eigenvalues_index_begin
     ie = n if eigenvalues_index_end is None else eigenvalues_index_end
 

@@ -1,0 +1,227 @@
+"""Distributed Hermitian eigensolver (single-node multi-GPU, RCCL/xGMI).
+
+Counterpart of the reference's distributed eigensolver pipeline
+(``eigensolver/reduction_to_band/impl.h:1150-1516``, ``band_to_tridiag/mc.h:
+990-1559``, ``tridiag_solver/impl.h:363-480``, both back-transforms) with an
+MI355X-first decomposition that exploits 288 GB HBM3E per GPU:
+
+* the O(n^3) stages are SHARDED: reduction-to-band trailing updates by
+  cyclic row chunks (each rank updates its rows; per-panel comm = one panel
+  all-reduce + one X all-reduce + a b x b all-reduce), and the two
+  back-transforms by eigenvector column stripes (embarrassingly parallel);
+* the O(n^2 b) / O(n^2) serial stages (panel QR, bulge chasing, D&C scalar
+  work) are REPLICATED in rank-lockstep — identical inputs give bitwise
+  identical results, so no broadcast is needed and no rank waits on another
+  (the reference pays p2p pipelines for the same stages because its targets
+  have ~1/10th the per-GPU memory);
+* collectives are a handful of large contiguous all-reduces per panel over
+  the full RCCL communicator — sized for xGMI's 7x153 GB/s point-to-point
+  fan-out rather than many small messages.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+import torch
+
+from ..types import UpLo
+from ..matrix.matrix import Matrix
+from ..comm.grid import CommGrid
+from ..comm import collectives as coll
+from .red2band import panel_qr_, t_factor, _herm_full_dense
+from .band2tridiag import band_to_tridiagonal, bt_band_to_tridiagonal, TridiagResult
+from .tridiag_dc import tridiagonal_eigensolver
+
+
+class _ChunkMap:
+    """Cyclic row chunks of size cb; chunk c owned by rank c % P."""
+
+    def __init__(self, n: int, cb: int, rank: int, world: int):
+        self.n = n
+        self.cb = cb
+        self.rank = rank
+        self.world = world
+        self.owned: List[Tuple[int, int]] = []
+        for c in range((n + cb - 1) // cb):
+            if c % world == rank:
+                self.owned.append((c * cb, min((c + 1) * cb, n)))
+
+    def owned_in(self, lo: int) -> List[Tuple[int, int]]:
+        """Owned row ranges clipped to [lo, n)."""
+        out = []
+        for a, b in self.owned:
+            if b > lo:
+                out.append((max(a, lo), b))
+        return out
+
+
+def _assemble(full: torch.Tensor, group) -> torch.Tensor:
+    """All-reduce-SUM assembly of a zero-filled-except-owned tensor."""
+    coll.all_reduce_sum(full, group)
+    return full
+
+
+def _red2band_dist(A: torch.Tensor, band: int, cm: _ChunkMap, group):
+    """Sharded dense reduction to band. A: [n, n] device; only the caller's
+    owned rows are authoritative. Returns the refl dict of red2band."""
+    n = A.shape[0]
+    dt, dev = A.dtype, A.device
+    taus_all, panels = [], []
+    for j0 in range(0, max(n - band, 0), band):
+        r0 = j0 + band
+        bw = min(band, n - j0 - band)
+        if bw <= 0:
+            break
+        m_p = n - r0
+        # assemble the panel from its row owners, replicate via all-reduce
+        P = torch.zeros((m_p, bw), dtype=dt, device=dev)
+        for a, b in cm.owned_in(r0):
+            P[a - r0:b - r0] = A[a:b, j0:j0 + bw]
+        _assemble(P, group)
+        nrefl = min(m_p, bw)
+        taus = torch.zeros(nrefl, dtype=dt, device=dev)
+        panel_qr_(P, taus)          # replicated, lockstep-identical
+        # write the factored panel back into owned rows (R + V storage)
+        for a, b in cm.owned_in(r0):
+            A[a:b, j0:j0 + bw] = P[a - r0:b - r0]
+        panels.append((j0, bw, nrefl))
+        taus_all.append(taus)
+        V = torch.tril(P[:, :nrefl], -1) + torch.eye(m_p, nrefl, dtype=dt, device=dev)
+        T = t_factor(V, taus)
+        W = V @ T                                   # replicated (m b^2)
+        # Y = A22 W by owned row slabs; X assembled by all-reduce
+        Y = torch.zeros((m_p, nrefl), dtype=dt, device=dev)
+        for a, b in cm.owned_in(r0):
+            Y[a - r0:b - r0] = A[a:b, r0:] @ W
+        # S = T^H (V^H Y): partial over owned rows then all-reduce
+        VHY = torch.zeros((nrefl, nrefl), dtype=dt, device=dev)
+        for a, b in cm.owned_in(r0):
+            VHY += V[a - r0:b - r0].mH @ Y[a - r0:b - r0]
+        _assemble(VHY, group)
+        S = T.mH @ VHY
+        X = torch.zeros((m_p, nrefl), dtype=dt, device=dev)
+        for a, b in cm.owned_in(r0):
+            X[a - r0:b - r0] = Y[a - r0:b - r0] - 0.5 * V[a - r0:b - r0] @ S
+        _assemble(X, group)
+        # trailing update on owned rows (full width of A22)
+        XH = X.mH
+        VH = V.mH
+        for a, b in cm.owned_in(r0):
+            A[a:b, r0:] -= V[a - r0:b - r0] @ XH + X[a - r0:b - r0] @ VH
+    return {"taus": taus_all, "panels": panels, "band": band}
+
+
+def _extract_band_dist(A: torch.Tensor, band: int, cm: _ChunkMap, group) -> torch.Tensor:
+    """Replicated [n, n_keep] dense band+V columns are NOT needed — only the
+    band diagonals; assembled from row owners."""
+    n = A.shape[0]
+    ld = 2 * band
+    store = torch.zeros((n, ld), dtype=A.dtype, device=A.device)
+    for a, b in cm.owned:
+        for d in range(min(band, n - 1) + 1):
+            lo = max(a, d)
+            if lo >= b:
+                continue
+            rows = torch.arange(lo, b, device=A.device)
+            store[rows - d, d] = A[rows, rows - d]
+    _assemble(store, group)
+    return store
+
+
+def _bt_red2band_dist(E: torch.Tensor, A: torch.Tensor, refl, cm: _ChunkMap,
+                      group) -> None:
+    """E <- Q E for the caller's eigenvector column stripe; V panels are
+    assembled from their row owners per panel (reverse order)."""
+    band = refl["band"]
+    n = A.shape[0]
+    dt, dev = A.dtype, A.device
+    for (j0, bw, nrefl), taus in zip(reversed(refl["panels"]), reversed(refl["taus"])):
+        r0 = j0 + band
+        m_p = n - r0
+        P = torch.zeros((m_p, nrefl), dtype=dt, device=dev)
+        for a, b in cm.owned_in(r0):
+            P[a - r0:b - r0] = A[a:b, j0:j0 + nrefl]
+        _assemble(P, group)
+        V = torch.tril(P, -1) + torch.eye(m_p, nrefl, dtype=dt, device=dev)
+        T = t_factor(V, taus)
+        W = T @ (V.mH @ E[r0:, :])
+        E[r0:, :] -= V @ W
+
+
+def hermitian_eigensolver_dist(uplo: UpLo, mat: Matrix, grid: CommGrid,
+                               band: int,
+                               eigenvalues_index_begin: int = 0,
+                               eigenvalues_index_end: Optional[int] = None
+                               ) -> Tuple[torch.Tensor, Matrix]:
+    """Distributed HEEV over the full communicator of ``grid``."""
+    assert uplo == UpLo.Lower
+    d = mat.dist
+    n = d.m
+    dev = mat.device
+    world = grid.world_size
+    rank = grid.rank
+    group = grid.full_group
+    ib = eigenvalues_index_begin
+    ie = n if eigenvalues_index_end is None else eigenvalues_index_end
+
+    # replicate the input (to_global all-reduces the tiled parts)
+    A = _herm_full_dense(mat.to_global())
+    cm = _ChunkMap(n, d.nb, rank, world)
+
+    refl = _red2band_dist(A, band, cm, group)
+
+    # band -> tridiagonal: replicated CPU chase (identical on every rank)
+    store = _extract_band_dist(A, band, cm, group).cpu()
+    from .band2tridiag import _slot_counts
+    counts = _slot_counts(n, band)
+    offsets = torch.zeros(n, dtype=torch.int64)
+    if n > 1:
+        offsets[1:] = torch.cumsum(counts, 0)[:-1]
+    total = int(counts.sum().item())
+    vstore = torch.zeros((max(total, 1), band + 1), dtype=store.dtype)
+    if n > 2:
+        from ..ops._ext import get_ext
+        get_ext().band_chase(store, band, vstore, offsets)
+    dvec = store[:, 0]
+    evec = store[: n - 1, 1] if n > 1 else store[:0, 1]
+    phases = None
+    if A.is_complex():
+        ph = torch.ones(n, dtype=A.dtype)
+        e_abs = evec.abs()
+        for j in range(n - 1):
+            aj = e_abs[j]
+            ph[j + 1] = ph[j] * (evec[j] / aj if aj > 0 else 1.0)
+        phases = ph
+        d_real, e_real = dvec.real.clone(), e_abs.to(dvec.real.dtype)
+    else:
+        d_real, e_real = dvec.clone(), evec.clone()
+    tri = TridiagResult(d=d_real, e=e_real, band=band, n=n, vstore=vstore,
+                        offsets=offsets, nslots=counts, phases=phases)
+
+    # tridiag D&C: replicated (scalar stages), GEMM merges on the local GPU
+    w, E_real = tridiagonal_eigensolver(tri.d, tri.e, device=dev)
+    w = w[ib:ie].clone()
+    nE = int(ie - ib)
+
+    # back-transforms on this rank's contiguous column stripe of [ib, ie)
+    c0 = ib + (nE * rank) // world
+    c1 = ib + (nE * (rank + 1)) // world
+    E_slice = E_real[:, c0:c1].to(mat.dtype).contiguous()
+    bt_band_to_tridiagonal(E_slice, tri)
+    _bt_red2band_dist(E_slice, A, refl, cm, group)
+
+    # assemble the full eigenvector block and scatter into the tiled output
+    E_full = torch.zeros((n, nE), dtype=mat.dtype, device=dev)
+    E_full[:, c0 - ib:c1 - ib] = E_slice
+    _assemble(E_full, group)
+
+    evecs = Matrix.create(n, max(nE, 1), d.mb, d.nb, dtype=mat.dtype,
+                          device=dev, grid=grid)
+    if nE:
+        full = torch.zeros((n, evecs.dist.n), dtype=mat.dtype, device=dev)
+        full[:, :nE] = E_full
+        evecs.set_from_global(full)
+    # leave the band + reflectors in mat (parity with the in-place contract)
+    mat.set_from_global(A)
+    return w, evecs

@@ -76,3 +76,50 @@ def test_gen_eigensolver_local(dtype):
     # B-orthogonality
     borth = (E.mH @ b0 @ E - torch.eye(n, dtype=dtype)).abs().max().item()
     assert borth < 1e-9 * n, f"borth={borth}"
+
+
+# ---------------- distributed (gloo) ----------------
+
+from dlaf_amd import CommGrid  # noqa: E402
+from dist_utils import run_distributed  # noqa: E402
+
+
+def _dist_eig_worker(rank, ws, gr, gc, n, nb, dtype_str, gen):
+    dtype = getattr(torch, dtype_str)
+    grid = CommGrid(gr, gc)
+    A = Matrix.create(n, n, nb, nb, dtype=dtype, grid=grid)
+    mutil.set_random_hermitian(A, seed=71)
+    a0 = _herm(A.to_global())
+    if gen:
+        B = Matrix.create(n, n, nb, nb, dtype=dtype, grid=grid)
+        mutil.set_random_hermitian_positive_definite(B, seed=72)
+        b0 = _herm(B.to_global())
+        w, evecs = hermitian_generalized_eigensolver(UpLo.Lower, A, B, grid)
+        E = evecs.to_global()
+        res = (a0 @ E - b0 @ E @ torch.diag(w.to(dtype))).abs().max().item()
+        wref = sl.eigh(a0.numpy(), b0.numpy(), eigvals_only=True)
+    else:
+        w, evecs = hermitian_eigensolver(UpLo.Lower, A, grid, band=nb)
+        E = evecs.to_global()
+        res = (a0 @ E - E @ torch.diag(w.to(dtype))).abs().max().item()
+        wref = np.linalg.eigvalsh(a0.numpy())
+    werr = np.abs(np.sort(w.numpy()) - wref).max()
+    return res, werr
+
+
+@pytest.mark.parametrize("gr,gc", [(1, 2), (2, 2)])
+@pytest.mark.parametrize("dtype_str", ["float64", "complex128"])
+def test_eigensolver_dist_cpu(gr, gc, dtype_str):
+    n = 32
+    out = run_distributed(_dist_eig_worker, gr * gc, args=(gr, gc, n, 8, dtype_str, False))
+    for res, werr in out:
+        assert res < 1e-9 * n, f"res={res}"
+        assert werr < 1e-10 * n, f"werr={werr}"
+
+
+def test_gen_eigensolver_dist_cpu():
+    n = 24
+    out = run_distributed(_dist_eig_worker, 2, args=(1, 2, n, 8, "float64", True))
+    for res, werr in out:
+        assert res < 1e-8 * n, f"res={res}"
+        assert werr < 1e-9 * n, f"werr={werr}"
