@@ -38,6 +38,41 @@ def bench_gemm(dtype=torch.float64, nb=512, ntiles=128, iters=10, opB=Op.Trans):
           f"{flops/dt_s/1e12:.2f} TFLOP/s")
 
 
+def bench_torch_bmm(dtype=torch.float64, nb=512, ntiles=128, iters=10):
+    """rocBLAS batched GEMM baseline for the same shape (library comparison)."""
+    dev = "cuda"
+    A = torch.randn(ntiles, nb, nb, dtype=dtype, device=dev) if not dtype.is_complex else (
+        torch.randn(ntiles, nb, nb, dtype=torch.float64, device=dev)
+        + 1j * torch.randn(ntiles, nb, nb, dtype=torch.float64, device=dev)).to(dtype)
+    C = torch.zeros_like(A)
+    for _ in range(3):
+        torch.bmm(A, A.mT, out=C)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        torch.bmm(A, A.mT, out=C)
+    torch.cuda.synchronize()
+    dt_s = (time.perf_counter() - t0) / iters
+    flops = 2.0 * ntiles * nb * nb * nb * (4 if dtype.is_complex else 1)
+    print(f"torch.bmm {dtype} nb={nb} ntiles={ntiles}: {dt_s*1e3:.2f} ms  "
+          f"{flops/dt_s/1e12:.2f} TFLOP/s")
+
+
+def bench_big_dgemm(dtype=torch.float64, n=16384, iters=5):
+    """Single large library DGEMM: the machine speed-of-light reference."""
+    a = torch.randn(n, n, dtype=dtype, device="cuda")
+    c = torch.zeros_like(a)
+    for _ in range(2):
+        torch.mm(a, a.mT, out=c)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        torch.mm(a, a.mT, out=c)
+    torch.cuda.synchronize()
+    dt_s = (time.perf_counter() - t0) / iters
+    print(f"torch.mm {dtype} n={n}: {dt_s*1e3:.1f} ms  {2*n**3/dt_s/1e12:.2f} TFLOP/s")
+
+
 def bench_potrf(dtype=torch.float64, nb=512, iters=20):
     a = torch.randn(nb, nb, dtype=dtype, device="cuda")
     a = a @ a.mH + nb * torch.eye(nb, dtype=dtype, device="cuda")
@@ -77,6 +112,9 @@ if __name__ == "__main__":
     bench_gemm(torch.float64, 512, 1024, iters=5)
     bench_gemm(torch.float32, 512, 128)
     bench_gemm(torch.complex128, 512, 64)
+    bench_torch_bmm(torch.float64, 512, 128)
+    bench_torch_bmm(torch.float64, 512, 1024, iters=5)
+    bench_big_dgemm(torch.float64, 16384)
     bench_potrf(torch.float64)
     bench_potrf(torch.complex128, 512)
     bench_trtri(torch.float64)
