@@ -1,5 +1,6 @@
 """Timing breakdown of the eigensolver pipeline on GPU (SYEV config 4 shape)."""
-import sys, time
+import os, sys, time
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 import torch
 from dlaf_amd import Matrix, UpLo
 from dlaf_amd.matrix import util as mutil
