@@ -40,7 +40,7 @@ def _leaf(d: np.ndarray, e: np.ndarray) -> Tuple[np.ndarray, np.ndarray]:
 
 
 def _secular_roots(d: torch.Tensor, z: torch.Tensor, rho: float,
-                   iters: int = 60) -> Tuple[torch.Tensor, torch.Tensor]:
+                   iters: int = 40) -> Tuple[torch.Tensor, torch.Tensor]:
     """Roots of 1 + rho sum z_i^2/(d_i - lam), rho > 0, d ascending, z != 0.
 
     Returns (shift_idx [k] int64, mu [k]): lam_j = d[shift_idx_j] + mu_j; the
@@ -70,17 +70,45 @@ def _secular_roots(d: torch.Tensor, z: torch.Tensor, rho: float,
                      torch.full_like(d, rho), hi)
     delta0 = d.unsqueeze(1) - shift.unsqueeze(0)     # [k poles, k roots], exact
     mu = 0.5 * (lo + hi)
+    # pole-split masks for the laed4-style two-pole rational model:
+    # psi = poles i <= j, phi = poles i > j
+    idx = torch.arange(k, device=dev)
+    below = idx.unsqueeze(1) <= idx.unsqueeze(0)     # [pole i, root j]
+    jj = idx                                          # left pole index = j
+    j2 = (idx + 1).clamp(max=k - 1)                   # right pole (last: unused)
+    last = idx == k - 1
     for _ in range(iters):
         diff = delta0 - mu.unsqueeze(0)              # d_i - lam_j
         t = z2.unsqueeze(1) / diff
-        g = 1.0 + rho * t.sum(0)
-        gp = rho * (t / diff).sum(0)                 # g' in lam; dg/dmu = +g'... careful
-        # g(mu) decreasing? f(lam) = 1 + rho sum z^2/(d - lam); f' = rho sum z^2/(d-lam)^2 > 0
-        step = g / gp.clamp_min(1e-300)
-        mu_n = mu - step
-        # update bracket by sign of g (f increasing in lam/mu)
-        lo = torch.where(g < 0, mu, lo)
-        hi = torch.where(g > 0, mu, hi)
+        t2 = t / diff
+        f = 1.0 + rho * t.sum(0)
+        psi_p = torch.where(below, t2, torch.zeros_like(t2)).sum(0)
+        phi_p = torch.where(below, torch.zeros_like(t2), t2).sum(0)
+        d1 = delta0[jj, idx] - mu                    # d_j - lam (negative side)
+        d2 = delta0[j2, idx] - mu                    # d_{j+1} - lam
+        P = rho * psi_p * d1 * d1
+        Q = rho * phi_p * d2 * d2
+        c = f - rho * psi_p * d1 - rho * phi_p * d2
+        # solve c + P/(d1 - s) + Q/(d2 - s) = 0 (interior);
+        # quadratic a s^2 + b s + c2 = 0
+        a = c
+        b = -(c * (d1 + d2) + P + Q)
+        c2 = c * d1 * d2 + P * d2 + Q * d1
+        disc = (b * b - 4.0 * a * c2).clamp_min(0.0).sqrt()
+        # stable quadratic roots; select the one inside the pole gap (d1, d2)
+        qq = -0.5 * (b + torch.where(b >= 0, disc, -disc))
+        safe_a = torch.where(a.abs() < 1e-300, torch.full_like(a, 1e-300), a)
+        safe_q = torch.where(qq.abs() < 1e-300, torch.full_like(qq, 1e-300), qq)
+        r1 = qq / safe_a
+        r2 = c2 / safe_q
+        in1 = (r1 > d1) & (r1 < d2)
+        s_int = torch.where(in1, r1, r2)
+        # last root: one-pole model c + P/(d1 - s) = 0
+        s_last = d1 + P / torch.where(c.abs() < 1e-300, torch.full_like(c, 1e-300), c)
+        s = torch.where(last, s_last, s_int)
+        mu_n = mu + s
+        lo = torch.where(f < 0, mu, lo)
+        hi = torch.where(f > 0, mu, hi)
         bad = ~torch.isfinite(mu_n) | (mu_n <= lo) | (mu_n >= hi)
         mu = torch.where(bad, 0.5 * (lo + hi), mu_n)
     return sidx, mu
