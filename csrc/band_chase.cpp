@@ -25,8 +25,10 @@
 
 #include <torch/extension.h>
 
+#include <atomic>
 #include <cmath>
 #include <complex>
+#include <thread>
 #include <vector>
 
 namespace {
@@ -155,6 +157,63 @@ void apply_left(int64_t m, int64_t nn, T tau, const T* v, T* a, int64_t ld,
   }
 }
 
+// One full sweep s (start + all chase steps). When `done` is non-null the
+// sweep PIPELINES against its predecessor (wavefront parallelization, the
+// reference's semaphore scheme mc.h:666-693): step t of sweep s may only run
+// once sweep s-1 has completed step t+3, which keeps the two sweeps' band
+// windows disjoint (gap b-1 columns).
+template <class T>
+void run_sweep(T* a, int64_t ld, int64_t size, int64_t b, T* vstore,
+               const int64_t* offsets, int64_t s, T* v, T* w,
+               std::atomic<int32_t>* done) {
+  const int64_t vstride = b + 1;
+  auto wait_pred = [&](int64_t t) {
+    if (done && s > 0) {
+      while (done[s - 1].load(std::memory_order_acquire) < t + 3)
+        std::this_thread::yield();
+    }
+  };
+  wait_pred(0);
+  int64_t n0 = std::min(size - s - 1, b);
+  T tau = hh_reflector(n0, a + 1 + s * ld);
+  {
+    T* slot = vstore + offsets[s] * vstride;
+    slot[0] = tau;
+    slot[1] = T(1);
+    for (int64_t i = 1; i < n0; ++i) slot[1 + i] = a[1 + i + s * ld];
+    for (int64_t i = n0; i < b; ++i) slot[1 + i] = T(0);
+  }
+  v[0] = T(1);
+  for (int64_t i = 1; i < n0; ++i) v[i] = a[1 + i + s * ld];
+  for (int64_t i = 1; i < n0; ++i) a[1 + i + s * ld] = T(0);
+
+  int64_t step = 0;
+  while (true) {
+    wait_pred(step + 1);
+    int64_t j = 1 + s + step * b;
+    int64_t nn = std::min(b, size - j);
+    int64_t m = std::min(b, size - b - j);
+    apply_two_sided(nn, tau, v, a + j * ld, ld, w);
+    if (m > 0) apply_right(m, nn, tau, v, a + j * ld, ld, nn, w);
+    if (m <= 1) break;
+    tau = hh_reflector(m, a + nn + j * ld);
+    ++step;
+    {
+      T* slot = vstore + (offsets[s] + step) * vstride;
+      slot[0] = tau;
+      slot[1] = T(1);
+      for (int64_t i = 1; i < m; ++i) slot[1 + i] = a[nn + i + j * ld];
+      for (int64_t i = m; i < b; ++i) slot[1 + i] = T(0);
+    }
+    v[0] = T(1);
+    for (int64_t i = 1; i < m; ++i) v[i] = a[nn + i + j * ld];
+    for (int64_t i = 1; i < m; ++i) a[nn + i + j * ld] = T(0);
+    apply_left(m, nn - 1, tau, v, a + (nn - 1) + (j + 1) * ld, ld, 0, w);
+    if (done) done[s].store(step, std::memory_order_release);
+  }
+  if (done) done[s].store(INT32_MAX, std::memory_order_release);
+}
+
 // Sequential chase of the whole band matrix.
 // band: [ld, size] column-major torch tensor (band.stride(1) == ld... we use
 //   contiguous [size, ld] row-major = column-major [ld, size]).
@@ -162,55 +221,28 @@ void apply_left(int64_t m, int64_t nn, T tau, const T* v, T* a, int64_t ld,
 //   slot k of sweep s lives at (offsets[s] + k) * (b + 1).
 template <class T>
 void chase_impl(T* a, int64_t ld, int64_t size, int64_t b, T* vstore,
-                const int64_t* offsets) {
-  std::vector<T> v(b + 1), w(b);
-  int64_t vstride = b + 1;
+                const int64_t* offsets, int nthreads) {
   // sweeps eliminate columns 0 .. size-3 (column size-2 is already tridiagonal)
-  for (int64_t s = 0; s + 3 <= size; ++s) {
-    // start sweep: eliminate column s below the first subdiagonal
-    int64_t n0 = std::min(size - s - 1, b);
-    T tau = hh_reflector(n0, a + 1 + s * ld);
-    // copy out reflector (v0 = 1 implicit in storage too: slot = tau, 1, tail)
-    {
-      T* slot = vstore + offsets[s] * vstride;
-      slot[0] = tau;
-      slot[1] = T(1);
-      for (int64_t i = 1; i < n0; ++i) slot[1 + i] = a[1 + i + s * ld];
-      for (int64_t i = n0; i < b; ++i) slot[1 + i] = T(0);
-    }
-    v[0] = T(1);
-    for (int64_t i = 1; i < n0; ++i) v[i] = a[1 + i + s * ld];
-    // zero the eliminated entries in storage (they are now implicit)
-    for (int64_t i = 1; i < n0; ++i) a[1 + i + s * ld] = T(0);
-
-    int64_t step = 0;
-    int64_t nref = n0;
-    while (true) {
-      int64_t j = 1 + s + step * b;
-      int64_t nn = std::min(b, size - j);
-      int64_t m = std::min(b, size - b - j);
-      // two-sided on the diag block with the current reflector (length nref)
-      apply_two_sided(nn, tau, v.data(), a + j * ld, ld, w.data());
-      if (m > 0) apply_right(m, nn, tau, v.data(), a + j * ld, ld, nn, w.data());
-      if (m <= 1) break;
-      // new reflector from the spilled column j (rows j+nn .. j+nn+m)
-      tau = hh_reflector(m, a + nn + j * ld);
-      ++step;
-      {
-        T* slot = vstore + (offsets[s] + step) * vstride;
-        slot[0] = tau;
-        slot[1] = T(1);
-        for (int64_t i = 1; i < m; ++i) slot[1 + i] = a[nn + i + j * ld];
-        for (int64_t i = m; i < b; ++i) slot[1 + i] = T(0);
-      }
-      v[0] = T(1);
-      for (int64_t i = 1; i < m; ++i) v[i] = a[nn + i + j * ld];
-      for (int64_t i = 1; i < m; ++i) a[nn + i + j * ld] = T(0);
-      nref = m;
-      apply_left(m, nn - 1, tau, v.data(), a + (nn - 1) + (j + 1) * ld, ld, 0,
-                 w.data());
-    }
+  int64_t nsweeps = size - 2;
+  if (nsweeps <= 0) return;
+  if (nthreads <= 1 || nsweeps < 4) {
+    std::vector<T> v(b + 1), w(b);
+    for (int64_t s = 0; s < nsweeps; ++s)
+      run_sweep(a, ld, size, b, vstore, offsets, s, v.data(), w.data(), nullptr);
+    return;
   }
+  std::vector<std::atomic<int32_t>> done(nsweeps);
+  for (auto& d : done) d.store(0, std::memory_order_relaxed);
+  std::vector<std::thread> pool;
+  for (int tid = 0; tid < nthreads; ++tid) {
+    pool.emplace_back([=, &done]() {
+      std::vector<T> v(b + 1), w(b);
+      for (int64_t s = tid; s < nsweeps; s += nthreads)
+        run_sweep(a, ld, size, b, vstore, offsets, s, v.data(), w.data(),
+                  done.data());
+    });
+  }
+  for (auto& t : pool) t.join();
 }
 
 }  // namespace
@@ -219,7 +251,11 @@ void chase_impl(T* a, int64_t ld, int64_t size, int64_t b, T* vstore,
 //   band[j][d] = A[j+d, j], d = 0..ld-1, ld >= 2b (bulge headroom).
 // vstore: [total_slots, b+1]; offsets: [size] int64 slot offsets per sweep.
 void band_chase(torch::Tensor band, int64_t b, torch::Tensor vstore,
-                torch::Tensor offsets) {
+                torch::Tensor offsets, int64_t nthreads) {
+  if (nthreads <= 0) {
+    nthreads = std::min<int64_t>(16, std::thread::hardware_concurrency());
+    if (nthreads < 1) nthreads = 1;
+  }
   TORCH_CHECK(!band.is_cuda(), "band_chase is a CPU stage (as the reference)");
   TORCH_CHECK(band.is_contiguous() && vstore.is_contiguous());
   int64_t size = band.size(0);
@@ -236,14 +272,15 @@ void band_chase(torch::Tensor band, int64_t b, torch::Tensor vstore,
           chase_impl(reinterpret_cast<std::complex<float>*>(band.data_ptr<T>()),
                      ld, size, b,
                      reinterpret_cast<std::complex<float>*>(vstore.data_ptr<T>()),
-                     offs);
+                     offs, (int)nthreads);
         } else if constexpr (std::is_same_v<T, c10::complex<double>>) {
           chase_impl(reinterpret_cast<std::complex<double>*>(band.data_ptr<T>()),
                      ld, size, b,
                      reinterpret_cast<std::complex<double>*>(vstore.data_ptr<T>()),
-                     offs);
+                     offs, (int)nthreads);
         } else {
-          chase_impl(band.data_ptr<T>(), ld, size, b, vstore.data_ptr<T>(), offs);
+          chase_impl(band.data_ptr<T>(), ld, size, b, vstore.data_ptr<T>(), offs,
+                     (int)nthreads);
         }
       });
 }

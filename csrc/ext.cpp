@@ -268,11 +268,13 @@ void factor_invert_block(torch::Tensor A, int64_t n, int64_t ld,
 }  // namespace
 
 void band_chase(torch::Tensor band, int64_t b, torch::Tensor vstore,
-                torch::Tensor offsets);
+                torch::Tensor offsets, int64_t nthreads);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("band_chase", &band_chase,
-        "CPU bulge chasing band->tridiag with reflector recording");
+        "CPU bulge chasing band->tridiag with reflector recording",
+        py::arg("band"), py::arg("b"), py::arg("vstore"), py::arg("offsets"),
+        py::arg("nthreads") = 0);
   m.def("batch_gemm", &batch_gemm,
         "fused batched tile GEMM: C[d] = alpha*op(A[d])op(B[d]) + beta*C[d]");
   m.def("potrf_block", &potrf_block, "single-workgroup Cholesky block factor");

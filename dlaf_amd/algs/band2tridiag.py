@@ -98,11 +98,21 @@ def band_to_tridiagonal(uplo: UpLo, band: int, mat: Matrix) -> TridiagResult:
                          offsets=offsets, nslots=counts, phases=phases)
 
 
-def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult) -> None:
+def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
+                           group_size: int = 64) -> None:
     """Back-transform E <- Q (D E) where A_band = Q (D T_real D^H) Q^H.
 
     E: [n, nE] device tensor of tridiagonal eigenvectors, updated in place.
-    Reference: ``eigensolver/bt_band_to_tridiag/impl.h:59-1031``.
+
+    Reference: ``eigensolver/bt_band_to_tridiag/impl.h:59-1031`` — including
+    its HH-apply GROUPING (``hh_apply_group_size``, tune.h): reflectors of
+    ``group_size`` consecutive sweeps at the same chase depth k form one
+    staircase V panel applied via a compact-WY T factor (3 GEMMs), so E is
+    traversed once per GROUP instead of once per sweep — arithmetic intensity
+    up by group_size. Ordering: groups in reverse sweep order; within a
+    group, windows k ascending (cross-window overlapping reflector pairs are
+    exactly those the ascending order keeps correctly ordered; the disjoint
+    pairs commute).
     """
     n, nE = E.shape
     b = tri.band
@@ -112,20 +122,46 @@ def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult) -> None:
     if n <= 2 or tri.vstore.numel() == 0:
         return
     V = tri.vstore.to(dev)
-    offsets = tri.offsets
-    counts = tri.nslots
-    # pad so every sweep's blocks are uniform [K, b]
-    Epad = torch.zeros((n + b, nE), dtype=E.dtype, device=dev)
+    offsets = tri.offsets.to(dev)
+    counts = tri.nslots.to(dev)
+    nsweeps = n - 2
+    G = max(1, min(group_size, nsweeps))
+    H = b + G - 1                       # window height
+    Epad = torch.zeros((n + b + G, nE), dtype=E.dtype, device=dev)
     Epad[:n] = E
-    for s in range(n - 3, -1, -1):
-        K = int(counts[s])
-        if K == 0:
+    eyeG = torch.eye(G, dtype=E.dtype, device=dev)
+    group_starts = list(range(0, nsweeps, G))
+    for s0 in reversed(group_starts):
+        Gc = min(G, nsweeps - s0)       # sweeps in this group
+        counts_g = counts[s0:s0 + Gc]
+        nwin = int(counts_g.max())
+        if nwin == 0:
             continue
-        o = int(offsets[s])
-        taus = V[o:o + K, 0]
-        Vs = V[o:o + K, 1:]                       # [K, b] (v0 = 1 stored)
-        r0 = 1 + s
-        seg = Epad[r0:r0 + K * b].view(K, b, nE)
-        w = Vs.conj().unsqueeze(1) @ seg          # [K, 1, nE]
-        seg -= taus.view(K, 1, 1) * (Vs.unsqueeze(2) @ w)
+        # staircase V panels for all windows: Vg[k, row, g]
+        Vg = torch.zeros((nwin, H, G), dtype=E.dtype, device=dev)
+        taus_g = torch.zeros((nwin, G), dtype=E.dtype, device=dev)
+        for g in range(Gc):
+            Kg = int(counts_g[g])
+            if Kg == 0:
+                continue
+            o = int(offsets[s0 + g])
+            taus_g[:Kg, g] = V[o:o + Kg, 0]
+            Vg[:Kg, g:g + b, g] = V[o:o + Kg, 1:]
+        # batched T factors: T = inv(diag(1/tau) + striu(V^H V)); tau=0 rows/cols vanish
+        Gram = Vg.mH @ Vg                                   # [nwin, G, G]
+        zc = taus_g == 0
+        safe = torch.where(zc, torch.ones_like(taus_g), taus_g)
+        M = torch.triu(Gram, 1) + torch.diag_embed(1.0 / safe)
+        mask = zc.unsqueeze(1) | zc.unsqueeze(2)
+        M = torch.where(mask, torch.zeros_like(M), M)
+        M = M + torch.diag_embed(torch.where(zc, torch.ones_like(safe),
+                                             torch.zeros_like(safe)))
+        T = torch.linalg.solve_triangular(M, eyeG.expand(nwin, G, G), upper=True)
+        T = torch.where(mask, torch.zeros_like(T), T)
+        # apply windows in ascending k (ordering constraint across overlaps)
+        for k in range(nwin):
+            base = 1 + s0 + k * b
+            seg = Epad[base:base + H]
+            W = T[k] @ (Vg[k].mH @ seg)
+            seg -= Vg[k] @ W
     E.copy_(Epad[:n])

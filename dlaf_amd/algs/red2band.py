@@ -37,7 +37,7 @@ def _herm_full_dense(a: torch.Tensor) -> torch.Tensor:
 
 
 def larfg(x: torch.Tensor) -> Tuple[torch.Tensor, complex]:
-    """LAPACK-convention Householder generator, in place.
+    """LAPACK-convention Householder generator, in place (host-sync variant).
 
     x[0] <- beta (real), x[1:] <- v tail (v0 = 1 implicit); returns tau.
     H = I - tau v v^H with H^H x = beta e1.
@@ -59,23 +59,53 @@ def larfg(x: torch.Tensor) -> Tuple[torch.Tensor, complex]:
     return x[0], tau
 
 
+def _larfg_device(x: torch.Tensor, taus: torch.Tensor, j: int) -> torch.Tensor:
+    """Branchless device larfg: no host synchronization.
+
+    Writes beta into x[0], v-tail into x[1:], tau into taus[j]; returns tau as
+    a 0-dim device tensor. Degenerate columns (nothing to eliminate) get
+    tau = 0 and x untouched, matching LAPACK.
+    """
+    alpha = x[0].clone()
+    tail = x[1:]
+    n2 = (tail.conj() * tail).sum().real if x.is_complex() else (tail * tail).sum()
+    a_re = alpha.real if x.is_complex() else alpha
+    a_im = alpha.imag if x.is_complex() else torch.zeros_like(a_re)
+    degen = (n2 == 0) & (a_im == 0)
+    mag = torch.sqrt(a_re * a_re + a_im * a_im + n2)
+    beta = torch.where(a_re < 0, mag, -mag)
+    safe_beta = torch.where(degen, torch.ones_like(beta), beta)
+    if x.is_complex():
+        tau = torch.complex((safe_beta - a_re) / safe_beta, -a_im / safe_beta)
+        denom = alpha - safe_beta.to(alpha.dtype)
+    else:
+        tau = (safe_beta - a_re) / safe_beta
+        denom = alpha - safe_beta
+    tau = torch.where(degen, torch.zeros_like(tau), tau)
+    scale = torch.where(degen, torch.ones_like(denom), 1.0 / denom)
+    tail.mul_(scale)
+    x[0] = torch.where(degen, alpha, beta.to(alpha.dtype))
+    taus[j] = tau
+    return tau
+
+
 def panel_qr_(P: torch.Tensor, taus: torch.Tensor) -> None:
     """In-place unblocked QR of a tall panel, LAPACK geqrf convention.
 
     R in the upper triangle, reflector tails below the diagonal (unit
-    implicit), taus filled. Device-resident column loop (the panel is the
-    O(n b^2) term; the O(n^2 b) work is in the caller's GEMMs).
+    implicit), taus filled. Fully device-resident: no host synchronization in
+    the column loop (the panel is the O(n b^2) term; the O(n^2 b) work is in
+    the caller's GEMMs).
     """
     m, nb = P.shape
+    one = torch.ones(1, dtype=P.dtype, device=P.device)
     for j in range(min(m, nb)):
-        _, tau = larfg(P[j:, j])
-        taus[j] = tau
-        if tau != 0.0 and j + 1 < nb:
+        tau = _larfg_device(P[:, j] if j == 0 else P[j:, j], taus, j)
+        if j + 1 < nb:
             # apply H^H = I - conj(tau) v v^H from the left (zgeqr2 convention)
-            v = torch.cat([torch.ones(1, dtype=P.dtype, device=P.device), P[j + 1:, j]])
+            v = torch.cat([one, P[j + 1:, j]])
             w = v.conj() @ P[j:, j + 1:]
-            ct = tau.conjugate() if isinstance(tau, complex) else tau
-            P[j:, j + 1:] -= ct * torch.outer(v, w)
+            P[j:, j + 1:] -= (tau.conj() * torch.outer(v, w))
 
 
 def t_factor(V: torch.Tensor, taus: torch.Tensor) -> torch.Tensor:
