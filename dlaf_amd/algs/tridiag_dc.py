@@ -111,6 +111,41 @@ def _secular_roots(d: torch.Tensor, z: torch.Tensor, rho: float,
         hi = torch.where(f > 0, mu, hi)
         bad = ~torch.isfinite(mu_n) | (mu_n <= lo) | (mu_n >= hi)
         mu = torch.where(bad, 0.5 * (lo + hi), mu_n)
+    # Illinois (modified regula falsi) polish: the rational step can creep on
+    # near-pole roots; Illinois has guaranteed bracket shrinkage and
+    # superlinear convergence (f is monotone increasing in mu). Seed the
+    # bracket with the rational iteration's (usually excellent) final point.
+    big = torch.full_like(mu, 1e300)
+    flo, fhi = -big, big
+    diff = delta0 - mu.unsqueeze(0)
+    fmu = 1.0 + rho * (z2.unsqueeze(1) / diff).sum(0)
+    neg0 = fmu < 0
+    lo = torch.where(neg0, mu, lo)
+    flo = torch.where(neg0, fmu, flo)
+    hi = torch.where(neg0, hi, mu)
+    fhi = torch.where(neg0, fhi, fmu)
+    side = torch.zeros_like(mu)           # -1 last update was lo, +1 was hi
+    for _ in range(24):
+        denom = fhi - flo
+        x = torch.where(denom.abs() > 0, (lo * fhi - hi * flo) / denom,
+                        0.5 * (lo + hi))
+        inside = (x > lo) & (x < hi) & torch.isfinite(x)
+        x = torch.where(inside, x, 0.5 * (lo + hi))
+        diff = delta0 - x.unsqueeze(0)
+        t = z2.unsqueeze(1) / diff
+        fx = 1.0 + rho * t.sum(0)
+        neg = fx < 0
+        # Illinois halving when the same endpoint is kept twice
+        fhi = torch.where(neg & (side < 0), 0.5 * fhi, fhi)
+        flo = torch.where(~neg & (side > 0), 0.5 * flo, flo)
+        lo = torch.where(neg, x, lo)
+        flo = torch.where(neg, fx, flo)
+        hi = torch.where(neg, hi, x)
+        fhi = torch.where(neg, fhi, fx)
+        side = torch.where(neg, -torch.ones_like(side), torch.ones_like(side))
+        mu = x
+    # final: return the bracket point with smaller |f|
+    mu = torch.where(flo.abs() < fhi.abs(), lo, hi)
     return sidx, mu
 
 
@@ -192,12 +227,13 @@ def _merge(w1, Q1, w2, Q2, rho, device):
         lam = dk[sidx] + mu
         # delta[i, j] = d_i - lam_j, via exact pole differences
         delta = (dk.unsqueeze(1) - dk[sidx].unsqueeze(0)) - mu.unsqueeze(0)
-        # Gu/Eisenstat z-hat: |zh_i|^2 = prod_j (lam_j - d_i) / prod_{j!=i} (d_j - d_i)
-        dd = dk.unsqueeze(1) - dk.unsqueeze(0)       # d_j - d_i at [i, j]... sign care
+        # Gu/Eisenstat z-hat:
+        #   zh_i^2 = prod_j (lam_j - d_i) / (rho * prod_{j!=i} (d_j - d_i))
+        dd = dk.unsqueeze(1) - dk.unsqueeze(0)       # d_j - d_i at [i, j]
         num = (-delta).abs().clamp_min(1e-300).log().sum(1)
         den_m = dd.abs().clamp_min(1e-300).log()
         den = den_m.sum(1) - torch.diagonal(den_m)
-        zh = torch.exp(0.5 * (num - den))
+        zh = torch.exp(0.5 * (num - den - math.log(rho_eff)))
         zh = torch.where(zk < 0, -zh, zh)
         # eigenvectors of the rank-1 system
         U = zh.unsqueeze(1) / delta                  # [k1, k1]
