@@ -270,7 +270,53 @@ void factor_invert_block(torch::Tensor A, int64_t n, int64_t ld,
 void band_chase(torch::Tensor band, int64_t b, torch::Tensor vstore,
                 torch::Tensor offsets, int64_t nthreads);
 
+extern "C" {
+int panel_qr_f64(double*, long, int, long, double*, double*, double*, hipStream_t);
+int panel_qr_f32(float*, long, int, long, float*, float*, float*, hipStream_t);
+int panel_qr_c128(double*, long, int, long, double*, double*, double*, hipStream_t);
+int panel_qr_c64(float*, long, int, long, float*, float*, float*, hipStream_t);
+}
+
+// Cooperative whole-panel QR (see csrc/panel_qr.hip). P: 2D device view with
+// unit column stride; taus: [min(m,nb)]; norms/wraw: zeroed workspaces.
+void panel_qr(torch::Tensor P, torch::Tensor taus, torch::Tensor norms,
+              torch::Tensor wraw) {
+  TORCH_CHECK(P.is_cuda() && P.dim() == 2 && P.stride(1) == 1);
+  long m = P.size(0);
+  int nb = (int)P.size(1);
+  long ldp = P.stride(0);
+  auto stream = at::cuda::getCurrentHIPStream().stream();
+  int rc = -1;
+  switch (P.scalar_type()) {
+    case torch::kFloat64:
+      rc = panel_qr_f64((double*)P.data_ptr(), m, nb, ldp,
+                        (double*)taus.data_ptr(), (double*)norms.data_ptr(),
+                        (double*)wraw.data_ptr(), stream);
+      break;
+    case torch::kFloat32:
+      rc = panel_qr_f32((float*)P.data_ptr(), m, nb, ldp,
+                        (float*)taus.data_ptr(), (float*)norms.data_ptr(),
+                        (float*)wraw.data_ptr(), stream);
+      break;
+    case torch::kComplexDouble:
+      rc = panel_qr_c128((double*)P.data_ptr(), m, nb, ldp,
+                         (double*)taus.data_ptr(), (double*)norms.data_ptr(),
+                         (double*)wraw.data_ptr(), stream);
+      break;
+    case torch::kComplexFloat:
+      rc = panel_qr_c64((float*)P.data_ptr(), m, nb, ldp,
+                        (float*)taus.data_ptr(), (float*)norms.data_ptr(),
+                        (float*)wraw.data_ptr(), stream);
+      break;
+    default:
+      TORCH_CHECK(false, "unsupported dtype");
+  }
+  TORCH_CHECK(rc == 0, "panel_qr cooperative launch failed, hipError ", rc);
+  HIP_CHECK(hipGetLastError());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("panel_qr", &panel_qr, "cooperative whole-panel QR (one launch)");
   m.def("band_chase", &band_chase,
         "CPU bulge chasing band->tridiag with reflector recording",
         py::arg("band"), py::arg("b"), py::arg("vstore"), py::arg("offsets"),

@@ -93,11 +93,19 @@ def panel_qr_(P: torch.Tensor, taus: torch.Tensor) -> None:
     """In-place unblocked QR of a tall panel, LAPACK geqrf convention.
 
     R in the upper triangle, reflector tails below the diagonal (unit
-    implicit), taus filled. Fully device-resident: no host synchronization in
-    the column loop (the panel is the O(n b^2) term; the O(n^2 b) work is in
-    the caller's GEMMs).
+    implicit), taus filled. GPU: ONE cooperative kernel launch for the whole
+    panel (csrc/panel_qr.hip — the column loop with its grid syncs runs on
+    device). CPU: torch column loop (the reference backend).
     """
     m, nb = P.shape
+    if P.is_cuda and P.stride(1) == 1 and m > 0 and nb > 0:
+        from ..ops._ext import get_ext
+        rdt = torch.float32 if P.dtype in (torch.float32, torch.complex64) \
+            else torch.float64
+        norms = torch.zeros(nb, dtype=rdt, device=P.device)
+        wraw = torch.zeros(2 * nb * nb, dtype=P.dtype, device=P.device)
+        get_ext().panel_qr(P, taus, norms, wraw)
+        return
     one = torch.ones(1, dtype=P.dtype, device=P.device)
     for j in range(min(m, nb)):
         tau = _larfg_device(P[:, j] if j == 0 else P[j:, j], taus, j)

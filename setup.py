@@ -21,6 +21,7 @@ setup(
                 "csrc/band_chase.cpp",
                 "csrc/gemm_tiles.hip",
                 "csrc/factor.hip",
+                "csrc/panel_qr.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3"],

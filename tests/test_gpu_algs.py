@@ -193,3 +193,28 @@ def test_max_norm_gpu():
     mutil.set_random(mat, seed=4)
     a = mat.to_global().cpu()
     assert abs(max_norm(mat) - a.abs().max().item()) < 1e-14
+
+
+def test_panel_qr_kernel_gpu():
+    """Cooperative whole-panel QR vs the CPU torch column loop."""
+    from dlaf_amd.algs.red2band import panel_qr_
+    for dtype in [torch.float64, torch.complex128, torch.float32]:
+        for (m, nb) in [(1024, 128), (777, 64), (128, 128), (100, 128)]:
+            g = torch.Generator().manual_seed(17)
+            P0 = torch.randn(m, nb, generator=g, dtype=torch.float64)
+            if dtype.is_complex:
+                P0 = (P0 + 1j * torch.randn(m, nb, generator=g, dtype=torch.float64))
+            P0 = P0.to(dtype)
+            ncols = min(m, nb)
+            Pc = P0.clone()
+            tc = torch.zeros(ncols, dtype=dtype)
+            panel_qr_(Pc, tc)
+            Pg = P0.cuda()
+            tg = torch.zeros(ncols, dtype=dtype, device="cuda")
+            panel_qr_(Pg, tg)
+            torch.cuda.synchronize()
+            tol = 1e-4 if dtype == torch.float32 else 1e-11
+            perr = (Pg.cpu() - Pc).abs().max().item()
+            terr = (tg.cpu() - tc).abs().max().item()
+            assert perr < tol * m, f"{dtype} {m}x{nb} P err={perr}"
+            assert terr < tol * m, f"{dtype} {m}x{nb} tau err={terr}"
