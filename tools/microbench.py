@@ -96,11 +96,11 @@ def bench_trtri(dtype=torch.float64, nb=512, iters=20):
         2 * nb * torch.eye(nb, dtype=dtype, device="cuda")
     T = torch.empty_like(L)
     for _ in range(3):
-        ops.trtri_tile(L, T)
+        ops.tri_inverse_full(L, lower=True)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
-        ops.trtri_tile(L, T)
+        ops.tri_inverse_full(L, lower=True)
     torch.cuda.synchronize()
     dt_s = (time.perf_counter() - t0) / iters
     print(f"trtri_tile {dtype} nb={nb}: {dt_s*1e6:.0f} us")
