@@ -221,8 +221,11 @@ def _merge(w1, Q1, w2, Q2, rho, device):
     Q_out = torch.empty_like(Qb)
 
     if k1 > 0:
-        dk = torch.from_numpy(dn[nd_idx]).to(device)
-        zk = torch.from_numpy(zn[nd_idx]).to(device)
+        # size hybrid: the secular solve is O(iters * k^2) in ~20 small tensor
+        # ops per iteration — launch-bound on GPU below a few thousand roots
+        sec_dev = device if k1 >= 2048 else torch.device("cpu")
+        dk = torch.from_numpy(dn[nd_idx]).to(sec_dev)
+        zk = torch.from_numpy(zn[nd_idx]).to(sec_dev)
         sidx, mu = _secular_roots(dk, zk, rho_eff)
         lam = dk[sidx] + mu
         # delta[i, j] = d_i - lam_j, via exact pole differences
@@ -238,8 +241,8 @@ def _merge(w1, Q1, w2, Q2, rho, device):
         # eigenvectors of the rank-1 system
         U = zh.unsqueeze(1) / delta                  # [k1, k1]
         U = U / torch.linalg.vector_norm(U, dim=0, keepdim=True)
-        lam_out = lam
-        V_nd = Qb[:, torch.from_numpy(nd_idx).to(device)] @ U.to(Qb.dtype)
+        lam_out = lam.to(device)
+        V_nd = Qb[:, torch.from_numpy(nd_idx).to(device)] @ U.to(Qb.dtype).to(device)
     else:
         lam_out = torch.empty(0, dtype=d_s.dtype, device=device)
         V_nd = torch.empty((n, 0), dtype=Qb.dtype, device=device)
