@@ -115,22 +115,38 @@ __global__ void panel_qr_kernel(T* __restrict__ P, long m, int nb, long ldp,
   for (int j = 0; j < ncols; ++j) {
     T* wj = wraw + (long)j * nb;
     T* wrow = wraw + (long)(nb + j) * nb;
-    // ---- phase A ----
+    // ---- phase A: each block OWNS whole output values (no cross-block
+    // atomics — an atomics formulation serializes on <= nb addresses and
+    // measured 48 ms/panel). Block q' reduces column q = j + q' over all
+    // tail rows; q' == 0 computes the norm of column j instead. ----
     {
-      R part = R(0);
-      for (long i = j + 1 + gid0; i < m; i += gstride) part += abs2v(P[i * ldp + j]);
-      __shared__ R sred[256];
-      sred[tid] = part;
-      __syncthreads();
-      for (int s = nthreads / 2; s > 0; s >>= 1) {
-        if (tid < s) sred[tid] += sred[tid + s];
+      __shared__ R sredr[256];
+      __shared__ T sredt[256];
+      for (int q = j + (int)blockIdx.x; q < nb; q += gridDim.x) {
+        if (q == j) {
+          R part = R(0);
+          for (long i = j + 1 + tid; i < m; i += nthreads)
+            part += abs2v(P[i * ldp + j]);
+          sredr[tid] = part;
+          __syncthreads();
+          for (int s = nthreads / 2; s > 0; s >>= 1) {
+            if (tid < s) sredr[tid] += sredr[tid + s];
+            __syncthreads();
+          }
+          if (tid == 0) norms[j] = sredr[0];
+        } else {
+          T part = ScalarTraits<T>::zero();
+          for (long i = j + 1 + tid; i < m; i += nthreads)
+            part += conjv(P[i * ldp + j]) * P[i * ldp + q];
+          sredt[tid] = part;
+          __syncthreads();
+          for (int s = nthreads / 2; s > 0; s >>= 1) {
+            if (tid < s) sredt[tid] += sredt[tid + s];
+            __syncthreads();
+          }
+          if (tid == 0) wj[q] = sredt[0];
+        }
         __syncthreads();
-      }
-      if (tid == 0 && sred[0] != R(0)) atomic_addT(&norms[j], sred[0]);
-      for (long idx = gid0; idx < (m - j - 1) * (long)(nb - j - 1); idx += gstride) {
-        long i = j + 1 + idx / (nb - j - 1);
-        int q = j + 1 + (int)(idx % (nb - j - 1));
-        atomic_addT(&wj[q], conjv(P[i * ldp + j]) * P[i * ldp + q]);
       }
       // stage row j and the diagonal
       for (long q = j + gid0; q < nb; q += gstride) wrow[q] = P[(long)j * ldp + q];

@@ -40,7 +40,7 @@ def _leaf(d: np.ndarray, e: np.ndarray) -> Tuple[np.ndarray, np.ndarray]:
 
 
 def _secular_roots(d: torch.Tensor, z: torch.Tensor, rho: float,
-                   iters: int = 40) -> Tuple[torch.Tensor, torch.Tensor]:
+                   iters: int = 18) -> Tuple[torch.Tensor, torch.Tensor]:
     """Roots of 1 + rho sum z_i^2/(d_i - lam), rho > 0, d ascending, z != 0.
 
     Returns (shift_idx [k] int64, mu [k]): lam_j = d[shift_idx_j] + mu_j; the
@@ -125,7 +125,7 @@ def _secular_roots(d: torch.Tensor, z: torch.Tensor, rho: float,
     hi = torch.where(neg0, hi, mu)
     fhi = torch.where(neg0, fhi, fmu)
     side = torch.zeros_like(mu)           # -1 last update was lo, +1 was hi
-    for _ in range(24):
+    for _ in range(14):
         denom = fhi - flo
         x = torch.where(denom.abs() > 0, (lo * fhi - hi * flo) / denom,
                         0.5 * (lo + hi))
@@ -221,9 +221,9 @@ def _merge(w1, Q1, w2, Q2, rho, device):
     Q_out = torch.empty_like(Qb)
 
     if k1 > 0:
-        # size hybrid: the secular solve is O(iters * k^2) in ~20 small tensor
-        # ops per iteration — launch-bound on GPU below a few thousand roots
-        sec_dev = device if k1 >= 2048 else torch.device("cpu")
+        # small merges stay on CPU tensors (launch overhead dominates);
+        # anything sizeable runs on the device
+        sec_dev = device if k1 >= 192 else torch.device("cpu")
         dk = torch.from_numpy(dn[nd_idx]).to(sec_dev)
         zk = torch.from_numpy(zn[nd_idx]).to(sec_dev)
         sidx, mu = _secular_roots(dk, zk, rho_eff)
