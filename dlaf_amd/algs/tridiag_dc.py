@@ -40,7 +40,7 @@ def _leaf(d: np.ndarray, e: np.ndarray) -> Tuple[np.ndarray, np.ndarray]:
 
 
 def _secular_roots(d: torch.Tensor, z: torch.Tensor, rho: float,
-                   iters: int = 18) -> Tuple[torch.Tensor, torch.Tensor]:
+                   iters: int = 40) -> Tuple[torch.Tensor, torch.Tensor]:
     """Roots of 1 + rho sum z_i^2/(d_i - lam), rho > 0, d ascending, z != 0.
 
     Returns (shift_idx [k] int64, mu [k]): lam_j = d[shift_idx_j] + mu_j; the
@@ -77,11 +77,17 @@ def _secular_roots(d: torch.Tensor, z: torch.Tensor, rho: float,
     jj = idx                                          # left pole index = j
     j2 = (idx + 1).clamp(max=k - 1)                   # right pole (last: unused)
     last = idx == k - 1
-    for _ in range(iters):
+    dscale = float(d.abs().max()) + rho
+    for it in range(iters):
         diff = delta0 - mu.unsqueeze(0)              # d_i - lam_j
         t = z2.unsqueeze(1) / diff
         t2 = t / diff
         f = 1.0 + rho * t.sum(0)
+        if it >= 12 and it % 6 == 0:
+            # vectorized convergence check (one sync every 6 iterations)
+            fp_est = rho * t2.sum(0)
+            if float((f.abs() / fp_est.clamp_min(1e-300)).max()) < 1e-15 * dscale:
+                break
         psi_p = torch.where(below, t2, torch.zeros_like(t2)).sum(0)
         phi_p = torch.where(below, torch.zeros_like(t2), t2).sum(0)
         d1 = delta0[jj, idx] - mu                    # d_j - lam (negative side)
@@ -125,7 +131,7 @@ def _secular_roots(d: torch.Tensor, z: torch.Tensor, rho: float,
     hi = torch.where(neg0, hi, mu)
     fhi = torch.where(neg0, fhi, fmu)
     side = torch.zeros_like(mu)           # -1 last update was lo, +1 was hi
-    for _ in range(14):
+    for _ in range(24):
         denom = fhi - flo
         x = torch.where(denom.abs() > 0, (lo * fhi - hi * flo) / denom,
                         0.5 * (lo + hi))
