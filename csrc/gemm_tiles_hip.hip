@@ -56,20 +56,20 @@ struct Mfma<float> {
 //   BN = 128 ("in-place-safe"): wave tile 64x64 -> a single column block spans
 //        the whole N of a panel-apply, making X = X * dinv^H safe in place
 //        (every workgroup reads all of its A rows before writing them).
-template <typename T, int OPA, int OPB, int BN>
+template <typename T, int OPA, int OPB, int BN, int BK = 16, bool DBUF = false>
 __launch_bounds__(256) __global__ void gemm_tiles_k(
     const GemmDesc* __restrict__ descs, const T* __restrict__ A,
     const T* __restrict__ B, T* __restrict__ C, int M, int N, int K, int lda,
     int ldb, int ldc, T alpha, T beta, int mblocks, int nblocks) {
-  constexpr int BM = 128, BK = 16;
+  constexpr int BM = 128;
   constexpr int LA = BM * BK / 256;  // elements staged per thread (A)
   constexpr int LB = BK * BN / 256;
   constexpr int WCW = BN / 2;        // wave tile columns
   constexpr int NFRAG = WCW / 16;
   using acc_t = typename Mfma<T>::acc_t;
 
-  __shared__ T As[BM][BK + 1];
-  __shared__ T Bs[BK][BN + 2];
+  __shared__ T As[DBUF ? 2 : 1][BM][BK + 1];
+  __shared__ T Bs[DBUF ? 2 : 1][BK][BN + 2];
 
   const int wg = blockIdx.x;
   const int per_desc = mblocks * nblocks;
@@ -135,10 +135,7 @@ __launch_bounds__(256) __global__ void gemm_tiles_k(
     }
   };
 
-  load_step(0, ra, rb);
-
-  for (int s = 0; s < total_steps; ++s) {
-    // regs -> LDS
+  auto stage = [&](int buf, T* va, T* vb) {
 #pragma unroll
     for (int j = 0; j < LA; ++j) {
       const int e = j * 256 + tid;
@@ -150,7 +147,7 @@ __launch_bounds__(256) __global__ void gemm_tiles_k(
         k = e / BM;
         i = e % BM;
       }
-      As[i][k] = ra[j];
+      As[buf][i][k] = va[j];
     }
 #pragma unroll
     for (int j = 0; j < LB; ++j) {
@@ -163,25 +160,48 @@ __launch_bounds__(256) __global__ void gemm_tiles_k(
         c = e / BK;
         k = e % BK;
       }
-      Bs[k][c] = rb[j];
+      Bs[buf][k][c] = vb[j];
     }
-    __syncthreads();
-    if (s + 1 < total_steps) load_step(s + 1, ra, rb);  // overlaps MFMA below
-
+  };
+  auto compute = [&](int buf) {
 #pragma unroll
     for (int ks = 0; ks < BK / 4; ++ks) {
       T af[4], bf[NFRAG];
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) af[mi] = As[wrow + mi * 16 + li][ks * 4 + lk];
+      for (int mi = 0; mi < 4; ++mi)
+        af[mi] = As[buf][wrow + mi * 16 + li][ks * 4 + lk];
 #pragma unroll
-      for (int ni = 0; ni < NFRAG; ++ni) bf[ni] = Bs[ks * 4 + lk][wcol + ni * 16 + li];
+      for (int ni = 0; ni < NFRAG; ++ni)
+        bf[ni] = Bs[buf][ks * 4 + lk][wcol + ni * 16 + li];
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
         for (int ni = 0; ni < NFRAG; ++ni)
           acc[mi][ni] = Mfma<T>::mma(af[mi], bf[ni], acc[mi][ni]);
     }
-    __syncthreads();  // LDS reuse barrier
+  };
+
+  load_step(0, ra, rb);
+
+  if constexpr (DBUF) {
+    // one barrier per K-step: compute buf s&1 while staging s+1 into 1-(s&1)
+    stage(0, ra, rb);
+    __syncthreads();
+    for (int s = 0; s < total_steps; ++s) {
+      const int cur = s & 1;
+      if (s + 1 < total_steps) load_step(s + 1, ra, rb);
+      compute(cur);
+      if (s + 1 < total_steps) stage(1 - cur, ra, rb);
+      __syncthreads();
+    }
+  } else {
+    for (int s = 0; s < total_steps; ++s) {
+      stage(0, ra, rb);
+      __syncthreads();
+      if (s + 1 < total_steps) load_step(s + 1, ra, rb);  // overlaps MFMA below
+      compute(0);
+      __syncthreads();  // LDS reuse barrier
+    }
   }
 
   // epilogue: C = alpha*acc + beta*C
@@ -397,6 +417,12 @@ void launch_real(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
                  int M, int N, int K, int lda, int ldb, int ldc, int opA,
                  int opB, T alpha, T beta, hipStream_t stream, int inplace) {
   if (ndesc <= 0 || M <= 0 || N <= 0) return;
+  // kernel variant (microbench lever): 0 = BK16 single-buffer (default),
+  // 1 = BK32 single-buffer, 2 = BK16 double-buffer (one barrier per step)
+  static const int variant = [] {
+    const char* v = getenv("DLAF_GEMM_VARIANT");
+    return v ? atoi(v) : 0;
+  }();
   // N <= 64 is a single column block even at BN=64, hence in-place safe.
   const int BN = (inplace && N > 64) ? 128 : 64;
   const int mblocks = (M + 127) / 128, nblocks = (N + BN - 1) / BN;
@@ -405,20 +431,31 @@ void launch_real(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
   // OP_C == OP_T for real scalars
   const int oa = (opA == OP_C) ? OP_T : opA;
   const int ob = (opB == OP_C) ? OP_T : opB;
+#define LAUNCH(OA, OB, BNv, BKv, DB)                                         \
+ hipLaunchKernelGGL(( gemm_tiles_k<T, OA, OB, BNv, BKv, DB>), dim3(grid), dim3(block), 0, stream,          \
+      descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks, nblocks)
 #define CASE(OA, OB)                                                        \
   if (oa == OA && ob == OB) {                                               \
-    if (inplace)                                                            \
-     hipLaunchKernelGGL(( gemm_tiles_k<T, OA, OB, 128>), dim3(grid), dim3(block), 0, stream,              \
-          descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks,     \
-          nblocks);                                                         \
-    else                                                                    \
-     hipLaunchKernelGGL(( gemm_tiles_k<T, OA, OB, 64>), dim3(grid), dim3(block), 0, stream,               \
-          descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks,     \
-          nblocks);                                                         \
+    if (inplace) {                                                          \
+      if (variant == 1)                                                     \
+        LAUNCH(OA, OB, 128, 32, false);                                     \
+      else if (variant == 2)                                                \
+        LAUNCH(OA, OB, 128, 16, true);                                      \
+      else                                                                  \
+        LAUNCH(OA, OB, 128, 16, false);                                     \
+    } else {                                                                \
+      if (variant == 1)                                                     \
+        LAUNCH(OA, OB, 64, 32, false);                                      \
+      else if (variant == 2)                                                \
+        LAUNCH(OA, OB, 64, 16, true);                                       \
+      else                                                                  \
+        LAUNCH(OA, OB, 64, 16, false);                                      \
+    }                                                                       \
     return;                                                                 \
   }
   CASE(OP_N, OP_N) CASE(OP_N, OP_T) CASE(OP_T, OP_N) CASE(OP_T, OP_T)
 #undef CASE
+#undef LAUNCH
 }
 
 template <typename T>
