@@ -259,15 +259,20 @@ def _dist_plan(mat: Matrix) -> _DescTable:
             rows_g = [d.global_tile_of_local((li, 0))[0] for li in range(li0, lr)]
             offs = np.array([mat.tile_offset((i, k)) for i in rows_g], dtype=np.int64)
             _trsm_plan_rows(table, ("k", k), offs, nb, nb, bsz)
-        # trailing on local tiles
+        # trailing on local tiles, split for lookahead: head = global col k+1
+        # (unblocks step k+1's panel), tail = the bulk
         lj0 = d.next_local_tile_col(k + 1)
+        ch, ah, bh = [], [], []
         c, a, b = [], [], []
         for lj in range(lj0, lc):
             j = d.global_tile_of_local((0, lj))[1]
+            dst = (ch, ah, bh) if j == k + 1 else (c, a, b)
             for li in range(d.next_local_tile_row(j), lr):
-                c.append(mat.local_tile_offset(li, lj))
-                a.append(li * ts)
-                b.append(lj * ts)
+                dst[0].append(mat.local_tile_offset(li, lj))
+                dst[1].append(li * ts)
+                dst[2].append(lj * ts)
+        if ch:
+            table.add(("k", k, "head"), _rows6(np.array(ch), np.array(ah), np.array(bh)))
         if c:
             table.add(("k", k, "trail"), _rows6(np.array(c), np.array(a), np.array(b)))
     table.upload(mat.device)
@@ -280,13 +285,14 @@ def _cholesky_dist(mat: Matrix, grid: CommGrid) -> None:
     nt = d.nr_tiles[0]
     nb = d.nb
     gpu = mat.device.type == "cuda"
+    if gpu:
+        _cholesky_dist_gpu(mat, grid)
+        return
     opc = _op_conj(mat.dtype)
 
-    table = _dist_plan(mat) if gpu else None
     col_panel = Panel(Panel.COL, d, mat.dtype, mat.device)  # L[i,k] by local row
     row_panel = Panel(Panel.ROW, d, mat.dtype, mat.device)  # L[j,k] by local col
     diag_ws = torch.zeros((nb, nb), dtype=mat.dtype, device=mat.device)
-    dinv = ops.dinv_workspace(nb, mat.dtype, mat.device) if gpu else None
 
     for k in range(nt):
         kr, kc = d.rank_of_tile((k, k))
@@ -297,25 +303,14 @@ def _cholesky_dist(mat: Matrix, grid: CommGrid) -> None:
         if on_panel_col:
             if d.rank_row == kr:
                 diag = mat.tile((k, k))
-                if gpu:
-                    ops.potrf_tile(diag, dinv)
-                else:
-                    ops.potrf_tile(diag, None)
+                ops.potrf_tile(diag, None)
             else:
                 diag = diag_ws
             if grid.col_group is not None:
                 coll.broadcast(diag, grid.global_rank_of_col_member(kr), grid.col_group)
-            if gpu and d.rank_row != kr:
-                _compute_dinv(diag, dinv)
-            rows = [d.global_tile_of_local((li, 0))[0] for li in range(li0, lr)]
-            if rows:
-                if gpu:
-                    _run_trsm_panel(table, ("k", k), mat.storage, diag, dinv, nb, opc)
-                else:
-                    for i in rows:
-                        _cpu_trsm_tile(diag, mat.tile((i, k)))
             for li in range(li0, lr):
                 i = d.global_tile_of_local((li, 0))[0]
+                _cpu_trsm_tile(diag, mat.tile((i, k)))
                 col_panel.slot(li).copy_(mat.tile((i, k)))
 
         # column panel: one contiguous broadcast along the row direction
@@ -340,22 +335,109 @@ def _cholesky_dist(mat: Matrix, grid: CommGrid) -> None:
                     row_panel.slot(lj), grid.global_rank_of_col_member(jr), grid.col_group
                 )
 
-        # trailing update on local tiles (global i >= j > k)
-        if gpu:
-            descs = table.get(("k", k, "trail"))
-            if descs is not None:
-                ops.gemm_fused(
-                    mat.storage, col_panel.storage, row_panel.storage, descs,
-                    nb, nb, nb, nb, nb, nb, Op.NoTrans, opc, -1.0, 1.0,
+        for lj in range(lj0, lc):
+            j = d.global_tile_of_local((0, lj))[1]
+            for li in range(d.next_local_tile_row(j), lr):
+                ops.gemm_tile(
+                    mat.local_tile((li, lj)), col_panel.slot(li), row_panel.slot(lj),
+                    Op.NoTrans, opc, -1.0, 1.0,
                 )
-        else:
+
+
+def _cholesky_dist_gpu(mat: Matrix, grid: CommGrid) -> None:
+    """Distributed GPU Cholesky with comm/compute-overlapped lookahead.
+
+    Two streams (the reference's priority-based lookahead,
+    ``factorization/cholesky/impl.h:280-282``, recast for fused kernels):
+      sp: panel path + RCCL collectives for step k — depends only on the
+          HEAD update of step k-1 (column k), so it overlaps the bulk
+          trailing GEMM of step k-1 running on su;
+      su: head(k) (column k+1 only) then tail(k) (columns >= k+2).
+    Panels are double-buffered; reuse of a buffer set waits on the tail
+    event of the step that last read it.
+    """
+    d = mat.dist
+    nt = d.nr_tiles[0]
+    nb = d.nb
+    opc = _op_conj(mat.dtype)
+    table = _dist_plan(mat)
+    panels = [
+        (Panel(Panel.COL, d, mat.dtype, mat.device), Panel(Panel.ROW, d, mat.dtype, mat.device)),
+        (Panel(Panel.COL, d, mat.dtype, mat.device), Panel(Panel.ROW, d, mat.dtype, mat.device)),
+    ]
+    diag_ws = torch.zeros((nb, nb), dtype=mat.dtype, device=mat.device)
+    dinv = ops.dinv_workspace(nb, mat.dtype, mat.device)
+
+    rt = get_runtime(mat.device)
+    sp, su = rt.hp_streams[0], rt.np_streams[0]
+    cur = torch.cuda.current_stream(mat.device)
+    sp.wait_stream(cur)
+    su.wait_stream(cur)
+    ev_head = [None] * (nt + 1)
+    ev_tail = [None] * (nt + 1)
+    lr = d.local_nr_tiles[0]
+    lc = d.local_nr_tiles[1]
+
+    for k in range(nt):
+        kr, kc = d.rank_of_tile((k, k))
+        on_panel_col = d.rank_col == kc
+        li0 = d.next_local_tile_row(k + 1)
+        lj0 = d.next_local_tile_col(k + 1)
+        col_panel, row_panel = panels[k % 2]
+
+        with torch.cuda.stream(sp):
+            if k >= 1 and ev_head[k - 1] is not None:
+                sp.wait_event(ev_head[k - 1])      # column k fully updated
+            if k >= 2 and ev_tail[k - 2] is not None:
+                sp.wait_event(ev_tail[k - 2])      # panel buffer set reuse
+            if on_panel_col:
+                if d.rank_row == kr:
+                    diag = mat.tile((k, k))
+                    ops.potrf_tile(diag, dinv)
+                else:
+                    diag = diag_ws
+                if grid.col_group is not None:
+                    coll.broadcast(diag, grid.global_rank_of_col_member(kr), grid.col_group)
+                if d.rank_row != kr:
+                    _compute_dinv(diag, dinv)
+                if li0 < lr:
+                    _run_trsm_panel(table, ("k", k), mat.storage, diag, dinv, nb, opc)
+                for li in range(li0, lr):
+                    i = d.global_tile_of_local((li, 0))[0]
+                    col_panel.slot(li).copy_(mat.tile((i, k)))
+            if grid.row_group is not None and lr - li0 > 0:
+                coll.broadcast(col_panel.range_view(li0, lr),
+                               grid.global_rank_of_row_member(kc), grid.row_group)
             for lj in range(lj0, lc):
                 j = d.global_tile_of_local((0, lj))[1]
-                for li in range(d.next_local_tile_row(j), lr):
-                    ops.gemm_tile(
-                        mat.local_tile((li, lj)), col_panel.slot(li), row_panel.slot(lj),
-                        Op.NoTrans, opc, -1.0, 1.0,
-                    )
+                jr = d.rank_of_tile_row(j)
+                if d.rank_row == jr:
+                    row_panel.slot(lj).copy_(col_panel.slot(d.next_local_tile_row(j)))
+                if grid.col_group is not None:
+                    coll.broadcast(row_panel.slot(lj),
+                                   grid.global_rank_of_col_member(jr), grid.col_group)
+            ev_b = torch.cuda.Event()
+            ev_b.record(sp)
+
+        with torch.cuda.stream(su):
+            su.wait_event(ev_b)
+            head = table.get(("k", k, "head"))
+            if head is not None:
+                ops.gemm_fused(mat.storage, col_panel.storage, row_panel.storage, head,
+                               nb, nb, nb, nb, nb, nb, Op.NoTrans, opc, -1.0, 1.0)
+            e = torch.cuda.Event()
+            e.record(su)
+            ev_head[k] = e
+            tail = table.get(("k", k, "trail"))
+            if tail is not None:
+                ops.gemm_fused(mat.storage, col_panel.storage, row_panel.storage, tail,
+                               nb, nb, nb, nb, nb, nb, Op.NoTrans, opc, -1.0, 1.0)
+            e2 = torch.cuda.Event()
+            e2.record(su)
+            ev_tail[k] = e2
+
+    cur.wait_stream(sp)
+    cur.wait_stream(su)
 
 
 def cholesky_factorization(uplo: UpLo, mat: Matrix, grid: Optional[CommGrid] = None) -> None:
