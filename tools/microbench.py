@@ -13,7 +13,7 @@ from dlaf_amd.ops import tile_ops as ops
 from dlaf_amd.types import Op
 
 
-def bench_gemm(dtype=torch.float64, nb=512, ntiles=128, iters=10, opB=Op.Trans):
+def bench_gemm(dtype=torch.float64, nb=512, ntiles=128, iters=10, opB=Op.Trans, inplace=False):
     dev = "cuda"
     A = torch.randn(ntiles, nb, nb, dtype=dtype, device=dev) if not dtype.is_complex else (
         torch.randn(ntiles, nb, nb, dtype=torch.float64, device=dev)
@@ -26,15 +26,15 @@ def bench_gemm(dtype=torch.float64, nb=512, ntiles=128, iters=10, opB=Op.Trans):
     dt = torch.from_numpy(descs).to(dev)
     # warmup
     for _ in range(3):
-        ops.gemm_fused(C, A, A, dt, nb, nb, nb, nb, nb, nb, Op.NoTrans, opB, -1.0, 1.0)
+        ops.gemm_fused(C, A, A, dt, nb, nb, nb, nb, nb, nb, Op.NoTrans, opB, -1.0, 1.0, inplace=inplace)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
-        ops.gemm_fused(C, A, A, dt, nb, nb, nb, nb, nb, nb, Op.NoTrans, opB, -1.0, 1.0)
+        ops.gemm_fused(C, A, A, dt, nb, nb, nb, nb, nb, nb, Op.NoTrans, opB, -1.0, 1.0, inplace=inplace)
     torch.cuda.synchronize()
     dt_s = (time.perf_counter() - t0) / iters
     flops = 2.0 * ntiles * nb * nb * nb * (4 if dtype.is_complex else 1)
-    print(f"gemm {dtype} nb={nb} ntiles={ntiles} opB={opB}: {dt_s*1e3:.2f} ms  "
+    print(f"gemm {dtype} nb={nb} ntiles={ntiles} opB={opB} bn128={inplace}: {dt_s*1e3:.2f} ms  "
           f"{flops/dt_s/1e12:.2f} TFLOP/s")
 
 
