@@ -441,8 +441,16 @@ def _cholesky_dist_gpu(mat: Matrix, grid: CommGrid) -> None:
 
 
 def cholesky_factorization(uplo: UpLo, mat: Matrix, grid: Optional[CommGrid] = None) -> None:
-    """In-place Cholesky factorization A = L L^H (Lower) of a tiled matrix."""
-    assert uplo == UpLo.Lower, "only Lower implemented (as in the reference miniapps)"
+    """In-place Cholesky: A = L L^H (Lower) or A = U^H U (Upper).
+
+    Upper runs through the U = L^H storage-transpose reduction (reference
+    ``factorization/cholesky/impl.h:317`` implements call_U natively)."""
+    if uplo == UpLo.Upper:
+        from ._uplo import transpose_storage
+        transpose_storage(mat)
+        cholesky_factorization(UpLo.Lower, mat, grid)
+        transpose_storage(mat)
+        return
     d = mat.dist
     assert d.m == d.n and d.mb == d.nb, "square matrix with square tiles required"
     g = grid if grid is not None else mat.grid

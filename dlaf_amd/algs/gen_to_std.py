@@ -57,7 +57,15 @@ def _hegst_diag_tile(a: torch.Tensor, l: torch.Tensor) -> None:
 def generalized_to_standard(uplo: UpLo, mat_a: Matrix, mat_l: Matrix,
                             grid: Optional[CommGrid] = None) -> None:
     """In-place HEGST (itype=1): A <- inv(L) A inv(L)^H, Lower storage."""
-    assert uplo == UpLo.Lower, "only Lower implemented (as the reference's C API)"
+    if uplo == UpLo.Upper:
+        # A <- U^-H A U^-1 == (with L = U^H) L^-1 A L^-H on transposed storage
+        from ._uplo import transpose_storage
+        transpose_storage(mat_a)
+        transpose_storage(mat_l)
+        generalized_to_standard(UpLo.Lower, mat_a, mat_l, grid)
+        transpose_storage(mat_a)
+        transpose_storage(mat_l)
+        return
     da, dl = mat_a.dist, mat_l.dist
     assert da.m == da.n and da.mb == da.nb
     assert (dl.m, dl.n, dl.mb, dl.nb) == (da.m, da.n, da.mb, da.nb)

@@ -48,7 +48,12 @@ def triangular_inverse(uplo: UpLo, diag: Diag, mat: Matrix,
     Reference: ``inverse/triangular/impl.h`` (Lower; Upper by symmetry is not
     provided by the reference miniapps either).
     """
-    assert uplo == UpLo.Lower, "only Lower implemented (as the reference's miniapps)"
+    if uplo == UpLo.Upper:
+        from ._uplo import transpose_storage
+        transpose_storage(mat)
+        triangular_inverse(UpLo.Lower, diag, mat, grid)
+        transpose_storage(mat)
+        return
     d = mat.dist
     assert d.m == d.n and d.mb == d.nb
     g = _trivial_grid(grid if grid is not None else mat.grid)
@@ -122,7 +127,12 @@ def inverse_from_cholesky_factor(uplo: UpLo, mat: Matrix,
     Reference: ``inverse/cholesky/impl.h:180-540`` (TRTRI then LAUUM-style
     assembly).
     """
-    assert uplo == UpLo.Lower, "only Lower implemented (as the reference miniapps)"
+    if uplo == UpLo.Upper:
+        from ._uplo import transpose_storage
+        transpose_storage(mat)
+        inverse_from_cholesky_factor(UpLo.Lower, mat, grid)
+        transpose_storage(mat)
+        return
     d = mat.dist
     assert d.m == d.n and d.mb == d.nb
     g = _trivial_grid(grid if grid is not None else mat.grid)

@@ -71,3 +71,15 @@ def test_cholesky_dist_cpu_dtypes(dtype_str):
     tol = 1e-3 if dtype_str == "float32" else 1e-10
     for e in errs:
         assert e < tol, f"err={e}"
+
+
+def test_cholesky_upper_local():
+    n, nb = 24, 8
+    mat = Matrix.create(n, n, nb, nb, dtype=torch.complex128)
+    mutil.set_random_hermitian_positive_definite(mat, seed=7)
+    a_ref = mat.to_global()
+    a_ref = torch.tril(a_ref) + torch.tril(a_ref, -1).mH
+    cholesky_factorization(UpLo.Upper, mat)
+    U = torch.triu(mat.to_global())
+    err = (U.mH @ U - a_ref).abs().max().item()
+    assert err < 1e-11 * n, f"err={err}"

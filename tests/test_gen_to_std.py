@@ -54,3 +54,26 @@ def test_hegst_dist_cpu_complex():
     errs = run_distributed(_dist_worker, 2, args=(2, 1, 18, 5, "complex128"))
     for e in errs:
         assert e < 1e-9, f"err={e}"
+
+
+def test_hegst_upper_local():
+    n, nb = 16, 4
+    dtype = torch.complex128
+    A = Matrix.create(n, n, nb, nb, dtype=dtype)
+    L = Matrix.create(n, n, nb, nb, dtype=dtype)
+    mutil.set_random_hermitian(A, seed=31)
+    mutil.set_random_hermitian_positive_definite(L, seed=32)
+    b = L.to_global()
+    b = torch.tril(b) + torch.tril(b, -1).mH
+    U = torch.linalg.cholesky(b).mH
+    L.set_from_global(torch.triu(U))
+    a0 = A.to_global()
+    a0h = torch.triu(a0) + torch.triu(a0, 1).mH
+    A.set_from_global(torch.triu(a0))
+    generalized_to_standard(UpLo.Upper, A, L)
+    uinv = torch.linalg.inv(U)
+    want = uinv.mH @ a0h @ uinv
+    got = A.to_global()
+    got = torch.triu(got) + torch.triu(got, 1).mH
+    err = (got - want).abs().max().item()
+    assert err < 1e-9 * n, f"err={err}"

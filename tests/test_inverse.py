@@ -88,3 +88,27 @@ def test_inverse_dist_cpu_complex(which):
     errs = run_distributed(_dist_worker, 2, args=(2, 1, which, 18, 5, "complex128"))
     for e in errs:
         assert e < 1e-9, f"err={e}"
+
+
+def test_trtri_upper_local():
+    n, nb = 18, 6
+    mat = Matrix.create(n, n, nb, nb, dtype=torch.float64)
+    mutil.set_random_hermitian_positive_definite(mat, seed=9)
+    a = mat.to_global()
+    triangular_inverse(UpLo.Upper, Diag.NonUnit, mat)
+    want = torch.linalg.inv(torch.triu(a))
+    err = (torch.triu(mat.to_global()) - torch.triu(want)).abs().max().item()
+    assert err < 1e-10 * n, f"err={err}"
+
+
+def test_potri_upper_local():
+    n, nb = 16, 4
+    mat = Matrix.create(n, n, nb, nb, dtype=torch.float64)
+    mutil.set_random_hermitian_positive_definite(mat, seed=10)
+    a = mat.to_global()
+    U = torch.linalg.cholesky(a).mH
+    mat.set_from_global(torch.triu(U))
+    inverse_from_cholesky_factor(UpLo.Upper, mat)
+    want = torch.linalg.inv(a)
+    err = (torch.triu(mat.to_global()) - torch.triu(want)).abs().max().item()
+    assert err < 1e-9 * n, f"err={err}"
