@@ -152,7 +152,10 @@ __global__ void panel_qr_kernel(T* __restrict__ P, long m, int nb, long ldp,
       for (long q = j + gid0; q < nb; q += gstride) wrow[q] = P[(long)j * ldp + q];
     }
     grid.sync();
-    // ---- phase B: trailing update (reads only staged row/diag + unscaled x) --
+    // ---- phase B: trailing update (reads only staged row/diag + unscaled x).
+    // Tail scaling, beta and tau writes are DEFERRED to the epilogue: column
+    // j is never read again by later columns, so leaving its tail unscaled
+    // halves the grid syncs (2 per column).
     {
       T tau, scale;
       R beta;
@@ -172,22 +175,22 @@ __global__ void panel_qr_kernel(T* __restrict__ P, long m, int nb, long ldp,
       }
     }
     grid.sync();
-    // ---- phase C: scale the stored tail, write beta and tau ----
-    {
-      T tau, scale;
-      R beta;
-      bool degen;
-      derive<T, R>(norms[j], wrow[j], tau, scale, beta, degen);
-      if (!degen) {
-        for (long i = j + 1 + gid0; i < m; i += gstride)
-          P[i * ldp + j] = P[i * ldp + j] * scale;
-      }
-      if (blockIdx.x == 0 && tid == 0) {
-        taus[j] = tau;
-        if (!degen) P[(long)j * ldp + j] = ScalarTraits<T>::from_real(beta);
-      }
+  }
+  // ---- epilogue: per-column tail scaling + beta/tau writes ----
+  for (int j = 0; j < ncols; ++j) {
+    T* wrow = wraw + (long)(nb + j) * nb;
+    T tau, scale;
+    R beta;
+    bool degen;
+    derive<T, R>(norms[j], wrow[j], tau, scale, beta, degen);
+    if (!degen) {
+      for (long i = j + 1 + gid0; i < m; i += gstride)
+        P[i * ldp + j] = P[i * ldp + j] * scale;
     }
-    grid.sync();
+    if (blockIdx.x == 0 && tid == 0) {
+      taus[j] = tau;
+      if (!degen) P[(long)j * ldp + j] = ScalarTraits<T>::from_real(beta);
+    }
   }
 }
 
@@ -207,7 +210,8 @@ extern "C" {
         &per_cu, reinterpret_cast<const void*>(&panel_qr_kernel<T>), threads, \
         0);                                                                   \
     int blocks = prop.multiProcessorCount * (per_cu > 0 ? per_cu : 1);        \
-    if (blocks > 1024) blocks = 1024;                                         \
+    /* cooperative grid.sync cost grows with the WG count: cap at 256 */      \
+    if (blocks > 256) blocks = 256;                                           \
     void* args[] = {(void*)&P,    (void*)&m,     (void*)&nb,  (void*)&ldp,    \
                     (void*)&taus, (void*)&norms, (void*)&wraw};               \
     hipError_t err = hipLaunchCooperativeKernel(                              \
