@@ -271,6 +271,8 @@ void band_chase(torch::Tensor band, int64_t b, torch::Tensor vstore,
                 torch::Tensor offsets, int64_t nthreads);
 
 extern "C" {
+void secular_roots_f64(const double*, const double*, int, double, long long*,
+                       double*, hipStream_t);
 int panel_qr_f64(double*, long, int, long, double*, double*, double*, hipStream_t);
 int panel_qr_f32(float*, long, int, long, float*, float*, float*, hipStream_t);
 int panel_qr_c128(double*, long, int, long, double*, double*, double*, hipStream_t);
@@ -317,6 +319,15 @@ void panel_qr(torch::Tensor P, torch::Tensor taus, torch::Tensor norms,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("panel_qr", &panel_qr, "cooperative whole-panel QR (one launch)");
+  m.def("secular_roots", [](torch::Tensor d, torch::Tensor z2, double rho,
+                            torch::Tensor sidx, torch::Tensor mu) {
+    TORCH_CHECK(d.is_cuda() && d.scalar_type() == torch::kFloat64);
+    auto stream = at::cuda::getCurrentHIPStream().stream();
+    secular_roots_f64(d.data_ptr<double>(), z2.data_ptr<double>(),
+                      (int)d.size(0), rho, (long long*)sidx.data_ptr<int64_t>(),
+                      mu.data_ptr<double>(), stream);
+    HIP_CHECK(hipGetLastError());
+  }, "D&C secular-equation roots, one thread per root");
   m.def("band_chase", &band_chase,
         "CPU bulge chasing band->tridiag with reflector recording",
         py::arg("band"), py::arg("b"), py::arg("vstore"), py::arg("offsets"),

@@ -51,6 +51,13 @@ def _secular_roots(d: torch.Tensor, z: torch.Tensor, rho: float,
     z2 = z * z
     if k == 1:
         return torch.zeros(1, dtype=torch.int64, device=dev), rho * z2
+    if dev.type == "cuda" and d.dtype == torch.float64:
+        # single-kernel path: one thread per root (csrc/secular.hip)
+        from ..ops._ext import get_ext
+        sidx = torch.empty(k, dtype=torch.int64, device=dev)
+        mu = torch.empty(k, dtype=torch.float64, device=dev)
+        get_ext().secular_roots(d.contiguous(), z2.contiguous(), float(rho), sidx, mu)
+        return sidx, mu
     # interval per root j: (d_j, d_{j+1}), last: (d_{k-1}, d_{k-1} + rho)
     d_lo = d
     d_hi = torch.cat([d[1:], (d[-1] + rho).reshape(1)])

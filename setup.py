@@ -22,6 +22,7 @@ setup(
                 "csrc/gemm_tiles.hip",
                 "csrc/factor.hip",
                 "csrc/panel_qr.hip",
+                "csrc/secular.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3"],

@@ -218,3 +218,27 @@ def test_panel_qr_kernel_gpu():
             terr = (tg.cpu() - tc).abs().max().item()
             assert perr < tol * m, f"{dtype} {m}x{nb} P err={perr}"
             assert terr < tol * m, f"{dtype} {m}x{nb} tau err={terr}"
+
+
+def test_secular_kernel_gpu():
+    """HIP one-thread-per-root secular solver vs the CPU torch reference."""
+    import dlaf_amd.algs.tridiag_dc as dc
+    g = torch.Generator().manual_seed(23)
+    for k in [5, 100, 700]:
+        d = torch.sort(torch.randn(k, generator=g, dtype=torch.float64))[0]
+        # well-separated + some clusters
+        z = torch.randn(k, generator=g, dtype=torch.float64)
+        z = z / z.norm()
+        rho = 1.7
+        sc, mc = dc._secular_roots(d, z, rho)             # CPU torch path
+        sg, mg = dc._secular_roots(d.cuda(), z.cuda(), rho)  # HIP kernel
+        torch.cuda.synchronize()
+        lam_c = d[sc] + mc
+        lam_g = (d.cuda()[sg] + mg).cpu()
+        err = (lam_c - lam_g).abs().max().item()
+        assert err < 1e-12 * max(1.0, d.abs().max().item()), f"k={k} err={err}"
+        # residual of the secular equation at the GPU roots
+        dd = d.unsqueeze(1) - d[sg.cpu()].unsqueeze(0)
+        f = 1.0 + rho * ((z * z).unsqueeze(1) / (dd - mg.cpu().unsqueeze(0))).sum(0)
+        fp = rho * ((z * z).unsqueeze(1) / (dd - mg.cpu().unsqueeze(0)) ** 2).sum(0)
+        assert (f.abs() / fp).max().item() < 1e-12, f"k={k}"
