@@ -38,6 +38,18 @@ from .algs.norm import max_norm  # noqa: F401
 from .algs.eigensolver import (  # noqa: F401
     hermitian_eigensolver,
     hermitian_generalized_eigensolver,
+    get_band_size,
 )
+from .algs.red2band import reduction_to_band, bt_reduction_to_band  # noqa: F401
+from .algs.band2tridiag import (  # noqa: F401
+    band_to_tridiagonal,
+    bt_band_to_tridiagonal,
+)
+from .algs.tridiag_dc import tridiagonal_eigensolver  # noqa: F401
+from .algs.permutations import permute_columns, permute_rows  # noqa: F401
+from .algs.redistribute import redistribute  # noqa: F401
+from .matrix.mirror import MatrixMirror, MatrixRef, save_matrix, load_matrix  # noqa: F401
+from .config import initialize, finalize, ScopedInitializer, get_tune_parameters  # noqa: F401
+from .utils import Timer, trace_range  # noqa: F401
 
 __version__ = "0.1.0"
