@@ -269,6 +269,8 @@ void factor_invert_block(torch::Tensor A, int64_t n, int64_t ld,
 
 void band_chase(torch::Tensor band, int64_t b, torch::Tensor vstore,
                 torch::Tensor offsets, int64_t nthreads);
+int64_t dc_deflate_scan(torch::Tensor d, torch::Tensor z, double rho,
+                        double tol, torch::Tensor deflated, torch::Tensor rots);
 
 extern "C" {
 void secular_roots_f64(const double*, const double*, int, double, long long*,
@@ -328,6 +330,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                       mu.data_ptr<double>(), stream);
     HIP_CHECK(hipGetLastError());
   }, "D&C secular-equation roots, one thread per root");
+  m.def("dc_deflate_scan", &dc_deflate_scan,
+        "sequential deflation scan for the D&C merge");
   m.def("band_chase", &band_chase,
         "CPU bulge chasing band->tridiag with reflector recording",
         py::arg("band"), py::arg("b"), py::arg("vstore"), py::arg("offsets"),

@@ -187,32 +187,22 @@ def _merge(w1, Q1, w2, Q2, rho, device):
     d_s, perm = torch.sort(d)
     z_s = z[perm]
 
-    # ---- deflation (host bookkeeping, O(k)) ----
-    dn = d_s.cpu().numpy().copy()
-    zn = z_s.cpu().numpy().copy()
+    # ---- deflation (host bookkeeping, O(k), sequential scan in C++) ----
+    dn = d_s.cpu().numpy().astype(np.float64).copy()
+    zn = z_s.cpu().numpy().astype(np.float64).copy()
     k = n
     dmax = max(np.abs(dn).max(), rho_eff) if k else 1.0
     tol = 8.0 * _EPS * max(dmax, 1e-300)
-    deflated = np.zeros(k, dtype=bool)
-    rots = []  # (i, j, c, s)
-    deflated |= np.abs(rho_eff * zn) <= tol
-    last = -1
-    for i in range(k):
-        if deflated[i]:
-            continue
-        if last >= 0 and (dn[i] - dn[last]) <= tol:
-            zi, zj = zn[last], zn[i]
-            r = math.hypot(zi, zj)
-            c, s = zj / r, -zi / r
-            # zero z[last], keep z[i] = r
-            zn[i] = r
-            zn[last] = 0.0
-            di, dj = dn[last], dn[i]
-            dn[last] = di * c * c + dj * s * s
-            dn[i] = di * s * s + dj * c * c
-            rots.append((last, i, c, s))
-            deflated[last] = True
-        last = i
+    from ..ops._ext import get_ext
+    dt_ = torch.from_numpy(dn)
+    zt_ = torch.from_numpy(zn)
+    deflated_t = torch.zeros(k, dtype=torch.bool)
+    rots_t = torch.zeros((max(k, 1), 4), dtype=torch.float64)
+    nrot = get_ext().dc_deflate_scan(dt_, zt_, float(rho_eff), float(tol),
+                                     deflated_t, rots_t)
+    deflated = deflated_t.numpy()
+    rots = [(int(r[0]), int(r[1]), float(r[2]), float(r[3]))
+            for r in rots_t[:nrot]]
 
     nd_idx = np.nonzero(~deflated)[0]
     df_idx = np.nonzero(deflated)[0]
@@ -234,9 +224,13 @@ def _merge(w1, Q1, w2, Q2, rho, device):
     Q_out = torch.empty_like(Qb)
 
     if k1 > 0:
-        # small merges stay on CPU tensors (launch overhead dominates);
-        # anything sizeable runs on the device
-        sec_dev = device if k1 >= 192 else torch.device("cpu")
+        # the HIP one-thread-per-root kernel handles every size in one
+        # launch; the torch formulation stays for CPU runs (small merges on
+        # host tensors, large on device)
+        if device.type == "cuda" and d_s.dtype == torch.float64:
+            sec_dev = device
+        else:
+            sec_dev = device if k1 >= 192 else torch.device("cpu")
         dk = torch.from_numpy(dn[nd_idx]).to(sec_dev)
         zk = torch.from_numpy(zn[nd_idx]).to(sec_dev)
         sidx, mu = _secular_roots(dk, zk, rho_eff)
