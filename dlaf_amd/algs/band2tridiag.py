@@ -99,7 +99,7 @@ def band_to_tridiagonal(uplo: UpLo, band: int, mat: Matrix) -> TridiagResult:
 
 
 def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
-                           group_size: int = 64) -> None:
+                           group_size: Optional[int] = None) -> None:
     """Back-transform E <- Q (D E) where A_band = Q (D T_real D^H) Q^H.
 
     E: [n, nE] device tensor of tridiagonal eigenvectors, updated in place.
@@ -114,6 +114,9 @@ def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
     exactly those the ascending order keeps correctly ordered; the disjoint
     pairs commute).
     """
+    if group_size is None:
+        from ..config import get_tune_parameters
+        group_size = get_tune_parameters().bt_band_to_tridiag_hh_apply_group_size
     n, nE = E.shape
     b = tri.band
     dev = E.device
