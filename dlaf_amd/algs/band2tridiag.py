@@ -38,18 +38,14 @@ class TridiagResult:
 
 
 def _slot_counts(n: int, b: int) -> torch.Tensor:
+    """Reflector slots per sweep: 1 + number of chase steps with m > 1
+    (closed form of the do_step loop; sweep s chases while
+    1 + s + t*b <= n - b - 2)."""
     counts = torch.zeros(n, dtype=torch.int64)
-    for s in range(max(n - 2, 0)):
-        k = 1
-        t = 0
-        while True:
-            j = 1 + s + t * b
-            m = min(b, n - b - j)
-            if m <= 1:
-                break
-            k += 1
-            t += 1
-        counts[s] = k
+    if n > 2:
+        s = torch.arange(n - 2)
+        tmax = torch.div(n - b - 3 - s, b, rounding_mode="floor")
+        counts[: n - 2] = 1 + torch.clamp(tmax + 1, min=0)
     return counts
 
 
@@ -63,12 +59,15 @@ def band_to_tridiagonal(uplo: UpLo, band: int, mat: Matrix) -> TridiagResult:
     g = mat.grid
     assert g is None or not g.distributed, \
         "distributed band_to_tridiagonal lands with the distributed eigensolver"
-    A = mat.to_global().cpu()
+    A = mat.to_global()
     b = band
     ld = 2 * b
-    store = torch.zeros((n, ld), dtype=A.dtype)
+    # extract the compact band on the device, move only n x 2b to the host
+    store_dev = torch.zeros((n, ld), dtype=A.dtype, device=A.device)
     for dd in range(min(b, n - 1) + 1):
-        store[: n - dd, dd] = torch.diagonal(A, -dd)
+        store_dev[: n - dd, dd] = torch.diagonal(A, -dd)
+    store = store_dev.cpu()
+    del A, store_dev
 
     counts = _slot_counts(n, b)
     offsets = torch.zeros(n, dtype=torch.int64)
