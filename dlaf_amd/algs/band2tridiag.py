@@ -66,6 +66,7 @@ def band_to_tridiagonal(uplo: UpLo, band: int, mat: Matrix) -> TridiagResult:
     store_dev = torch.zeros((n, ld), dtype=A.dtype, device=A.device)
     for dd in range(min(b, n - 1) + 1):
         store_dev[: n - dd, dd] = torch.diagonal(A, -dd)
+    is_cplx = A.is_complex()
     store = store_dev.cpu()
     del A, store_dev
 
@@ -74,14 +75,14 @@ def band_to_tridiagonal(uplo: UpLo, band: int, mat: Matrix) -> TridiagResult:
     if n > 1:
         offsets[1:] = torch.cumsum(counts, 0)[:-1]
     total = int(counts.sum().item())
-    vstore = torch.zeros((max(total, 1), b + 1), dtype=A.dtype)
+    vstore = torch.zeros((max(total, 1), b + 1), dtype=store.dtype)
     if n > 2:
         get_ext().band_chase(store, b, vstore, offsets)
 
     dvec = store[:, 0]
     evec = store[: n - 1, 1] if n > 1 else store[:0, 1]
     phases = None
-    if A.is_complex():
+    if is_cplx:
         ph = torch.ones(n, dtype=A.dtype)
         e_abs = evec.abs()
         for j in range(n - 1):
