@@ -83,7 +83,7 @@ def band_to_tridiagonal(uplo: UpLo, band: int, mat: Matrix) -> TridiagResult:
     evec = store[: n - 1, 1] if n > 1 else store[:0, 1]
     phases = None
     if is_cplx:
-        ph = torch.ones(n, dtype=A.dtype)
+        ph = torch.ones(n, dtype=store.dtype)
         e_abs = evec.abs()
         for j in range(n - 1):
             aj = e_abs[j]
