@@ -253,10 +253,9 @@ void chase_impl(T* a, int64_t ld, int64_t size, int64_t b, T* vstore,
 void band_chase(torch::Tensor band, int64_t b, torch::Tensor vstore,
                 torch::Tensor offsets, int64_t nthreads) {
   if (nthreads <= 0) {
-    // wavefront depth supports ~(n/b)/3 concurrent sweeps; GPU-node hosts
-    // have aplenty cores (measured 256), the dependency stagger is the
-    // effective limit
-    nthreads = std::min<int64_t>(64, std::thread::hardware_concurrency());
+    // measured sweet spot: 16 (64 threads ran 3x SLOWER at n=20k b=128 —
+    // the shared band working set thrashes caches across too many cores)
+    nthreads = std::min<int64_t>(16, std::thread::hardware_concurrency());
     if (nthreads < 1) nthreads = 1;
   }
   TORCH_CHECK(!band.is_cuda(), "band_chase is a CPU stage (as the reference)");
