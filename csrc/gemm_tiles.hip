@@ -70,7 +70,17 @@ __launch_bounds__(256) __global__ void gemm_tiles_k(
   __shared__ T As[DBUF ? 2 : 1][BM][BK + 1];
   __shared__ T Bs[DBUF ? 2 : 1][BK][BN + 2];
 
-  const int wg = blockIdx.x;
+  int wg = blockIdx.x;
+  if constexpr (DBUF) {
+    // (variant 2 doubles as the XCD-swizzle experiment) remap so each XCD's
+    // round-robin share becomes a CONTIGUOUS block-index range: blocks of one
+    // desc then co-reside in one XCD's L2 and share A/B tile lines.
+    const int nx = 8;
+    const int g = gridDim.x;
+    const int per = g / nx, rem8 = g % nx;
+    const int xcd = wg % nx, pos = wg / nx;
+    wg = xcd * per + (xcd < rem8 ? xcd : rem8) + pos;
+  }
   const int per_desc = mblocks * nblocks;
   const GemmDesc d = descs[wg / per_desc];
   const int rem = wg % per_desc;
