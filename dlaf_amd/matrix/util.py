@@ -130,3 +130,17 @@ def lower_triangle_dominant(mat: Matrix, seed: int = 0) -> None:
         return v + diag.to(v.dtype)
 
     set_elementwise(mat, fn)
+
+
+def set_random_hermitian_banded(mat: Matrix, band: int, seed: int = 0) -> None:
+    """Random Hermitian with zero entries outside |i-j| <= band (reference
+    ``util_matrix.h`` set_random_hermitian_banded, band2tridiag test input)."""
+    assert mat.dist.m == mat.dist.n and mat.dist.mb == mat.dist.nb
+    base = _hermitian_fn(mat, seed, 0.0)
+
+    def fn(I, J):
+        v = base(I, J)
+        mask = (I - J).abs() <= band
+        return torch.where(mask, v, torch.zeros_like(v))
+
+    set_elementwise(mat, fn)
