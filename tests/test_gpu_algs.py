@@ -242,3 +242,19 @@ def test_secular_kernel_gpu():
         f = 1.0 + rho * ((z * z).unsqueeze(1) / (dd - mg.cpu().unsqueeze(0))).sum(0)
         fp = rho * ((z * z).unsqueeze(1) / (dd - mg.cpu().unsqueeze(0)) ** 2).sum(0)
         assert (f.abs() / fp).max().item() < 1e-12, f"k={k}"
+
+
+def test_eigensolver_gpu_float32():
+    n, nb = 768, 256
+    mat = Matrix.create(n, n, nb, nb, dtype=torch.float32, device="cuda")
+    mutil.set_random_hermitian(mat, seed=35)
+    a0 = _herm(mat.to_global().cpu()).to(torch.float64)
+    w, evecs = hermitian_eigensolver(UpLo.Lower, mat)
+    torch.cuda.synchronize()
+    E = evecs.to_global().cpu().to(torch.float64)
+    w = w.cpu().to(torch.float64)
+    scale = max(1.0, w.abs().max().item())
+    res = (a0 @ E - E @ torch.diag(w)).abs().max().item()
+    assert res < 1e-3 * n * scale, f"res={res}"
+    orth = (E.mT @ E - torch.eye(n, dtype=torch.float64)).abs().max().item()
+    assert orth < 1e-3 * n, f"orth={orth}"
