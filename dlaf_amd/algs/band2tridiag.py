@@ -159,7 +159,7 @@ def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
         M = torch.where(mask, torch.zeros_like(M), M)
         M = M + torch.diag_embed(torch.where(zc, torch.ones_like(safe),
                                              torch.zeros_like(safe)))
-        T = torch.linalg.solve_triangular(M, eyeG.expand(nwin, G, G), upper=True)
+        T = torch.linalg.solve_triangular(M, eyeG.expand(nwin, G, G).contiguous(), upper=True)
         T = torch.where(mask, torch.zeros_like(T), T)
         # apply windows in ascending k (ordering constraint across overlaps)
         for k in range(nwin):

@@ -26,7 +26,7 @@ class TuneParameters:
     """Runtime-tunable knobs (reference ``tune.h:114-168``)."""
     eigensolver_min_band: int = 100
     band_to_tridiag_1d_block_size_base: int = 8192
-    bt_band_to_tridiag_hh_apply_group_size: int = 64
+    bt_band_to_tridiag_hh_apply_group_size: int = 128  # measured best at n=20k
     tridiag_rank1_num_threads: int = 0          # 0 = auto
     red2band_panel_num_threads: int = 0
     tfactor_num_streams: int = 4
