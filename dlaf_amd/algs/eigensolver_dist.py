@@ -104,11 +104,12 @@ def _red2band_dist(A: torch.Tensor, band: int, cm: _ChunkMap, group):
         for a, b in cm.owned_in(r0):
             X[a - r0:b - r0] = Y[a - r0:b - r0] - 0.5 * V[a - r0:b - r0] @ S
         _assemble(X, group)
-        # trailing update on owned rows (full width of A22)
+        # trailing update on owned rows (full width of A22), in place
         XH = X.mH
         VH = V.mH
         for a, b in cm.owned_in(r0):
-            A[a:b, r0:] -= V[a - r0:b - r0] @ XH + X[a - r0:b - r0] @ VH
+            A[a:b, r0:].addmm_(V[a - r0:b - r0], XH, beta=1, alpha=-1)
+            A[a:b, r0:].addmm_(X[a - r0:b - r0], VH, beta=1, alpha=-1)
     return {"taus": taus_all, "panels": panels, "band": band}
 
 
@@ -146,7 +147,7 @@ def _bt_red2band_dist(E: torch.Tensor, A: torch.Tensor, refl, cm: _ChunkMap,
         V = torch.tril(P, -1) + torch.eye(m_p, nrefl, dtype=dt, device=dev)
         T = t_factor(V, taus)
         W = T @ (V.mH @ E[r0:, :])
-        E[r0:, :] -= V @ W
+        E[r0:, :].addmm_(V, W, beta=1, alpha=-1)
 
 
 def hermitian_eigensolver_dist(uplo: UpLo, mat: Matrix, grid: CommGrid,

@@ -164,7 +164,9 @@ def reduction_to_band_dense(A: torch.Tensor, band: int):
         Y = A22 @ (V @ T)                      # hemm (A22 kept full Hermitian)
         S = T.mH @ (V.mH @ Y)
         X = Y - 0.5 * V @ S
-        A22 -= V @ X.mH + X @ V.mH
+        # in-place rank-2b update: no m_t x m_t temporaries
+        A22.addmm_(V, X.mH, beta=1, alpha=-1)
+        A22.addmm_(X, V.mH, beta=1, alpha=-1)
     return taus_all, panels
 
 
@@ -203,4 +205,4 @@ def bt_reduction_to_band(E: torch.Tensor, mat_v: Matrix, refl) -> None:
         V = torch.tril(P, -1) + torch.eye(m_p, nrefl, dtype=A.dtype, device=A.device)
         T = t_factor(V, taus)
         W = T @ (V.mH @ E[r0:, :])
-        E[r0:, :] -= V @ W
+        E[r0:, :].addmm_(V, W, beta=1, alpha=-1)
