@@ -265,12 +265,15 @@ def _merge(w1, Q1, w2, Q2, rho, device):
 
 
 def tridiagonal_eigensolver(d: torch.Tensor, e: torch.Tensor,
-                            device=None, leaf: int = 64):
+                            device=None, leaf: int = None):
     """Eigendecomposition of a real symmetric tridiagonal matrix.
 
     Returns (evals [n] fp, evecs [n, n]) on ``device``. Reference:
     ``eigensolver/tridiag_solver/impl.h:198-278`` (local).
     """
+    if leaf is None:
+        import os
+        leaf = int(os.environ.get("DLAF_DC_LEAF", "128"))
     if device is None:
         device = d.device
     dn = d.detach().cpu().numpy().astype(np.float64).copy()
