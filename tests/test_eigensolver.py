@@ -123,3 +123,26 @@ def test_gen_eigensolver_dist_cpu():
     for res, werr in out:
         assert res < 1e-8 * n, f"res={res}"
         assert werr < 1e-9 * n, f"werr={werr}"
+
+
+def _dist_partial_worker(rank, ws, gr, gc):
+    grid = CommGrid(gr, gc)
+    n, nb = 32, 8
+    A = Matrix.create(n, n, nb, nb, dtype=torch.float64, grid=grid)
+    mutil.set_random_hermitian(A, seed=91)
+    a0 = _herm(A.to_global())
+    ib, ie = 4, 20
+    w, evecs = hermitian_eigensolver(UpLo.Lower, A, grid, band=8,
+                                     eigenvalues_index_begin=ib,
+                                     eigenvalues_index_end=ie)
+    E = evecs.to_global()[:, : ie - ib]
+    wref = np.linalg.eigvalsh(a0.numpy())
+    werr = np.abs(np.sort(w.numpy()) - wref[ib:ie]).max()
+    res = (a0 @ E - E @ torch.diag(w)).abs().max().item()
+    return res, werr
+
+
+def test_eigensolver_dist_partial_spectrum():
+    for res, werr in run_distributed(_dist_partial_worker, 2, args=(1, 2)):
+        assert res < 1e-9 * 32, f"res={res}"
+        assert werr < 1e-10 * 32, f"werr={werr}"
