@@ -130,6 +130,8 @@ __global__ void secular_kernel(const T* __restrict__ d,
       side = 1;
     }
     if (fx == T(0)) break;
+    // converged: bracket width at relative machine precision
+    if (hi - lo <= T(1e-16) * (fabs(lo) + fabs(hi) + T(1e-300))) break;
   }
   mu = (fabs(flo) < fabs(fhi)) ? lo : hi;
   sidx[j] = p;

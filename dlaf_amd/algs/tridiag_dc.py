@@ -273,7 +273,7 @@ def tridiagonal_eigensolver(d: torch.Tensor, e: torch.Tensor,
     """
     if leaf is None:
         import os
-        leaf = int(os.environ.get("DLAF_DC_LEAF", "128"))
+        leaf = int(os.environ.get("DLAF_DC_LEAF", "64"))
     if device is None:
         device = d.device
     dn = d.detach().cpu().numpy().astype(np.float64).copy()
