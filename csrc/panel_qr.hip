@@ -210,8 +210,13 @@ extern "C" {
         &per_cu, reinterpret_cast<const void*>(&panel_qr_kernel<T>), threads, \
         0);                                                                   \
     int blocks = prop.multiProcessorCount * (per_cu > 0 ? per_cu : 1);        \
-    /* cooperative grid.sync cost grows with the WG count: cap at 256 */      \
-    if (blocks > 256) blocks = 256;                                           \
+    /* cooperative grid.sync cost grows with the WG count (measured ~40us at \
+       256 WGs); the phase-B work is small, so favour fewer, fuller WGs */    \
+    static const int cap = [] {                                               \
+      const char* v = getenv("DLAF_PANEL_QR_BLOCKS");                         \
+      return v ? atoi(v) : 128;                                               \
+    }();                                                                      \
+    if (blocks > cap) blocks = cap;                                           \
     void* args[] = {(void*)&P,    (void*)&m,     (void*)&nb,  (void*)&ldp,    \
                     (void*)&taus, (void*)&norms, (void*)&wraw};               \
     hipError_t err = hipLaunchCooperativeKernel(                              \
