@@ -56,7 +56,8 @@ struct Mfma<float> {
 //   BN = 128 ("in-place-safe"): wave tile 64x64 -> a single column block spans
 //        the whole N of a panel-apply, making X = X * dinv^H safe in place
 //        (every workgroup reads all of its A rows before writing them).
-template <typename T, int OPA, int OPB, int BN, int BK = 16, bool DBUF = false>
+template <typename T, int OPA, int OPB, int BN, int BK = 16, bool DBUF = false,
+          bool GUARD = true>
 __launch_bounds__(256) __global__ void gemm_tiles_k(
     const GemmDesc* __restrict__ descs, const T* __restrict__ A,
     const T* __restrict__ B, T* __restrict__ C, int M, int N, int K, int lda,
@@ -121,10 +122,14 @@ __launch_bounds__(256) __global__ void gemm_tiles_k(
         i = e % BM;
       }
       const int gi = i0 + i, gk = k0 + k;
-      T v = T(0);
-      if (gi < M && gk < K)
-        v = (OPA == OP_N) ? Ab[(int64_t)gi * lda + gk] : Ab[(int64_t)gk * lda + gi];
-      va[j] = v;
+      if constexpr (GUARD) {
+        T v = T(0);
+        if (gi < M && gk < K)
+          v = (OPA == OP_N) ? Ab[(int64_t)gi * lda + gk] : Ab[(int64_t)gk * lda + gi];
+        va[j] = v;
+      } else {
+        va[j] = (OPA == OP_N) ? Ab[(int64_t)gi * lda + gk] : Ab[(int64_t)gk * lda + gi];
+      }
     }
 #pragma unroll
     for (int j = 0; j < LB; ++j) {
@@ -138,10 +143,14 @@ __launch_bounds__(256) __global__ void gemm_tiles_k(
         k = e % BK;
       }
       const int gk = k0 + k, gc = j0 + c;
-      T v = T(0);
-      if (gk < K && gc < N)
-        v = (OPB == OP_N) ? Bb[(int64_t)gk * ldb + gc] : Bb[(int64_t)gc * ldb + gk];
-      vb[j] = v;
+      if constexpr (GUARD) {
+        T v = T(0);
+        if (gk < K && gc < N)
+          v = (OPB == OP_N) ? Bb[(int64_t)gk * ldb + gc] : Bb[(int64_t)gc * ldb + gk];
+        vb[j] = v;
+      } else {
+        vb[j] = (OPB == OP_N) ? Bb[(int64_t)gk * ldb + gc] : Bb[(int64_t)gc * ldb + gk];
+      }
     }
   };
 
@@ -225,7 +234,7 @@ __launch_bounds__(256) __global__ void gemm_tiles_k(
       for (int r = 0; r < 4; ++r) {
         const int row = i0 + wrow + mi * 16 + Mfma<T>::acc_row(lk, r);
         const int col = j0 + wcol + ni * 16 + li;
-        if (row < M && col < N) {
+        if (!GUARD || (row < M && col < N)) {
           const int64_t off = (int64_t)row * ldc + col;
           T out = alpha * (T)v[r];
           if (beta != T(0)) out += beta * Cb[off];
@@ -441,9 +450,18 @@ void launch_real(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
   // OP_C == OP_T for real scalars
   const int oa = (opA == OP_C) ? OP_T : opA;
   const int ob = (opB == OP_C) ? OP_T : opB;
+  const bool full = (M % 128 == 0) && (N % BN == 0) && (K % 16 == 0);
 #define LAUNCH(OA, OB, BNv, BKv, DB)                                         \
- hipLaunchKernelGGL(( gemm_tiles_k<T, OA, OB, BNv, BKv, DB>), dim3(grid), dim3(block), 0, stream,          \
-      descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks, nblocks)
+  do {                                                                       \
+    if (full && BKv == 16)                                                   \
+     hipLaunchKernelGGL(( gemm_tiles_k<T, OA, OB, BNv, BKv, DB, false>), dim3(grid), dim3(block), 0,         \
+          stream, descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta,     \
+                    mblocks, nblocks);                                       \
+    else                                                                     \
+     hipLaunchKernelGGL(( gemm_tiles_k<T, OA, OB, BNv, BKv, DB, true>), dim3(grid), dim3(block), 0,          \
+          stream, descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta,     \
+                    mblocks, nblocks);                                       \
+  } while (0)
 #define CASE(OA, OB)                                                        \
   if (oa == OA && ob == OB) {                                               \
     if (inplace) {                                                          \
