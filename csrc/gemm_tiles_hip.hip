@@ -450,7 +450,11 @@ void launch_real(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
   // OP_C == OP_T for real scalars
   const int oa = (opA == OP_C) ? OP_T : opA;
   const int ob = (opB == OP_C) ? OP_T : opB;
-  const bool full = (M % 128 == 0) && (N % BN == 0) && (K % 16 == 0);
+  static const int full_opt = [] {
+    const char* v = getenv("DLAF_GEMM_FULLOPT");
+    return v ? atoi(v) : 1;
+  }();
+  const bool full = full_opt && (M % 128 == 0) && (N % BN == 0) && (K % 16 == 0);
 #define LAUNCH(OA, OB, BNv, BKv, DB)                                         \
   do {                                                                       \
     if (full && BKv == 16)                                                   \
