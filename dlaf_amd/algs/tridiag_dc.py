@@ -208,20 +208,50 @@ def _merge(w1, Q1, w2, Q2, rho, device):
     df_idx = np.nonzero(deflated)[0]
     k1 = len(nd_idx)
 
-    # ---- build permuted Q (and apply rotations) on device ----
-    Qb = torch.zeros((n, n), dtype=Q1.dtype, device=device)
-    Qb[:n1, :n1] = Q1
-    Qb[n1:, n1:] = Q2
-    Qb = Qb[:, perm]
-    for (i, j, c, s) in rots:
-        gi = Qb[:, i].clone()
-        gj = Qb[:, j].clone()
-        # Qb <- Qb G with G = [[c, -s], [s, c]]; z' = G^T z zeroes component i
-        Qb[:, i] = c * gi + s * gj
-        Qb[:, j] = -s * gi + c * gj
+    # ---- column assembly without materializing the full permuted Qb ----
+    # (the reference's multiplyEigenvectors also avoids a dense re-layout,
+    # merge.h:974-1076). Columns are gathered straight from the Q1/Q2 blocks;
+    # rotation-involved columns are staged, rotated in order, and patched in.
+    perm_np = perm.cpu().numpy()
 
-    w_out = torch.empty(n, dtype=d_s.dtype, device=device)
-    Q_out = torch.empty_like(Qb)
+    def gather_cols(cols_np):
+        m = len(cols_np)
+        out = torch.zeros((n, m), dtype=Q1.dtype, device=device)
+        if m == 0:
+            return out
+        pc = perm_np[cols_np]
+        top = pc < n1
+        ti = np.nonzero(top)[0]
+        bi = np.nonzero(~top)[0]
+        if len(ti):
+            out[:n1, torch.from_numpy(ti).to(device)] = \
+                Q1[:, torch.from_numpy(pc[ti]).to(device)]
+        if len(bi):
+            out[n1:, torch.from_numpy(bi).to(device)] = \
+                Q2[:, torch.from_numpy(pc[bi] - n1).to(device)]
+        return out
+
+    rot_cols = sorted({c for (i, j, _, _) in rots for c in (i, j)})
+    rot_pos = {c: p for p, c in enumerate(rot_cols)}
+    R = gather_cols(np.array(rot_cols, dtype=np.int64))
+    for (i, j, c, s) in rots:
+        pi, pj = rot_pos[i], rot_pos[j]
+        gi = R[:, pi].clone()
+        gj = R[:, pj].clone()
+        # R <- R G with G = [[c, -s], [s, c]]; z' = G^T z zeroes component i
+        R[:, pi] = c * gi + s * gj
+        R[:, pj] = -s * gi + c * gj
+
+    def patch_rotated(out, cols_np):
+        pos = [(p, rot_pos[c]) for p, c in enumerate(cols_np) if c in rot_pos]
+        if pos:
+            dst = torch.tensor([p for p, _ in pos], dtype=torch.int64, device=device)
+            src = torch.tensor([q for _, q in pos], dtype=torch.int64, device=device)
+            out[:, dst] = R[:, src]
+        return out
+
+    Qnd = patch_rotated(gather_cols(nd_idx), nd_idx)
+    Qdf = patch_rotated(gather_cols(df_idx), df_idx)
 
     if k1 > 0:
         # the HIP one-thread-per-root kernel handles every size in one
@@ -249,18 +279,24 @@ def _merge(w1, Q1, w2, Q2, rho, device):
         U = zh.unsqueeze(1) / delta                  # [k1, k1]
         U = U / torch.linalg.vector_norm(U, dim=0, keepdim=True)
         lam_out = lam.to(device)
-        V_nd = Qb[:, torch.from_numpy(nd_idx).to(device)] @ U.to(Qb.dtype).to(device)
+        V_nd = Qnd @ U.to(Qnd.dtype).to(device)
     else:
         lam_out = torch.empty(0, dtype=d_s.dtype, device=device)
-        V_nd = torch.empty((n, 0), dtype=Qb.dtype, device=device)
+        V_nd = torch.empty((n, 0), dtype=Qnd.dtype, device=device)
 
     all_vals = torch.cat([lam_out, torch.from_numpy(dn[df_idx]).to(device)])
     if negate:
         all_vals = -all_vals
     order = torch.argsort(all_vals)
     w_out = all_vals[order]
-    V = torch.cat([V_nd, Qb[:, torch.from_numpy(df_idx).to(device)]], dim=1)
-    Q_out = V[:, order]
+    # scatter the nd / deflated columns straight to their final positions
+    invp = torch.empty(n, dtype=torch.int64, device=device)
+    invp[order] = torch.arange(n, device=device)
+    Q_out = torch.empty((n, n), dtype=Qnd.dtype, device=device)
+    if k1:
+        Q_out.index_copy_(1, invp[:k1], V_nd)
+    if n - k1:
+        Q_out.index_copy_(1, invp[k1:], Qdf)
     return w_out, Q_out
 
 
