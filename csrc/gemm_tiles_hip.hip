@@ -457,10 +457,14 @@ void launch_real(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
     const char* v = getenv("DLAF_GEMM_FULLOPT");
     return v ? atoi(v) : 0;
   }();
-  const bool full = full_opt && (M % 128 == 0) && (N % BN == 0) && (K % 16 == 0);
+  // NOTE: the guard-free test must use the TEMPLATE tile width (BNv), not
+  // the runtime BN: the inplace branch instantiates the 128-wide kernel even
+  // when the grid is sized with BN=64 (N <= 64), relying on the guards to
+  // mask columns >= N.
+  const bool fullMK = full_opt && (M % 128 == 0) && (K % 16 == 0);
 #define LAUNCH(OA, OB, BNv, BKv, DB)                                         \
   do {                                                                       \
-    if (full && BKv == 16)                                                   \
+    if (fullMK && (N % BNv == 0) && BKv == 16)                               \
      hipLaunchKernelGGL(( gemm_tiles_k<T, OA, OB, BNv, BKv, DB, false>), dim3(grid), dim3(block), 0,         \
           stream, descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta,     \
                     mblocks, nblocks);                                       \
