@@ -67,30 +67,12 @@ def main():
         if device.type == "cuda":
             torch.cuda.synchronize(device)
 
-    def step_eager():
+    def step():
         mat.storage.copy_(pristine)
         cholesky_factorization(UpLo.Lower, mat, grid)
 
-    step = step_eager
-    use_graph = device.type == "cuda" and world_size == 1
     for _ in range(args.warmup):
-        step_eager()
-    if use_graph:
-        # capture the whole factorization step (restore + panel/trailing
-        # sequence incl. the lookahead streams) into ONE hipGraph: removes
-        # per-launch host overhead from the timed region
-        g = torch.cuda.CUDAGraph()
-        try:
-            with torch.cuda.graph(g):
-                step_eager()
-
-            def step_graph():
-                g.replay()
-
-            step = step_graph
-        except Exception as e:
-            print(f"# hipGraph capture unavailable ({e}); eager path", flush=True)
-            step = step_eager
+        step()
     barrier_sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
@@ -130,7 +112,6 @@ def main():
                 "parallelism": f"grid{gr}x{gc}",
                 "global_batch": 1,
                 "seq_len": n,
-                "hipgraph": step is not step_eager,
             },
         }))
 
