@@ -12,7 +12,7 @@ from typing import Optional, Tuple
 
 import torch
 
-from ..types import UpLo, Op, Side, Diag, is_complex, real_dtype
+from ..types import UpLo, Op, Side, Diag
 from ..matrix.matrix import Matrix
 from ..comm.grid import CommGrid
 from .red2band import reduction_to_band, bt_reduction_to_band

@@ -30,7 +30,7 @@ from ..matrix.matrix import Matrix
 from ..comm.grid import CommGrid
 from ..comm import collectives as coll
 from .red2band import panel_qr_, t_factor, _herm_full_dense
-from .band2tridiag import band_to_tridiagonal, bt_band_to_tridiagonal, TridiagResult
+from .band2tridiag import bt_band_to_tridiagonal, TridiagResult
 from .tridiag_dc import tridiagonal_eigensolver
 
 

@@ -26,7 +26,6 @@ from . import _panels as pan
 from .triangular import (
     _bcast_B_col,
     _bcast_B_row,
-    _t,
     _trivial_grid,
     _update_tiles,
 )

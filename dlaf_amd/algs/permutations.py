@@ -16,7 +16,6 @@ import torch
 
 from ..matrix.matrix import Matrix
 from ..comm.grid import CommGrid
-from ..comm import collectives as coll
 
 
 def permute_columns(src: Matrix, perm: torch.Tensor, dst: Matrix,

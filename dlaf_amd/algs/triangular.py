@@ -21,7 +21,6 @@ from __future__ import annotations
 
 from typing import List, Optional, Tuple
 
-import numpy as np
 import torch
 
 from ..types import Side, UpLo, Op, Diag

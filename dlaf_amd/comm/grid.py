@@ -15,7 +15,7 @@ which case it is the trivial 1x1 grid (local-only algorithms).
 
 from __future__ import annotations
 
-from typing import Dict, Optional, Tuple
+from typing import Optional, Tuple
 
 import torch
 import torch.distributed as dist

@@ -9,7 +9,7 @@ same role — reproducing solver inputs/outputs for debugging).
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 

@@ -10,8 +10,6 @@ from __future__ import annotations
 
 import io
 
-import torch
-
 from .matrix import Matrix
 
 

@@ -22,7 +22,6 @@ import torch
 import torch.distributed as dist
 
 from dlaf_amd import CommGrid, Matrix
-from dlaf_amd.types import dtype_char
 
 _DTYPES = {"s": torch.float32, "d": torch.float64,
            "c": torch.complex64, "z": torch.complex128}

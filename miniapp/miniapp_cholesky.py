@@ -5,7 +5,7 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 import torch
 from _harness import run_miniapp, random_spd
-from dlaf_amd import UpLo, cholesky_factorization, max_norm
+from dlaf_amd import UpLo, cholesky_factorization
 from dlaf_amd.types import total_ops
 
 

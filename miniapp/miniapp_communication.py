@@ -6,7 +6,6 @@ import sys, os, time
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 import torch
-import torch.distributed as dist
 from _harness import common_parser, MiniappCtx
 from dlaf_amd.comm import collectives as coll
 

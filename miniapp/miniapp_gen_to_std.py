@@ -3,7 +3,6 @@
 import sys, os
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
-import torch
 from _harness import run_miniapp, random_spd, random_herm
 from dlaf_amd import UpLo, generalized_to_standard, cholesky_factorization
 from dlaf_amd.types import total_ops

@@ -22,7 +22,6 @@ def bench_gemm(dtype=torch.float64, nb=512, ntiles=128, iters=10, opB=Op.Trans, 
     ts = nb * nb
     offs = [i * ts for i in range(ntiles)]
     descs = ops.make_descs(offs, offs, offs)
-    import numpy as np
     dt = torch.from_numpy(descs).to(dev)
     # warmup
     for _ in range(3):
