@@ -183,20 +183,33 @@ __launch_bounds__(256) __global__ void gemm_tiles_k(
     }
   };
   auto compute = [&](int buf) {
+    // software-pipelined fragment loads: ds_reads for step ks+1 issue while
+    // the MFMAs of step ks run (MfmaUtil measured 57% without this — the
+    // read->mma dependency chain stalls the MAI pipe between steps)
+    T af[2][4], bf[2][NFRAG];
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+      af[0][mi] = As[buf][wrow + mi * 16 + li][lk];
+#pragma unroll
+    for (int ni = 0; ni < NFRAG; ++ni)
+      bf[0][ni] = Bs[buf][lk][wcol + ni * 16 + li];
 #pragma unroll
     for (int ks = 0; ks < BK / 4; ++ks) {
-      T af[4], bf[NFRAG];
+      const int cur = ks & 1;
+      if (ks + 1 < BK / 4) {
+        const int nxt = 1 - cur;
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-        af[mi] = As[buf][wrow + mi * 16 + li][ks * 4 + lk];
+        for (int mi = 0; mi < 4; ++mi)
+          af[nxt][mi] = As[buf][wrow + mi * 16 + li][(ks + 1) * 4 + lk];
 #pragma unroll
-      for (int ni = 0; ni < NFRAG; ++ni)
-        bf[ni] = Bs[buf][ks * 4 + lk][wcol + ni * 16 + li];
+        for (int ni = 0; ni < NFRAG; ++ni)
+          bf[nxt][ni] = Bs[buf][(ks + 1) * 4 + lk][wcol + ni * 16 + li];
+      }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
         for (int ni = 0; ni < NFRAG; ++ni)
-          acc[mi][ni] = Mfma<T>::mma(af[mi], bf[ni], acc[mi][ni]);
+          acc[mi][ni] = Mfma<T>::mma(af[cur][mi], bf[cur][ni], acc[mi][ni]);
     }
   };
 
