@@ -83,3 +83,14 @@ def test_cholesky_upper_local():
     U = torch.triu(mat.to_global())
     err = (U.mH @ U - a_ref).abs().max().item()
     assert err < 1e-11 * n, f"err={err}"
+
+
+def test_cholesky_complex64_local():
+    n, nb = 24, 8
+    mat = Matrix.create(n, n, nb, nb, dtype=torch.complex64)
+    mutil.set_random_hermitian_positive_definite(mat, seed=3)
+    a_ref = mat.to_global().to(torch.complex128)
+    cholesky_factorization(UpLo.Lower, mat)
+    got = torch.tril(mat.to_global()).to(torch.complex128)
+    want = torch.linalg.cholesky(a_ref)
+    assert (got - want).abs().max().item() < 1e-3

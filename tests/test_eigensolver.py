@@ -146,3 +146,16 @@ def test_eigensolver_dist_partial_spectrum():
     for res, werr in run_distributed(_dist_partial_worker, 2, args=(1, 2)):
         assert res < 1e-9 * 32, f"res={res}"
         assert werr < 1e-10 * 32, f"werr={werr}"
+
+
+def test_eigensolver_local_complex64():
+    n, nb = 48, 16
+    mat = Matrix.create(n, n, nb, nb, dtype=torch.complex64)
+    mutil.set_random_hermitian(mat, seed=77)
+    a0 = _herm(mat.to_global()).to(torch.complex128)
+    w, evecs = hermitian_eigensolver(UpLo.Lower, mat, band=8)
+    E = evecs.to_global().to(torch.complex128)
+    res = (a0 @ E - E @ torch.diag(w.to(torch.complex128))).abs().max().item()
+    assert res < 1e-2, f"res={res}"
+    wref = np.linalg.eigvalsh(a0.numpy())
+    assert np.abs(np.sort(w.numpy().astype(np.float64)) - wref).max() < 1e-2
