@@ -234,6 +234,73 @@ def _check_args(side: Side, A: Matrix, B: Matrix) -> None:
 # TRSM
 # ---------------------------------------------------------------------------
 
+def _trsm_lln_local_gpu(diagv: Diag, A: Matrix, B: Matrix) -> None:
+    """Local GPU fast path for Left-Lower-NoTrans (the BASELINE TRSM case):
+    lookahead schedule mirroring Cholesky — the row-(k+1) solve runs on the
+    high-priority stream overlapped with the bulk trailing update, descs are
+    read straight off the matrix storage (no panel staging)."""
+    from ..runtime import get_runtime
+    da, db = A.dist, B.dist
+    nt = da.nr_tiles[0]
+    nb = da.nb
+    ntc = db.nr_tiles[1]
+    unit = diagv == Diag.Unit
+    # precompute all diagonal-block inverses (off the critical path)
+    invs = [ops.tri_inverse_full(A.tile((k, k)), lower=True, unit=unit)
+            for k in range(nt)]
+    rt = get_runtime(B.device)
+    sp, su = rt.hp_streams[0], rt.np_streams[0]
+    cur = torch.cuda.current_stream(B.device)
+    sp.wait_stream(cur)
+    su.wait_stream(cur)
+    rowbufs = [torch.empty((ntc, nb, nb), dtype=B.dtype, device=B.device)
+               for _ in range(2)]
+    ev_tail = [None] * nt
+
+    def row_offs(i):
+        return [B.tile_offset((i, j)) for j in range(ntc)]
+
+    for k in range(nt):
+        rowbuf = rowbufs[k % 2]
+        with torch.cuda.stream(sp):
+            # row k needs tail(k-2) (rows >= k), and buffer k%2 was last read
+            # by tail(k-2) — one event covers both
+            if k >= 2 and ev_tail[k - 2] is not None:
+                sp.wait_event(ev_tail[k - 2])
+            # solve row k: rowbuf = inv @ B[k, :], copy back
+            offs = row_offs(k)
+            c = [j * nb * nb for j in range(ntc)]
+            ops.gemm_fused(rowbuf, invs[k], B.storage, ops.make_descs(c, [0] * ntc, offs),
+                           nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, 1.0, 0.0)
+            for j in range(ntc):
+                B.tile((k, j)).copy_(rowbuf[j])
+            ev_s = torch.cuda.Event()
+            ev_s.record(sp)
+            # head: update row k+1 eagerly so the next solve is unblocked
+            if k + 1 < nt:
+                ch = row_offs(k + 1)
+                ak = [A.tile_offset((k + 1, k))] * ntc
+                ops.gemm_fused(B.storage, A.storage, rowbuf, ops.make_descs(ch, ak, c),
+                               nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, -1.0, 1.0)
+        if k + 2 < nt:
+            with torch.cuda.stream(su):
+                su.wait_event(ev_s)
+                ct, at, bt = [], [], []
+                for i in range(k + 2, nt):
+                    ct += row_offs(i)
+                    at += [A.tile_offset((i, k))] * ntc
+                    bt += [j * nb * nb for j in range(ntc)]
+                ops.gemm_fused(B.storage, A.storage, rowbuf, ops.make_descs(ct, at, bt),
+                               nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, -1.0, 1.0)
+                e = torch.cuda.Event()
+                e.record(su)
+                ev_tail[k] = e
+        else:
+            ev_tail[k] = None
+    cur.wait_stream(sp)
+    cur.wait_stream(su)
+
+
 def triangular_solver(side: Side, uplo: UpLo, op: Op, diag: Diag, alpha,
                       A: Matrix, B: Matrix, grid: Optional[CommGrid] = None) -> None:
     """In-place solve op(A) X = alpha B (Left) or X op(A) = alpha B (Right).
@@ -248,6 +315,10 @@ def triangular_solver(side: Side, uplo: UpLo, op: Op, diag: Diag, alpha,
     notrans = op == Op.NoTrans
     if alpha != 1:
         B.storage.mul_(alpha)
+    if (side == Side.Left and lower and notrans and not g.distributed
+            and B.device.type == "cuda"):
+        _trsm_lln_local_gpu(diag, A, B)
+        return
 
     da, db = A.dist, B.dist
     nt = da.nr_tiles[0]
