@@ -255,6 +255,7 @@ def _trsm_lln_local_gpu(diagv: Diag, A: Matrix, B: Matrix) -> None:
     su.wait_stream(cur)
     rowbufs = [torch.empty((ntc, nb, nb), dtype=B.dtype, device=B.device)
                for _ in range(2)]
+    ev_head = [None] * nt
     ev_tail = [None] * nt
 
     def row_offs(i):
@@ -263,8 +264,11 @@ def _trsm_lln_local_gpu(diagv: Diag, A: Matrix, B: Matrix) -> None:
     for k in range(nt):
         rowbuf = rowbufs[k % 2]
         with torch.cuda.stream(sp):
-            # row k needs tail(k-2) (rows >= k), and buffer k%2 was last read
-            # by tail(k-2) — one event covers both
+            # row k's last pending updates: head(k-1) (column k-1 term) and
+            # tail(k-2) (all earlier columns); ev_tail[k-2] also frees buffer
+            # k%2 (last read by tail(k-2)).
+            if k >= 1 and ev_head[k - 1] is not None:
+                sp.wait_event(ev_head[k - 1])
             if k >= 2 and ev_tail[k - 2] is not None:
                 sp.wait_event(ev_tail[k - 2])
             # solve row k: rowbuf = inv @ B[k, :], copy back
@@ -276,27 +280,29 @@ def _trsm_lln_local_gpu(diagv: Diag, A: Matrix, B: Matrix) -> None:
                 B.tile((k, j)).copy_(rowbuf[j])
             ev_s = torch.cuda.Event()
             ev_s.record(sp)
-            # head: update row k+1 eagerly so the next solve is unblocked
-            if k + 1 < nt:
+        # all accumulations into rows > k run on su (no cross-stream writes):
+        # head (row k+1, small — unblocks solve(k+1)) first, then the bulk tail
+        if k + 1 < nt:
+            with torch.cuda.stream(su):
+                su.wait_event(ev_s)
                 ch = row_offs(k + 1)
                 ak = [A.tile_offset((k + 1, k))] * ntc
                 ops.gemm_fused(B.storage, A.storage, rowbuf, ops.make_descs(ch, ak, c),
                                nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, -1.0, 1.0)
-        if k + 2 < nt:
-            with torch.cuda.stream(su):
-                su.wait_event(ev_s)
-                ct, at, bt = [], [], []
-                for i in range(k + 2, nt):
-                    ct += row_offs(i)
-                    at += [A.tile_offset((i, k))] * ntc
-                    bt += [j * nb * nb for j in range(ntc)]
-                ops.gemm_fused(B.storage, A.storage, rowbuf, ops.make_descs(ct, at, bt),
-                               nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, -1.0, 1.0)
-                e = torch.cuda.Event()
-                e.record(su)
-                ev_tail[k] = e
-        else:
-            ev_tail[k] = None
+                eh = torch.cuda.Event()
+                eh.record(su)
+                ev_head[k] = eh
+                if k + 2 < nt:
+                    ct, at, bt = [], [], []
+                    for i in range(k + 2, nt):
+                        ct += row_offs(i)
+                        at += [A.tile_offset((i, k))] * ntc
+                        bt += [j * nb * nb for j in range(ntc)]
+                    ops.gemm_fused(B.storage, A.storage, rowbuf, ops.make_descs(ct, at, bt),
+                                   nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, -1.0, 1.0)
+                    e = torch.cuda.Event()
+                    e.record(su)
+                    ev_tail[k] = e
     cur.wait_stream(sp)
     cur.wait_stream(su)
 
