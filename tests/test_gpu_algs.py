@@ -53,6 +53,27 @@ def test_trsm_gpu(dtype, side, uplo, op):
 
 
 @pytest.mark.parametrize("dtype", [torch.float64, torch.complex128])
+@pytest.mark.parametrize("diag", [Diag.NonUnit, Diag.Unit])
+def test_trsm_lln_fastpath_gpu(dtype, diag):
+    """Left-Lower-NoTrans lookahead fast path: non-divisible sizes (padded
+    edge tiles), alpha scaling, unit diagonal."""
+    m, n, nb, alpha = 1100, 900, 256, 0.5
+    A = Matrix.create(m, m, nb, nb, dtype=dtype, device="cuda")
+    B = Matrix.create(m, n, nb, nb, dtype=dtype, device="cuda")
+    mutil.set_random_hermitian_positive_definite(A, seed=5)
+    mutil.set_random(B, seed=6)
+    a, b0 = A.to_global().cpu(), B.to_global().cpu()
+    triangular_solver(Side.Left, UpLo.Lower, Op.NoTrans, diag, alpha, A, B)
+    torch.cuda.synchronize()
+    tri = torch.tril(a)
+    if diag == Diag.Unit:
+        tri = tri - torch.diag_embed(tri.diagonal()) + torch.eye(m, dtype=tri.dtype)
+    want = torch.linalg.solve(tri, alpha * b0)
+    err = (B.to_global().cpu() - want).abs().max().item()
+    assert err < 1e-9 * (m + n), f"err={err}"
+
+
+@pytest.mark.parametrize("dtype", [torch.float64, torch.complex128])
 def test_trmm_gpu(dtype):
     m, n, nb = 1024, 512, 256
     A = Matrix.create(m, m, nb, nb, dtype=dtype, device="cuda")
