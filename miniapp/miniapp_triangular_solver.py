@@ -5,6 +5,7 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 from _harness import run_miniapp, random_spd, random_general
 from dlaf_amd import Side, UpLo, Op, Diag, triangular_solver
+from dlaf_amd import Matrix, triangular_multiplication
 from dlaf_amd.types import total_ops
 
 
@@ -15,7 +16,10 @@ def extra(p):
 def setup(ctx):
     m = ctx.opts.matrix_size
     n = ctx.opts.n or m
-    return {"a": random_spd(ctx), "b": random_general(ctx, m, n), "n": n}
+    st = {"a": random_spd(ctx), "b": random_general(ctx, m, n), "n": n}
+    if ctx.opts.check_result != "none":
+        st["b0"] = st["b"].storage.clone()
+    return st
 
 
 def run(ctx, st):
@@ -30,5 +34,25 @@ def flops(ctx):
     return total_ops(ctx.dtype, add_mul, add_mul)
 
 
+def check(ctx, st, result):
+    """Residual of the solve, verified by multiplying back on the GPU:
+    max|L·X - B0| / (m · max|B0|) — avoids an O(m^3) host reference."""
+    b = st["b"]
+    bx = Matrix.create(b.dist.m, b.dist.n, b.dist.mb, b.dist.nb,
+                       dtype=b.dtype, device=b.device, grid=ctx.grid)
+    bx.storage.copy_(b.storage)
+    triangular_multiplication(Side.Left, UpLo.Lower, Op.NoTrans, Diag.NonUnit,
+                              1.0, st["a"], bx, ctx.comm_grid)
+    import torch
+    diff = (bx.storage - st["b0"]).abs().max()
+    scale = st["b0"].abs().max()
+    if ctx.grid.distributed:
+        import torch.distributed as dist
+        dist.all_reduce(diff, op=dist.ReduceOp.MAX)
+        dist.all_reduce(scale, op=dist.ReduceOp.MAX)
+    return (diff / (ctx.opts.matrix_size * scale)).item()
+
+
 if __name__ == "__main__":
-    run_miniapp("miniapp_triangular_solver", setup, run, flops, extra=extra)
+    run_miniapp("miniapp_triangular_solver", setup, run, flops, check=check,
+                extra=extra)
