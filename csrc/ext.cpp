@@ -15,6 +15,13 @@
 
 #include "kernels.h"
 
+// csrc/rocblas_batch.cpp
+void lib_gemm_batched(torch::Tensor dt, torch::Tensor ptrC, torch::Tensor ptrA,
+                      torch::Tensor ptrB, int64_t M, int64_t N, int64_t K,
+                      int64_t lda, int64_t ldb, int64_t ldc, int64_t opA,
+                      int64_t opB, double alpha_re, double alpha_im,
+                      double beta_re, double beta_im);
+
 namespace {
 
 #define HIP_CHECK(expr)                                                     \
@@ -342,6 +349,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("factor_invert_block", &factor_invert_block,
         "fused single-workgroup [factor+]invert of a diagonal block");
   m.def("trtri_lower", &trtri_lower, "lower-triangular block inverse");
+  m.def("lib_gemm_batched", &lib_gemm_batched,
+        "rocBLAS pointer-array batched GEMM (uniform tile shape)");
   m.def("potrf_tile", &potrf_tile,
         "in-place tile Cholesky + diagonal-block inverses");
   m.attr("POTRF_BSZ_REAL") = 64;

@@ -86,6 +86,31 @@ def gemm_fused(
     )
 
 
+def ptr_array(base: torch.Tensor, offs) -> torch.Tensor:
+    """Device int64 array of addresses base.data_ptr() + off*itemsize for the
+    rocBLAS pointer-array batched path. ``offs`` in element units."""
+    if not torch.is_tensor(offs):
+        offs = torch.as_tensor(offs, dtype=torch.int64)
+    offs = offs.to(base.device, torch.int64, non_blocking=True)
+    return offs * base.element_size() + base.data_ptr()
+
+
+def gemm_batched_lib(dt_ref: torch.Tensor, ptrC: torch.Tensor, ptrA: torch.Tensor,
+                     ptrB: torch.Tensor, M: int, N: int, K: int,
+                     lda: int, ldb: int, ldc: int, opA: Op, opB: Op,
+                     alpha, beta) -> None:
+    """rocBLAS batched GEMM over precomputed device pointer arrays.
+
+    Uniform-shape batches only (one K-product per C tile, no aliasing among
+    C entries). ~30% faster than the fused kernel on plain nb=512 f64 tiles
+    (57.8 vs 44.1 TF); mixed-ktiles phases keep ``gemm_fused``.
+    """
+    ar, ai = _alpha_parts(alpha)
+    br, bi = _alpha_parts(beta)
+    get_ext().lib_gemm_batched(dt_ref, ptrC, ptrA, ptrB, M, N, K,
+                               lda, ldb, ldc, _opc(opA), _opc(opB), ar, ai, br, bi)
+
+
 def _t(x: torch.Tensor, op: Op) -> torch.Tensor:
     if op is Op.NoTrans:
         return x

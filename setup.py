@@ -23,7 +23,9 @@ setup(
                 "csrc/factor.hip",
                 "csrc/panel_qr.hip",
                 "csrc/secular.hip",
+                "csrc/rocblas_batch.cpp",
             ],
+            libraries=["rocblas"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3"],

@@ -113,7 +113,57 @@ if __name__ == "__main__":
     bench_gemm(torch.complex128, 512, 64)
     bench_torch_bmm(torch.float64, 512, 128)
     bench_torch_bmm(torch.float64, 512, 1024, iters=5)
+    if "--lib" in sys.argv:
+        bench_lib_batched(torch.float64)
+        bench_lib_batched(torch.complex128, nt=44)
     bench_big_dgemm(torch.float64, 16384)
     bench_potrf(torch.float64)
     bench_potrf(torch.complex128, 512)
     bench_trtri(torch.float64)
+
+
+def bench_lib_batched(dtype=torch.float64, nb=512, nt=63, iters=5):
+    """rocBLAS batched vs fused kernel on the POTRF trailing geometry:
+    C[i,j] -= P[i] @ P[j]^T over the lower triangle (i >= j)."""
+    from dlaf_amd.ops import tile_ops as ops
+    from dlaf_amd.types import Op
+    dev = "cuda"
+    panel = torch.randn(nt, nb, nb, dtype=dtype, device=dev) if not dtype.is_complex \
+        else torch.randn(nt, nb, nb, dtype=dtype, device=dev)
+    items = [(i, j) for j in range(nt) for i in range(j, nt)]
+    ntc = len(items)
+    C = torch.randn(ntc, nb, nb, dtype=dtype, device=dev)
+    C2 = C.clone()
+    offc = [e * nb * nb for e in range(ntc)]
+    offa = [i * nb * nb for i, _ in items]
+    offb = [j * nb * nb for _, j in items]
+    pC, pA, pB = (ops.ptr_array(C, offc), ops.ptr_array(panel, offa),
+                  ops.ptr_array(panel, offb))
+    pC2 = ops.ptr_array(C2, offc)
+    d = ops.make_descs(offc, offa, offb)
+    opB = Op.ConjTrans if dtype.is_complex else Op.Trans
+    # numerics: lib vs fused
+    ops.gemm_fused(C.reshape(-1), panel.reshape(-1), panel.reshape(-1), d,
+                   nb, nb, nb, nb, nb, nb, Op.NoTrans, opB, -1.0, 1.0)
+    ops.gemm_batched_lib(C2, pC2, pA, pB, nb, nb, nb, nb, nb, nb,
+                         Op.NoTrans, opB, -1.0, 1.0)
+    torch.cuda.synchronize()
+    print(f"lib vs fused max diff: {(C - C2).abs().max().item():.3e}")
+    import time
+    for tag, fn in (
+        ("fused", lambda: ops.gemm_fused(C.reshape(-1), panel.reshape(-1),
+                                         panel.reshape(-1), d, nb, nb, nb, nb, nb, nb,
+                                         Op.NoTrans, opB, -1.0, 1.0)),
+        ("rocblas_batched", lambda: ops.gemm_batched_lib(
+            C, pC, pA, pB, nb, nb, nb, nb, nb, nb, Op.NoTrans, opB, -1.0, 1.0)),
+    ):
+        fn(); torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            fn()
+        torch.cuda.synchronize()
+        dt_s = (time.perf_counter() - t0) / iters
+        mul = 4 if dtype.is_complex else 1
+        fl = 2.0 * nb**3 * ntc * mul
+        print(f"trailing {tag} {dtype} nb={nb} tiles={ntc}: {dt_s*1e3:.2f} ms  "
+              f"{fl/dt_s/1e12:.2f} TFLOP/s", flush=True)
