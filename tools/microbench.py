@@ -105,22 +105,6 @@ def bench_trtri(dtype=torch.float64, nb=512, iters=20):
     print(f"trtri_tile {dtype} nb={nb}: {dt_s*1e6:.0f} us")
 
 
-if __name__ == "__main__":
-    print(torch.cuda.get_device_name(0))
-    bench_gemm(torch.float64, 512, 128)
-    bench_gemm(torch.float64, 512, 1024, iters=5)
-    bench_gemm(torch.float32, 512, 128)
-    bench_gemm(torch.complex128, 512, 64)
-    bench_torch_bmm(torch.float64, 512, 128)
-    bench_torch_bmm(torch.float64, 512, 1024, iters=5)
-    if "--lib" in sys.argv:
-        bench_lib_batched(torch.float64)
-        bench_lib_batched(torch.complex128, nt=44)
-    bench_big_dgemm(torch.float64, 16384)
-    bench_potrf(torch.float64)
-    bench_potrf(torch.complex128, 512)
-    bench_trtri(torch.float64)
-
 
 def bench_lib_batched(dtype=torch.float64, nb=512, nt=63, iters=5):
     """rocBLAS batched vs fused kernel on the POTRF trailing geometry:
@@ -167,3 +151,19 @@ def bench_lib_batched(dtype=torch.float64, nb=512, nt=63, iters=5):
         fl = 2.0 * nb**3 * ntc * mul
         print(f"trailing {tag} {dtype} nb={nb} tiles={ntc}: {dt_s*1e3:.2f} ms  "
               f"{fl/dt_s/1e12:.2f} TFLOP/s", flush=True)
+
+if __name__ == "__main__":
+    print(torch.cuda.get_device_name(0))
+    bench_gemm(torch.float64, 512, 128)
+    bench_gemm(torch.float64, 512, 1024, iters=5)
+    bench_gemm(torch.float32, 512, 128)
+    bench_gemm(torch.complex128, 512, 64)
+    bench_torch_bmm(torch.float64, 512, 128)
+    bench_torch_bmm(torch.float64, 512, 1024, iters=5)
+    if "--lib" in sys.argv:
+        bench_lib_batched(torch.float64)
+        bench_lib_batched(torch.complex128, nt=44)
+    bench_big_dgemm(torch.float64, 16384)
+    bench_potrf(torch.float64)
+    bench_potrf(torch.complex128, 512)
+    bench_trtri(torch.float64)
