@@ -213,13 +213,13 @@ def _cholesky_local(mat: Matrix) -> None:
                     if k >= 1 and ev_tail[k - 1] is not None:
                         sp.wait_event(ev_tail[k - 1])
                     ops.gemm_fused(st, st, st, head, nb, nb, nb, nb, nb, nb,
-                                   Op.NoTrans, opc, -1.0, 1.0)
+                                   Op.NoTrans, opc, -1.0, 1.0, uniform=True)
             tail = table.get(("k", k, "tail"))
             if tail is not None:
                 with torch.cuda.stream(su):
                     su.wait_event(ev_p)
                     ops.gemm_fused(st, st, st, tail, nb, nb, nb, nb, nb, nb,
-                                   Op.NoTrans, opc, -1.0, 1.0)
+                                   Op.NoTrans, opc, -1.0, 1.0, uniform=True)
                     ev = torch.cuda.Event()
                     ev.record(su)
                     ev_tail[k] = ev
@@ -424,14 +424,16 @@ def _cholesky_dist_gpu(mat: Matrix, grid: CommGrid) -> None:
             head = table.get(("k", k, "head"))
             if head is not None:
                 ops.gemm_fused(mat.storage, col_panel.storage, row_panel.storage, head,
-                               nb, nb, nb, nb, nb, nb, Op.NoTrans, opc, -1.0, 1.0)
+                               nb, nb, nb, nb, nb, nb, Op.NoTrans, opc, -1.0, 1.0,
+                               uniform=True)
             e = torch.cuda.Event()
             e.record(su)
             ev_head[k] = e
             tail = table.get(("k", k, "trail"))
             if tail is not None:
                 ops.gemm_fused(mat.storage, col_panel.storage, row_panel.storage, tail,
-                               nb, nb, nb, nb, nb, nb, Op.NoTrans, opc, -1.0, 1.0)
+                               nb, nb, nb, nb, nb, nb, Op.NoTrans, opc, -1.0, 1.0,
+                               uniform=True)
             e2 = torch.cuda.Event()
             e2.record(su)
             ev_tail[k] = e2

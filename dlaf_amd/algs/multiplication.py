@@ -99,7 +99,8 @@ def hermitian_multiplication(side: Side, uplo: UpLo, alpha, A: Matrix, B: Matrix
                 if lcC:
                     if dev.type == "cuda":
                         ops.gemm_fused(C.storage, hd, rowpB.storage, ops.make_descs(c, [0] * len(c), b),
-                                       nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, alpha, 1.0)
+                                       nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, alpha, 1.0,
+                                       uniform=True)
                     else:
                         for lj in range(lcC):
                             j = dc.global_tile_of_local((0, lj))[1]
@@ -135,7 +136,8 @@ def hermitian_multiplication(side: Side, uplo: UpLo, alpha, A: Matrix, B: Matrix
                 if lrC:
                     if dev.type == "cuda":
                         ops.gemm_fused(C.storage, colpB.storage, hd, ops.make_descs(c, a, [0] * len(c)),
-                                       nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, alpha, 1.0)
+                                       nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, alpha, 1.0,
+                                       uniform=True)
                     else:
                         for li in range(lrC):
                             i = dc.global_tile_of_local((li, 0))[0]

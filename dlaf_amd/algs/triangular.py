@@ -59,7 +59,7 @@ def _update_tiles(mat: Matrix, items: List[Tuple[int, int]], colp: Panel, rowp: 
         a = [colp.offset(li) for li, _ in items]
         b = [rowp.offset(lj) for _, lj in items]
         ops.gemm_fused(mat.storage, colp.storage, rowp.storage, ops.make_descs(c, a, b),
-                       nb, nb, nb, nb, nb, nb, opA, opB, alpha, beta)
+                       nb, nb, nb, nb, nb, nb, opA, opB, alpha, beta, uniform=True)
     else:
         for li, lj in items:
             ops.gemm_tile(mat.local_tile((li, lj)), colp.slot(li), rowp.slot(lj),
@@ -78,7 +78,8 @@ def _left_mul_row(mat: Matrix, k: int, M: torch.Tensor, opM: Op, rowp: Panel,
         a = [0] * len(c)
         b = [mat.tile_offset((k, d.global_tile_of_local((0, lj))[1])) for lj in range(lj0, lj1)]
         ops.gemm_fused(rowp.storage, M, mat.storage, ops.make_descs(c, a, b),
-                       nb, nb, nb, M.stride(0), nb, nb, opM, Op.NoTrans, 1.0, 0.0)
+                       nb, nb, nb, M.stride(0), nb, nb, opM, Op.NoTrans, 1.0, 0.0,
+                       uniform=True)
         for lj in range(lj0, lj1):
             j = d.global_tile_of_local((0, lj))[1]
             mat.tile((k, j)).copy_(rowp.slot(lj))
@@ -120,7 +121,8 @@ def _col_from_panel_mul(mat: Matrix, k: int, colp: Panel, M: torch.Tensor, opM: 
         c = [mat.tile_offset((d.global_tile_of_local((li, 0))[0], k)) for li in range(li0, li1)]
         a = [colp.offset(li) for li in range(li0, li1)]
         ops.gemm_fused(mat.storage, colp.storage, M, ops.make_descs(c, a, [0] * len(c)),
-                       nb, nb, nb, nb, M.stride(0), nb, Op.NoTrans, opM, 1.0, 0.0)
+                       nb, nb, nb, nb, M.stride(0), nb, Op.NoTrans, opM, 1.0, 0.0,
+                       uniform=True)
     else:
         for li in range(li0, li1):
             i = d.global_tile_of_local((li, 0))[0]
@@ -138,7 +140,8 @@ def _row_from_panel_mul(mat: Matrix, k: int, rowp: Panel, M: torch.Tensor, opM: 
         c = [mat.tile_offset((k, d.global_tile_of_local((0, lj))[1])) for lj in range(lj0, lj1)]
         b = [rowp.offset(lj) for lj in range(lj0, lj1)]
         ops.gemm_fused(mat.storage, M, rowp.storage, ops.make_descs(c, [0] * len(c), b),
-                       nb, nb, nb, M.stride(0), nb, nb, opM, Op.NoTrans, 1.0, 0.0)
+                       nb, nb, nb, M.stride(0), nb, nb, opM, Op.NoTrans, 1.0, 0.0,
+                       uniform=True)
     else:
         for lj in range(lj0, lj1):
             j = d.global_tile_of_local((0, lj))[1]
@@ -275,7 +278,8 @@ def _trsm_lln_local_gpu(diagv: Diag, A: Matrix, B: Matrix) -> None:
             offs = row_offs(k)
             c = [j * nb * nb for j in range(ntc)]
             ops.gemm_fused(rowbuf, invs[k], B.storage, ops.make_descs(c, [0] * ntc, offs),
-                           nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, 1.0, 0.0)
+                           nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, 1.0, 0.0,
+                           uniform=True)
             for j in range(ntc):
                 B.tile((k, j)).copy_(rowbuf[j])
             ev_s = torch.cuda.Event()
@@ -288,7 +292,8 @@ def _trsm_lln_local_gpu(diagv: Diag, A: Matrix, B: Matrix) -> None:
                 ch = row_offs(k + 1)
                 ak = [A.tile_offset((k + 1, k))] * ntc
                 ops.gemm_fused(B.storage, A.storage, rowbuf, ops.make_descs(ch, ak, c),
-                               nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, -1.0, 1.0)
+                               nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, -1.0, 1.0,
+                               uniform=True)
                 eh = torch.cuda.Event()
                 eh.record(su)
                 ev_head[k] = eh
@@ -299,7 +304,8 @@ def _trsm_lln_local_gpu(diagv: Diag, A: Matrix, B: Matrix) -> None:
                         at += [A.tile_offset((i, k))] * ntc
                         bt += [j * nb * nb for j in range(ntc)]
                     ops.gemm_fused(B.storage, A.storage, rowbuf, ops.make_descs(ct, at, bt),
-                                   nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, -1.0, 1.0)
+                                   nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, -1.0, 1.0,
+                                   uniform=True)
                     e = torch.cuda.Event()
                     e.record(su)
                     ev_tail[k] = e

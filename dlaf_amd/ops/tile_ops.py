@@ -68,18 +68,35 @@ def gemm_fused(
     alpha,
     beta,
     inplace: bool = False,
+    uniform: bool = False,
 ) -> None:
     """GPU fused batched GEMM. ``descs``: np.ndarray [n,6] or device tensor.
 
     Set ``inplace=True`` when a desc's C block aliases its A block (panel
     applies) — it selects the kernel geometry that reads all of A before
     writing C.
+
+    Set ``uniform=True`` when the caller guarantees ktiles == 1 and no two
+    descs share a C block: real-dtype batches then route to rocBLAS
+    pointer-array batched GEMM (65.7 vs 45.3 TF on the POTRF trailing shape,
+    profiles/microbench_r1.log). Complex stays on the fused CDNA4 kernel
+    (rocblas_zgemm_batched measured 2.7x SLOWER than it).
     """
     ext = get_ext()
     if isinstance(descs, np.ndarray):
         descs = torch.from_numpy(descs).to(C_base.device, non_blocking=True)
     ar, ai = _alpha_parts(alpha)
     br, bi = _alpha_parts(beta)
+    if (uniform and not inplace and C_base.is_cuda
+            and C_base.dtype in (torch.float64, torch.float32)):
+        dd = descs.view(-1, 6)
+        es = C_base.element_size()
+        pc = dd[:, 0] * es + C_base.data_ptr()
+        pa = dd[:, 1] * es + A_base.data_ptr()
+        pb = dd[:, 2] * es + B_base.data_ptr()
+        ext.lib_gemm_batched(C_base, pc, pa, pb, M, N, K, lda, ldb, ldc,
+                             _opc(opA), _opc(opB), ar, ai, br, bi)
+        return
     ext.batch_gemm(
         C_base.reshape(-1), A_base.reshape(-1), B_base.reshape(-1), descs,
         M, N, K, lda, ldb, ldc, _opc(opA), _opc(opB), ar, ai, br, bi, inplace,
