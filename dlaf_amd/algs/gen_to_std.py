@@ -106,7 +106,7 @@ def generalized_to_standard(uplo: UpLo, mat_a: Matrix, mat_l: Matrix,
                               mat_a.tile_offset((k, d.global_tile_of_local((0, lj))[1])))
                              for lj in range(lj_prev)]
                     ops.gemm_items(rowpAp.storage, linv, mat_a.storage, items, nb,
-                                   Op.NoTrans, Op.NoTrans, 1.0, 0.0)
+                                   Op.NoTrans, Op.NoTrans, 1.0, 0.0, uniform=True)
                     for lj in range(lj_prev):
                         j = d.global_tile_of_local((0, lj))[1]
                         mat_a.tile((k, j)).copy_(rowpAp.slot(lj))
@@ -123,7 +123,7 @@ def generalized_to_standard(uplo: UpLo, mat_a: Matrix, mat_l: Matrix,
             items = [(mat_a.local_tile_offset(li, lj), colpL.offset(li), rowpAp.offset(lj))
                      for li in range(li0, lr) for lj in range(lj_prev)]
             ops.gemm_items(mat_a.storage, colpL.storage, rowpAp.storage, items, nb,
-                           Op.NoTrans, Op.NoTrans, -1.0, 1.0)
+                           Op.NoTrans, Op.NoTrans, -1.0, 1.0, uniform=True)
 
         # 2) diagonal tile transform
         if (d.rank_row, d.rank_col) == (kr, kc):
@@ -145,7 +145,7 @@ def generalized_to_standard(uplo: UpLo, mat_a: Matrix, mat_l: Matrix,
             hA = _herm_full(Akk, lower=True)
             ops.gemm_items(mat_a.storage, mat_l.storage, hA,
                            [(o, lo, 0) for o, lo in zip(offs, loffs)], nb,
-                           Op.NoTrans, Op.NoTrans, -0.5, 1.0)
+                           Op.NoTrans, Op.NoTrans, -0.5, 1.0, uniform=True)
         # broadcast the updated A panel (direct + transposed)
         pan.bcast_col_panel(mat_a, g, k, li0, lr, colpA)
         pan.transpose_col_to_row(d, g, colpA, rowpA, lj0, lc)
@@ -159,9 +159,9 @@ def generalized_to_standard(uplo: UpLo, mat_a: Matrix, mat_l: Matrix,
         trip_a = [(c, colpA.offset(li), rowpL.offset(lj)) for c, li, lj in items]
         trip_l = [(c, colpL.offset(li), rowpA.offset(lj)) for c, li, lj in items]
         ops.gemm_items(mat_a.storage, colpA.storage, rowpL.storage, trip_a, nb,
-                       Op.NoTrans, opc, -1.0, 1.0)
+                       Op.NoTrans, opc, -1.0, 1.0, uniform=True)
         ops.gemm_items(mat_a.storage, colpL.storage, rowpA.storage, trip_l, nb,
-                       Op.NoTrans, opc, -1.0, 1.0)
+                       Op.NoTrans, opc, -1.0, 1.0, uniform=True)
 
         # 5) second -1/2 L[:,k] A[k,k] correction
         if d.rank_col == kc and li0 < lr:
@@ -172,4 +172,4 @@ def generalized_to_standard(uplo: UpLo, mat_a: Matrix, mat_l: Matrix,
                      for li in range(li0, lr)]
             ops.gemm_items(mat_a.storage, mat_l.storage, hA,
                            [(o, lo, 0) for o, lo in zip(offs, loffs)], nb,
-                           Op.NoTrans, Op.NoTrans, -0.5, 1.0)
+                           Op.NoTrans, Op.NoTrans, -0.5, 1.0, uniform=True)

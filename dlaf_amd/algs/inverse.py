@@ -25,6 +25,7 @@ from __future__ import annotations
 
 from typing import Optional
 
+import numpy as np
 import torch
 
 from ..types import UpLo, Diag, Op, is_complex
@@ -90,13 +91,33 @@ def triangular_inverse(uplo: UpLo, diag: Diag, mat: Matrix,
         pan.transpose_col_to_row(d, g, colpL, rowpL, lj0, lc)
         if li0 < lr:
             acc.storage[li0:lr].zero_()
-            items = []
-            for li in range(li0, lr):
-                i = d.global_tile_of_local((li, 0))[0]
-                for lj in range(lj0, d.next_local_tile_col(i + 1)):
-                    items.append((acc.offset(li), mat.local_tile_offset(li, lj), rowpL.offset(lj)))
-            ops.gemm_items(acc.storage, mat.storage, rowpL.storage, items, nb,
-                           Op.NoTrans, Op.NoTrans, 1.0, 1.0)
+            if mat.device.type == "cuda":
+                # one desc per output row with ktiles accumulation: the k-sum
+                # over j MUST live inside a single desc — descs sharing a C
+                # block race (each workgroup does a non-atomic C += update)
+                rows6 = []
+                ts = nb * nb
+                for li in range(li0, lr):
+                    i = d.global_tile_of_local((li, 0))[0]
+                    cnt = d.next_local_tile_col(i + 1) - lj0
+                    if cnt > 0:
+                        rows6.append([acc.offset(li),
+                                      mat.local_tile_offset(li, lj0),
+                                      rowpL.offset(lj0), cnt, ts, ts])
+                if rows6:
+                    ops.gemm_fused(acc.storage, mat.storage, rowpL.storage,
+                                   np.array(rows6, dtype=np.int64),
+                                   nb, nb, nb, nb, nb, nb,
+                                   Op.NoTrans, Op.NoTrans, 1.0, 1.0)
+            else:
+                items = []
+                for li in range(li0, lr):
+                    i = d.global_tile_of_local((li, 0))[0]
+                    for lj in range(lj0, d.next_local_tile_col(i + 1)):
+                        items.append((acc.offset(li), mat.local_tile_offset(li, lj),
+                                      rowpL.offset(lj)))
+                ops.gemm_items(acc.storage, mat.storage, rowpL.storage, items, nb,
+                               Op.NoTrans, Op.NoTrans, 1.0, 1.0)
             if g.row_group is not None:
                 coll.reduce_sum(acc.range_view(li0, lr), g.global_rank_of_row_member(kc),
                                 g.row_group)
@@ -110,7 +131,7 @@ def triangular_inverse(uplo: UpLo, diag: Diag, mat: Matrix,
                      for li in range(li0, lr)]
                 items = list(zip(c, [acc.offset(li) for li in range(li0, lr)], [0] * len(c)))
                 ops.gemm_items(mat.storage, acc.storage, inv, items, nb,
-                               Op.NoTrans, Op.NoTrans, -1.0, 0.0)
+                               Op.NoTrans, Op.NoTrans, -1.0, 0.0, uniform=True)
             kr = d.rank_of_tile_row(k)
             if d.rank_row == kr:
                 mat.tile((k, k)).copy_(inv)
@@ -160,6 +181,6 @@ def inverse_from_cholesky_factor(uplo: UpLo, mat: Matrix,
                 trip = (mat.local_tile_offset(li, lj), colpX.offset(li), rowpX.offset(lj))
                 (first if i == k else accum).append(trip)
         ops.gemm_items(mat.storage, colpX.storage, rowpX.storage, first, nb,
-                       opc, Op.NoTrans, 1.0, 0.0)
+                       opc, Op.NoTrans, 1.0, 0.0, uniform=True)
         ops.gemm_items(mat.storage, colpX.storage, rowpX.storage, accum, nb,
-                       opc, Op.NoTrans, 1.0, 1.0)
+                       opc, Op.NoTrans, 1.0, 1.0, uniform=True)

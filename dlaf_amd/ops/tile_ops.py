@@ -229,18 +229,22 @@ def gemm_items(
     alpha,
     beta,
     inplace: bool = False,
+    uniform: bool = False,
 ) -> None:
     """Tile-triple GEMM over element offsets into flat bases, CPU or GPU.
 
-    ``items``: sequence of (c_off, a_off, b_off). GPU: ONE fused kernel launch.
-    CPU: loop over reshaped nb x nb views (the reference MC backend analog).
+    ``items``: sequence of (c_off, a_off, b_off). GPU: ONE fused kernel launch
+    (``uniform=True`` additionally allows the rocBLAS batched route — caller
+    guarantees unique C offsets). CPU: loop over reshaped nb x nb views (the
+    reference MC backend analog).
     """
     if not len(items):
         return
     if C_base.is_cuda:
         c, a, b = zip(*items)
         gemm_fused(C_base, A_base, B_base, make_descs(c, a, b),
-                   nb, nb, nb, nb, nb, nb, opA, opB, alpha, beta, inplace=inplace)
+                   nb, nb, nb, nb, nb, nb, opA, opB, alpha, beta, inplace=inplace,
+                   uniform=uniform)
         return
     ts = nb * nb
     cv, av, bv = C_base.reshape(-1), A_base.reshape(-1), B_base.reshape(-1)
