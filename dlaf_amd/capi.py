@@ -111,15 +111,33 @@ def _unwrap_local(mat: Matrix, a_local) -> None:
         a_local[:] = t.numpy()
 
 
+def _potrf_info(mat: Matrix) -> int:
+    """info > 0 when the input was not positive definite: a failed pivot
+    sqrt produces NaN on that diagonal (reference counterpart: the
+    cusolver-info device assert, ``src/cusolver/assert_info.cu:35``; here
+    the check is an O(n) diagonal scan returned as ScaLAPACK-style info)."""
+    nt = mat.dist.nr_tiles[0]
+    for k in range(nt):
+        if mat.dist.rank_of_tile((k, k)) != (mat.dist.rank_row, mat.dist.rank_col):
+            continue
+        d = mat.tile((k, k)).diagonal()
+        bad = torch.isnan(d.real if d.is_complex() else d)
+        if bool(bad.any()):
+            return k * mat.dist.mb + int(bad.int().argmax()) + 1
+    return 0
+
+
 def dlaf_cholesky_factorization(ctx: int, uplo: str, a_local, desc: DLAF_descriptor) -> int:
-    """``dlaf_cholesky_factorization_{s,d,c,z}`` analog; returns info (0 = ok)."""
+    """``dlaf_cholesky_factorization_{s,d,c,z}`` analog; returns info (0 = ok,
+    > 0 = leading minor of that order not positive definite)."""
     assert uplo.upper() == "L"
     grid = _grid(ctx)
     dev = _device_for(grid)
     mat = _wrap_local(a_local, desc, grid, dev)
     cholesky_factorization(UpLo.Lower, mat, grid if grid.distributed else None)
+    info = _potrf_info(mat)
     _unwrap_local(mat, a_local)
-    return 0
+    return info
 
 
 def dlaf_inverse_from_cholesky_factor(ctx: int, uplo: str, a_local,

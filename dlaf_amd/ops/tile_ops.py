@@ -199,8 +199,13 @@ def potrf_tile(tile: torch.Tensor, dinv: Optional[torch.Tensor] = None) -> Optio
         ddesc = _potrf_descs(n, tile.stride(0), tile.dtype, tile.device)
         get_ext().potrf_tile(tile, n, tile.stride(0), dinv, ddesc)
         return dinv
-    L = torch.linalg.cholesky(tile)
+    L, inf = torch.linalg.cholesky_ex(tile)
     tile.copy_(L)
+    if int(inf) > 0:
+        # match the GPU kernel's failure mode (sqrt of a bad pivot -> NaN on
+        # that diagonal) so callers detect non-SPD inputs uniformly via the
+        # diagonal scan in capi._potrf_info
+        tile[int(inf) - 1, int(inf) - 1] = float("nan")
     return None
 
 

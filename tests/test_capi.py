@@ -73,3 +73,16 @@ def test_capi_potrf_dist():
     errs = run_distributed(_dist_capi_worker, 4, args=(2, 2))
     for e in errs:
         assert e < 1e-10, f"err={e}"
+
+
+def test_potrf_info_non_spd():
+    """Non-positive-definite input reports ScaLAPACK-style info > 0."""
+    import numpy as np
+    from dlaf_amd import capi
+    n, nb = 64, 32
+    ctx = capi.dlaf_create_grid(1, 1)
+    a = -np.eye(n, order="F")  # negative definite
+    info = capi.dlaf_cholesky_factorization(
+        ctx, "L", a, capi.DLAF_descriptor(n, n, nb, nb, ld=n))
+    assert info > 0
+    capi.dlaf_free_grid(ctx)
