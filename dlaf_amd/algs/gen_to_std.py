@@ -85,6 +85,13 @@ def generalized_to_standard(uplo: UpLo, mat_a: Matrix, mat_l: Matrix,
     l_ws = torch.empty((nb, nb), dtype=dt, device=dev)
     a_ws = torch.empty((nb, nb), dtype=dt, device=dev)
 
+    # hoist the diagonal-block inversions (L is read-only here; each
+    # tri_inverse_full is a multi-launch recursion off the per-k critical path)
+    linvs = []
+    for k in range(nt):
+        Lkk = pan.bcast_diag_to_all(mat_l, g, k, l_ws)
+        linvs.append(ops.tri_inverse_full(Lkk, lower=True))
+
     for k in range(nt):
         kr, kc = d.rank_of_tile((k, k))
         li0k = d.next_local_tile_row(k)
@@ -100,7 +107,7 @@ def generalized_to_standard(uplo: UpLo, mat_a: Matrix, mat_l: Matrix,
         # 1) deferred forward substitution: finalize row k of panels j < k
         if k > 0:
             if d.rank_row == kr:
-                linv = ops.tri_inverse_full(Lkk, lower=True)
+                linv = linvs[k]
                 if dev.type == "cuda":
                     items = [(rowpAp.offset(lj), 0,
                               mat_a.tile_offset((k, d.global_tile_of_local((0, lj))[1])))
@@ -132,7 +139,7 @@ def generalized_to_standard(uplo: UpLo, mat_a: Matrix, mat_l: Matrix,
         # 3) panel k partial update (ranks in the owning column)
         Akk = pan.bcast_diag_to_col(mat_a, g, k, a_ws)
         if d.rank_col == kc and li0 < lr:
-            linv = ops.tri_inverse_full(Lkk, lower=True)
+            linv = linvs[k]
             offs = [mat_a.tile_offset((d.global_tile_of_local((li, 0))[0], k))
                     for li in range(li0, lr)]
             loffs = [mat_l.tile_offset((d.global_tile_of_local((li, 0))[0], k))

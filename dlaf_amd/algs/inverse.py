@@ -79,6 +79,16 @@ def triangular_inverse(uplo: UpLo, diag: Diag, mat: Matrix,
             if d.is_tile_local((k, k)):
                 saved_diag[k] = mat.tile((k, k)).diagonal().clone()
 
+    # hoist ALL diagonal-block inversions out of the sequential k-loop: they
+    # only read original L diagonal tiles (never touched before step k), and
+    # on GPU each tri_inverse_full is a multi-launch recursion that otherwise
+    # sits on the critical path of every step
+    invs = {}
+    for k in range(nt - 1, -1, -1):
+        dtile = pan.bcast_diag_to_col(mat, g, k, diag_ws)
+        if d.rank_col == d.rank_of_tile_col(k):
+            invs[k] = ops.tri_inverse_full(dtile, lower=True, unit=unit)
+
     for k in range(nt - 1, -1, -1):
         kc = d.rank_of_tile_col(k)
         li0 = d.next_local_tile_row(k + 1)
@@ -121,10 +131,9 @@ def triangular_inverse(uplo: UpLo, diag: Diag, mat: Matrix,
             if g.row_group is not None:
                 coll.reduce_sum(acc.range_view(li0, lr), g.global_rank_of_row_member(kc),
                                 g.row_group)
-        # 2) invert diagonal block on the owning column, write column k
-        dtile = pan.bcast_diag_to_col(mat, g, k, diag_ws)
+        # 2) apply the precomputed diagonal-block inverse, write column k
         if d.rank_col == kc:
-            inv = ops.tri_inverse_full(dtile, lower=True, unit=unit)
+            inv = invs[k]
             if li0 < lr:
                 # X[i,k] = -acc[i] @ inv  (write into matrix tiles)
                 c = [mat.tile_offset((d.global_tile_of_local((li, 0))[0], k))
