@@ -199,26 +199,13 @@ def _cholesky_local(mat: Matrix) -> None:
         sp.wait_stream(cur)
         su.wait_stream(cur)
         ev_tail = [None] * nt
-        # panel scratch: X = A_panel @ inv(L_kk)^H lands here in ONE uniform
-        # batched GEMM (rocBLAS for real dtypes), then one strided copy back —
-        # replaces the 16-launch per-64-block TRSM chain on the critical path
-        pscratch = torch.empty((nt, nb, nb), dtype=mat.dtype, device=mat.device)
-        ts = nb * nb
         for k in range(nt):
             with torch.cuda.stream(sp):
                 if k >= 2 and ev_tail[k - 2] is not None:
                     sp.wait_event(ev_tail[k - 2])
                 diag = mat.tile((k, k))
                 ops.potrf_tile(diag, dinv)
-                cnt = nt - 1 - k
-                if cnt > 0:
-                    tinv = ops.tri_inverse_full(diag, lower=True)
-                    c = [j * ts for j in range(cnt)]
-                    a = [mat.tile_offset((i, k)) for i in range(k + 1, nt)]
-                    ops.gemm_fused(pscratch, st, tinv, ops.make_descs(c, a, [0] * cnt),
-                                   nb, nb, nb, nb, tinv.stride(0), nb,
-                                   Op.NoTrans, opc, 1.0, 0.0, uniform=True)
-                    mat.storage[k + 1:nt, k].copy_(pscratch[:cnt])
+                _run_trsm_panel(table, ("k", k), st, diag, dinv, nb, opc)
                 ev_p = torch.cuda.Event()
                 ev_p.record(sp)
                 head = table.get(("k", k, "head"))
