@@ -114,9 +114,10 @@ def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
     exactly those the ascending order keeps correctly ordered; the disjoint
     pairs commute).
     """
+    from ..config import get_tune_parameters
     if group_size is None:
-        from ..config import get_tune_parameters
         group_size = get_tune_parameters().bt_band_to_tridiag_hh_apply_group_size
+    merge = max(1, get_tune_parameters().bt_band_to_tridiag_window_merge)
     n, nE = E.shape
     b = tri.band
     dev = E.device
@@ -161,10 +162,57 @@ def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
                                              torch.zeros_like(safe)))
         T = torch.linalg.solve_triangular(M, eyeG.expand(nwin, G, G).contiguous(), upper=True)
         T = torch.where(mask, torch.zeros_like(T), T)
-        # apply windows in ascending k (ordering constraint across overlaps)
-        for k in range(nwin):
-            base = 1 + s0 + k * b
-            seg = Epad[base:base + H]
-            W = T[k] @ (Vg[k].mH @ seg)
-            seg -= Vg[k] @ W
+        # Merge m consecutive windows into one block-WY apply: the product
+        # Q_{k+m-1}...Q_k of compact-WY transforms is itself compact-WY with
+        #   Vc = [V_k .. V_{k+m-1}] (window j shifted j*b rows down) and
+        #   Tc[j, :j] = -T_j (V_j^H [V_k..V_{j-1}]) Tc[:j, :j]
+        # (windows with row distance >= ceil(H/b) have zero overlap, so the
+        # cross-Grams S vanish beyond two sub-block-diagonals). This keeps
+        # the ordering constraint exactly while making the E-traversal GEMMs
+        # m-times wider and the serial window chain m-times shorter.
+        m = max(1, min(merge, nwin))
+        if m > 1:
+            Hm = H + (m - 1) * b
+            nmerge = (nwin + m - 1) // m
+            # pad the window axis to a multiple of m with zero windows
+            if nmerge * m != nwin:
+                pad = nmerge * m - nwin
+                Vg = torch.cat([Vg, torch.zeros((pad, H, G), dtype=Vg.dtype, device=dev)])
+                T = torch.cat([T, torch.zeros((pad, G, G), dtype=T.dtype, device=dev)])
+            # Vc[km, row, j*G+g] = Vg[km*m+j, row - j*b, g]
+            Vgm = Vg.view(nmerge, m, H, G)
+            Vc = torch.zeros((nmerge, Hm, m * G), dtype=Vg.dtype, device=dev)
+            for j in range(m):
+                Vc[:, j * b:j * b + H, j * G:(j + 1) * G] = Vgm[:, j]
+            Tm = T.view(nmerge, m, G, G)
+            Tc = torch.zeros((nmerge, m * G, m * G), dtype=T.dtype, device=dev)
+            for j in range(m):
+                Tc[:, j * G:(j + 1) * G, j * G:(j + 1) * G] = Tm[:, j]
+            for j in range(1, m):
+                # S_ji = V_j^H V_i with row alignment (j-i)*b; zero if no overlap
+                S = torch.zeros((nmerge, G, j * G), dtype=T.dtype, device=dev)
+                for i in range(j):
+                    d0 = (j - i) * b
+                    if d0 < H:
+                        S[:, :, i * G:(i + 1) * G] = (
+                            Vgm[:, j, :H - d0].mH @ Vgm[:, i, d0:])
+                row = -(Tm[:, j] @ (S @ Tc[:, :j * G, :j * G]))
+                Tc[:, j * G:(j + 1) * G, :j * G] = row
+            for km in range(nmerge):
+                k0 = km * m
+                mc = min(m, nwin - k0)
+                base = 1 + s0 + k0 * b
+                hseg = H + (mc - 1) * b
+                w = mc * G
+                seg = Epad[base:base + hseg]
+                Vck = Vc[km, :hseg, :w]
+                W = Tc[km, :w, :w] @ (Vck.mH @ seg)
+                seg -= Vck @ W
+        else:
+            # apply windows in ascending k (ordering constraint across overlaps)
+            for k in range(nwin):
+                base = 1 + s0 + k * b
+                seg = Epad[base:base + H]
+                W = T[k] @ (Vg[k].mH @ seg)
+                seg -= Vg[k] @ W
     E.copy_(Epad[:n])

@@ -61,3 +61,27 @@ def test_band2tridiag_backtransform(dtype, n, b):
     assert res < 1e-12 * n * scale, f"res={res}"
     orth = (E.mH @ E - torch.eye(n, dtype=dtype)).abs().max().item()
     assert orth < 1e-12 * n, f"orth={orth}"
+
+
+@pytest.mark.parametrize("dtype", [torch.float64, torch.complex128])
+def test_bt_window_merge_equivalence(dtype):
+    """Merged block-WY application (window_merge m) is algebraically exact:
+    identical output to the unmerged ascending-k chain."""
+    from dlaf_amd.config import get_tune_parameters
+    n, b, nE = 230, 8, 13
+    mat, _ = _band_matrix(n, b, dtype, seed=45)
+    tri = band_to_tridiagonal(UpLo.Lower, b, mat)
+    tp = get_tune_parameters()
+    outs = {}
+    for m in (1, 3, 4):
+        torch.manual_seed(7)
+        E = torch.randn(n, nE, dtype=torch.float64).to(dtype)
+        old = tp.bt_band_to_tridiag_window_merge
+        tp.bt_band_to_tridiag_window_merge = m
+        try:
+            bt_band_to_tridiagonal(E, tri, group_size=16)
+        finally:
+            tp.bt_band_to_tridiag_window_merge = old
+        outs[m] = E
+    assert torch.allclose(outs[1], outs[4], atol=1e-12, rtol=1e-12)
+    assert torch.allclose(outs[1], outs[3], atol=1e-12, rtol=1e-12)
