@@ -27,8 +27,12 @@ class TuneParameters:
     eigensolver_min_band: int = 100
     band_to_tridiag_1d_block_size_base: int = 8192
     bt_band_to_tridiag_hh_apply_group_size: int = 128  # measured best at n=20k
-    bt_band_to_tridiag_window_merge: int = 4   # consecutive WY windows merged
-                                               # into one block-WY apply
+    bt_band_to_tridiag_window_merge: int = 1   # consecutive WY windows merged
+                                               # into one block-WY apply;
+                                               # measured neutral at m=4 and
+                                               # worse at m=8 on n=20000 (the
+                                               # staircase zero-padding grows
+                                               # the GEMM flops ~2x at m=4)
     tridiag_rank1_num_threads: int = 0          # 0 = auto
     red2band_panel_num_threads: int = 0
     tfactor_num_streams: int = 4
