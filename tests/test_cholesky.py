@@ -94,3 +94,11 @@ def test_cholesky_complex64_local():
     got = torch.tril(mat.to_global()).to(torch.complex128)
     want = torch.linalg.cholesky(a_ref)
     assert (got - want).abs().max().item() < 1e-3
+
+
+@pytest.mark.timeout(600)
+def test_cholesky_dist_cpu_2x4():
+    """The driver's 8-GPU SCALE grid shape (2x4), 8 gloo ranks."""
+    errs = run_distributed(_dist_cholesky_worker, 8, args=(2, 4, 40, 8, "float64"))
+    for e in errs:
+        assert e < 1e-11 * 40, f"err={e}"
