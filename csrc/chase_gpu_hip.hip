@@ -1,0 +1,319 @@
+// GPU bulge chase: persistent single-wave workgroups, one sweep per WG in
+// round-robin, wavefront-pipelined through agent-scope flags.
+//
+// Mirrors csrc/band_chase.cpp's run_sweep unit-for-unit (same windows, same
+// publish protocol: unit u of sweep s needs done[s-1] >= u+3), so the sweep
+// windows stay disjoint by construction. Band layout identical: row-major
+// [size, ld] with band[j][d] = A[j+d, j], ld >= 2b.
+//
+// Reference counterpart: the CPU sweep workers of
+// eigensolver/band_to_tridiag/mc.h:666-693 — the reference keeps this stage
+// on the host; here the band (n x 2b), the reflector store and the flags all
+// stay in HBM and ~min(W, n/3b) sweeps run concurrently, one wavefront each.
+//
+// Cross-WG visibility uses the MI355X idiom (per-XCD L2s are not coherent;
+// see MI355X_MICROARCH.md "Workgroup dispatch, XCD placement &
+// inter-workgroup visibility"):
+//   producer: plain stores -> __syncthreads -> lane0 fence(release, agent)
+//             -> s_waitcnt vmcnt(0) -> relaxed agent flag store
+//   consumer: relaxed agent poll -> fence(acquire, agent) -> __syncthreads
+// Every spin is BOUNDED: on exhaustion the WG raises a global abort flag and
+// all WGs drain, so a scheduling pathology surfaces as a host error, never a
+// hung GPU.
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+#include "cplx.h"
+
+namespace {
+
+constexpr int BMAX = 64;      // wave size == max band width handled per lane
+constexpr int SPIN_LIMIT = 1 << 23;
+
+template <typename T>
+__device__ inline typename ScalarTraits<T>::real_t sabs2(T v) {
+  if constexpr (sizeof(T) == sizeof(typename ScalarTraits<T>::real_t))
+    return v * v;
+  else
+    return v.abs2();
+}
+
+template <typename R>
+__device__ inline R wave_sum(R x) {
+  for (int off = 32; off > 0; off >>= 1) x += __shfl_down(x, off, 64);
+  return __shfl(x, 0, 64);
+}
+
+template <typename T>
+__device__ inline T wave_bcast0(T x) {
+  if constexpr (sizeof(T) == sizeof(typename ScalarTraits<T>::real_t))
+    return __shfl(x, 0, 64);
+  else
+    return T{__shfl(x.re, 0, 64), __shfl(x.im, 0, 64)};
+}
+
+template <typename T>
+__device__ inline T wave_sum_t(T x) {
+  if constexpr (sizeof(T) == sizeof(typename ScalarTraits<T>::real_t)) {
+    return wave_sum(x);
+  } else {
+    return T{wave_sum(x.re), wave_sum(x.im)};
+  }
+}
+
+template <typename T>
+__device__ inline bool is_zero(T v) {
+  if constexpr (sizeof(T) == sizeof(typename ScalarTraits<T>::real_t))
+    return v == T(0);
+  else
+    return v.re == 0 && v.im == 0;
+}
+
+// One-wave LAPACK-style larfg over lane-held x (lane p owns x_p, p < n).
+// Returns tau (broadcast); x scaled in place (x_0 <- beta).
+template <typename T>
+__device__ T wave_reflector(int n, T& x, int lane) {
+  using TR = ScalarTraits<T>;
+  using R = typename TR::real_t;
+  if (n <= 1) return TR::zero();
+  R xn2 = wave_sum((lane >= 1 && lane < n) ? sabs2(x) : R(0));
+  T alpha = wave_bcast0(x);
+  R a_re = TR::real(alpha);
+  R a_im;
+  if constexpr (sizeof(T) == sizeof(R))
+    a_im = 0;
+  else
+    a_im = alpha.im;
+  if (xn2 == 0 && a_im == 0) return TR::zero();
+  R beta = -sqrt(a_re * a_re + a_im * a_im + xn2);
+  if (a_re < 0) beta = -beta;
+  T tau;
+  if constexpr (sizeof(T) == sizeof(R))
+    tau = (beta - a_re) / beta;
+  else
+    tau = T((beta - a_re) / beta, -a_im / beta);
+  T scale = TR::recip(alpha - TR::from_real(beta));
+  if (lane >= 1 && lane < n) x = x * scale;
+  if (lane == 0) x = TR::from_real(beta);
+  return tau;
+}
+
+// The band window ops. A = a + j*ld (element pointer), blk(p,q) at
+// A[(d0 + p - q) + q*ld]. All lanes participate; lane p owns row p (or
+// column q for the left apply). v/w broadcast through LDS.
+template <typename T>
+__device__ void unit_two_sided(int nn, T tau, const T* vl, T* A, int64_t ld,
+                               int lane, T* wl) {
+  using TR = ScalarTraits<T>;
+  if (is_zero(tau) || nn <= 0) return;
+  T w = TR::zero();
+  const int p = lane;
+  if (p < nn) {
+    for (int q = 0; q < nn; ++q) {
+      if (q < p)
+        w += A[(p - q) + (int64_t)q * ld] * vl[q];
+      else if (q == p)
+        w += TR::from_real(TR::real(A[(int64_t)p * ld])) * vl[p];
+      else
+        w += TR::conj(A[(q - p) + (int64_t)p * ld]) * vl[q];
+    }
+  }
+  T vhu = wave_sum_t((p < nn) ? TR::conj(vl[p]) * w : TR::zero());
+  using R = typename TR::real_t;
+  R t2 = sabs2(tau) / R(2);
+  T half = vhu * t2;  // |tau|^2/2 * (v^H u)
+  T wp = tau * w - half * ((p < nn) ? vl[p] : TR::zero());
+  __syncthreads();
+  wl[lane] = (p < nn) ? wp : TR::zero();
+  __syncthreads();
+  if (p < nn) {
+    for (int q = 0; q <= p; ++q) {
+      T* e = &A[(p - q) + (int64_t)q * ld];
+      *e -= vl[p] * TR::conj(wl[q]) + wl[p] * TR::conj(vl[q]);
+    }
+  }
+}
+
+template <typename T>
+__device__ void unit_apply_right(int m, int nn, T tau, const T* vl, T* A,
+                                 int64_t ld, int d0, int lane) {
+  using TR = ScalarTraits<T>;
+  if (is_zero(tau) || m <= 0 || nn <= 0) return;
+  const int p = lane;
+  if (p < m) {
+    T s = TR::zero();
+    for (int q = 0; q < nn; ++q) s += A[(d0 + p - q) + (int64_t)q * ld] * vl[q];
+    s = s * tau;
+    for (int q = 0; q < nn; ++q)
+      A[(d0 + p - q) + (int64_t)q * ld] -= s * TR::conj(vl[q]);
+  }
+}
+
+template <typename T>
+__device__ void unit_apply_left(int m, int nn, T tau, const T* vl, T* A,
+                                int64_t ld, int lane) {
+  using TR = ScalarTraits<T>;
+  if (is_zero(tau) || m <= 0 || nn <= 0) return;
+  const int q = lane;
+  if (q < nn) {
+    T s = TR::zero();
+    for (int p = 0; p < m; ++p) s += TR::conj(vl[p]) * A[(p - q) + (int64_t)q * ld];
+    s = s * TR::conj(tau);
+    for (int p = 0; p < m; ++p) A[(p - q) + (int64_t)q * ld] -= s * vl[p];
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(64) void chase_gpu_k(
+    T* a, int64_t ld, int64_t size, int64_t b, T* vstore,
+    const int64_t* offsets, int32_t* done, int32_t* abortf, int64_t nsweeps) {
+  using TR = ScalarTraits<T>;
+  const int lane = threadIdx.x;
+  const int64_t vstride = b + 1;
+  __shared__ T vl[BMAX], wl[BMAX];
+  __shared__ int ok_s;
+
+  // consumer side: bounded relaxed poll, then one agent acquire
+  auto wait_flag = [&](int64_t s, int32_t need) -> bool {
+    if (lane == 0) {
+      int ok = 1;
+      if (s > 0) {
+        int spins = 0;
+        while (__hip_atomic_load(&done[s - 1], __ATOMIC_RELAXED,
+                                 __HIP_MEMORY_SCOPE_AGENT) < need) {
+          if (++spins > SPIN_LIMIT) {
+            __hip_atomic_store(abortf, 1, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+            ok = 0;
+            break;
+          }
+          if ((spins & 1023) == 0 &&
+              __hip_atomic_load(abortf, __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_AGENT) != 0) {
+            ok = 0;
+            break;
+          }
+          __builtin_amdgcn_s_sleep(8);
+        }
+      }
+      if (ok) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+      ok_s = ok;
+    }
+    __syncthreads();
+    return ok_s != 0;
+  };
+  // producer side: release everything this WG stored, then publish the flag
+  auto publish = [&](int64_t s, int32_t val) {
+    __syncthreads();
+    if (lane == 0) {
+      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __hip_atomic_store(&done[s], val, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+    }
+    __syncthreads();
+  };
+
+  for (int64_t s = blockIdx.x; s < nsweeps; s += gridDim.x) {
+    // ---- unit 0: initial reflector of column s
+    if (!wait_flag(s, 3)) return;
+    const int64_t n0l = size - s - 1;
+    const int n0 = (int)(n0l < b ? n0l : b);
+    T x = (lane < n0) ? a[1 + lane + s * ld] : TR::zero();
+    T tau = wave_reflector(n0, x, lane);
+    {
+      T* slot = vstore + offsets[s] * vstride;
+      if (lane == 0) slot[0] = tau;
+      T vv = (lane == 0) ? TR::from_real(1)
+                         : ((lane < n0) ? x : TR::zero());
+      if (lane < b) slot[1 + lane] = (lane < n0) ? vv : TR::zero();
+      __syncthreads();
+      vl[lane] = vv;
+      __syncthreads();
+      // band writeback: position 0 = beta, tail zeroed
+      if (lane == 0 && n0 > 0) a[1 + s * ld] = x;
+      if (lane >= 1 && lane < n0) a[1 + lane + s * ld] = TR::zero();
+    }
+    int32_t step = 0;
+    while (true) {
+      if (!wait_flag(s, (int32_t)(step + 1) + 3)) return;
+      const int64_t j = 1 + s + (int64_t)step * b;
+      const int64_t nnl = size - j, ml = size - b - j;
+      const int nn = (int)(nnl < b ? nnl : b);
+      const int m = (int)(ml < b ? ml : b);
+      unit_two_sided(nn, tau, vl, a + j * ld, ld, lane, wl);
+      if (m > 0) unit_apply_right(m, nn, tau, vl, a + j * ld, ld, nn, lane);
+      if (m <= 1) break;
+      T x2 = (lane < m) ? a[nn + lane + j * ld] : TR::zero();
+      tau = wave_reflector(m, x2, lane);
+      ++step;
+      {
+        T* slot = vstore + (offsets[s] + step) * vstride;
+        if (lane == 0) slot[0] = tau;
+        T vv = (lane == 0) ? TR::from_real(1)
+                           : ((lane < m) ? x2 : TR::zero());
+        if (lane < b) slot[1 + lane] = (lane < m) ? vv : TR::zero();
+        __syncthreads();
+        vl[lane] = vv;
+        __syncthreads();
+        if (lane == 0 && m > 0) a[nn + j * ld] = x2;
+        if (lane >= 1 && lane < m) a[nn + lane + j * ld] = TR::zero();
+      }
+      unit_apply_left(m, nn - 1, tau, vl, a + (nn - 1) + (j + 1) * ld, ld, lane);
+      publish(s, step);
+    }
+    publish(s, INT32_MAX);
+  }
+}
+
+template <typename T>
+void launch_chase(T* a, int64_t ld, int64_t size, int64_t b, T* vstore,
+                  const int64_t* offsets, int32_t* done, int32_t* abortf,
+                  hipStream_t stream) {
+  int64_t nsweeps = size - 2;
+  if (nsweeps <= 0) return;
+  int per_cu = 0;
+  (void)hipOccupancyMaxActiveBlocksPerMultiprocessor(
+      &per_cu, reinterpret_cast<const void*>(chase_gpu_k<T>), 64, 0);
+  if (per_cu < 1) per_cu = 1;
+  hipDeviceProp_t prop;
+  (void)hipGetDeviceProperties(&prop, 0);
+  int64_t W = (int64_t)per_cu * prop.multiProcessorCount;
+  if (W > nsweeps) W = nsweeps;
+  if (W > 1024) W = 1024;  // residency is the correctness envelope; stay low
+ hipLaunchKernelGGL(( chase_gpu_k<T>), dim3(dim3((uint32_t)W)), dim3(dim3(64)), 0, stream, 
+      a, ld, size, b, vstore, offsets, done, abortf, nsweeps);
+}
+
+}  // namespace
+
+extern "C" {
+
+void chase_gpu_f64(double* a, int64_t ld, int64_t size, int64_t b,
+                   double* vstore, const int64_t* offsets, int32_t* done,
+                   int32_t* abortf, hipStream_t stream) {
+  launch_chase(a, ld, size, b, vstore, offsets, done, abortf, stream);
+}
+void chase_gpu_f32(float* a, int64_t ld, int64_t size, int64_t b, float* vstore,
+                   const int64_t* offsets, int32_t* done, int32_t* abortf,
+                   hipStream_t stream) {
+  launch_chase(a, ld, size, b, vstore, offsets, done, abortf, stream);
+}
+void chase_gpu_c128(double* a, int64_t ld, int64_t size, int64_t b,
+                    double* vstore, const int64_t* offsets, int32_t* done,
+                    int32_t* abortf, hipStream_t stream) {
+  launch_chase(reinterpret_cast<cplx<double>*>(a), ld, size, b,
+               reinterpret_cast<cplx<double>*>(vstore), offsets, done, abortf,
+               stream);
+}
+void chase_gpu_c64(float* a, int64_t ld, int64_t size, int64_t b, float* vstore,
+                   const int64_t* offsets, int32_t* done, int32_t* abortf,
+                   hipStream_t stream) {
+  launch_chase(reinterpret_cast<cplx<float>*>(a), ld, size, b,
+               reinterpret_cast<cplx<float>*>(vstore), offsets, done, abortf,
+               stream);
+}
+
+}  // extern "C"

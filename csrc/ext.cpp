@@ -15,6 +15,18 @@
 
 #include "kernels.h"
 
+// csrc/chase_gpu.hip
+extern "C" {
+void chase_gpu_f64(double*, int64_t, int64_t, int64_t, double*, const int64_t*,
+                   int32_t*, int32_t*, hipStream_t);
+void chase_gpu_f32(float*, int64_t, int64_t, int64_t, float*, const int64_t*,
+                   int32_t*, int32_t*, hipStream_t);
+void chase_gpu_c128(double*, int64_t, int64_t, int64_t, double*, const int64_t*,
+                    int32_t*, int32_t*, hipStream_t);
+void chase_gpu_c64(float*, int64_t, int64_t, int64_t, float*, const int64_t*,
+                   int32_t*, int32_t*, hipStream_t);
+}
+
 // csrc/rocblas_batch.cpp
 void lib_gemm_batched(torch::Tensor dt, torch::Tensor ptrC, torch::Tensor ptrA,
                       torch::Tensor ptrB, int64_t M, int64_t N, int64_t K,
@@ -349,6 +361,40 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("factor_invert_block", &factor_invert_block,
         "fused single-workgroup [factor+]invert of a diagonal block");
   m.def("trtri_lower", &trtri_lower, "lower-triangular block inverse");
+  m.def("band_chase_gpu", [](torch::Tensor band, int64_t b, torch::Tensor vstore,
+                             torch::Tensor offsets, torch::Tensor done,
+                             torch::Tensor abortf) {
+    TORCH_CHECK(band.is_cuda() && vstore.is_cuda() && offsets.is_cuda() &&
+                done.is_cuda() && abortf.is_cuda(), "device tensors required");
+    TORCH_CHECK(b <= 64, "band_chase_gpu handles b <= 64");
+    TORCH_CHECK(done.scalar_type() == at::kInt && abortf.scalar_type() == at::kInt);
+    const int64_t size = band.size(0), ld = band.size(1);
+    auto s = (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+    switch (band.scalar_type()) {
+      case at::kDouble:
+        chase_gpu_f64(band.data_ptr<double>(), ld, size, b,
+                      vstore.data_ptr<double>(), offsets.data_ptr<int64_t>(),
+                      done.data_ptr<int32_t>(), abortf.data_ptr<int32_t>(), s);
+        break;
+      case at::kFloat:
+        chase_gpu_f32(band.data_ptr<float>(), ld, size, b,
+                      vstore.data_ptr<float>(), offsets.data_ptr<int64_t>(),
+                      done.data_ptr<int32_t>(), abortf.data_ptr<int32_t>(), s);
+        break;
+      case at::kComplexDouble:
+        chase_gpu_c128((double*)band.data_ptr(), ld, size, b,
+                       (double*)vstore.data_ptr(), offsets.data_ptr<int64_t>(),
+                       done.data_ptr<int32_t>(), abortf.data_ptr<int32_t>(), s);
+        break;
+      case at::kComplexFloat:
+        chase_gpu_c64((float*)band.data_ptr(), ld, size, b,
+                      (float*)vstore.data_ptr(), offsets.data_ptr<int64_t>(),
+                      done.data_ptr<int32_t>(), abortf.data_ptr<int32_t>(), s);
+        break;
+      default:
+        TORCH_CHECK(false, "unsupported dtype");
+    }
+  }, "GPU bulge chase (persistent wavefront workgroups)");
   m.def("lib_gemm_batched", &lib_gemm_batched,
         "rocBLAS pointer-array batched GEMM (uniform tile shape)");
   m.def("potrf_tile", &potrf_tile,

@@ -17,6 +17,8 @@ from __future__ import annotations
 from dataclasses import dataclass
 from typing import Optional
 
+import os
+
 import torch
 
 from ..types import UpLo
@@ -49,6 +51,72 @@ def _slot_counts(n: int, b: int) -> torch.Tensor:
     return counts
 
 
+
+def chase_band(store: torch.Tensor, b: int) -> TridiagResult:
+    """Run the bulge chase on a compact band ``store`` [n, 2b] (consumed in
+    place) and package the TridiagResult.
+
+    GPU path (``csrc/chase_gpu.hip``): persistent wavefront workgroups chase
+    ~n/3b sweeps concurrently with the band, reflectors and flags resident in
+    HBM — the reference keeps this stage on the host
+    (``band_to_tridiag/mc.h:666-693``); here the host loop disappears
+    entirely. Deterministic (fixed wave-reduction order), so rank-replicated
+    distributed use stays lockstep-identical. Bounded spins: a scheduling
+    pathology sets an abort flag and the band is re-chased on the CPU from a
+    saved copy instead of hanging. ``DLAF_GPU_CHASE=0`` forces the CPU path.
+    """
+    from ..ops._ext import get_ext
+    n = store.shape[0]
+    is_cplx = store.is_complex()
+    counts = _slot_counts(n, b)
+    offsets = torch.zeros(n, dtype=torch.int64)
+    if n > 1:
+        offsets[1:] = torch.cumsum(counts, 0)[:-1]
+    total = int(counts.sum().item())
+    vstore = None
+    if (n > 2 and store.is_cuda and b <= 64
+            and os.environ.get("DLAF_GPU_CHASE", "1") != "0"):
+        dev = store.device
+        vstore = torch.zeros((max(total, 1), b + 1), dtype=store.dtype, device=dev)
+        done = torch.zeros(n, dtype=torch.int32, device=dev)
+        abortf = torch.zeros(1, dtype=torch.int32, device=dev)
+        backup = store.clone()
+        get_ext().band_chase_gpu(store, b, vstore, offsets.to(dev), done, abortf)
+        torch.cuda.synchronize()
+        if int(abortf.item()) != 0:
+            import warnings
+            warnings.warn("GPU band chase aborted (bounded spin exhausted); "
+                          "falling back to the CPU chase")
+            store.copy_(backup)
+            vstore = None
+        del backup, done, abortf
+    if vstore is None:
+        store_h = store.cpu() if store.is_cuda else store
+        vstore = torch.zeros((max(total, 1), b + 1), dtype=store_h.dtype)
+        if n > 2:
+            get_ext().band_chase(store_h, b, vstore, offsets)
+        store = store_h
+    dvec = store[:, 0]
+    evec = store[: n - 1, 1] if n > 1 else store[:0, 1]
+    phases = None
+    if is_cplx:
+        e_abs = evec.abs()
+        safe = torch.where(e_abs > 0, e_abs, torch.ones_like(e_abs))
+        factor = torch.where(e_abs > 0, evec / safe, torch.ones_like(evec))
+        ph = torch.ones(n, dtype=store.dtype, device=store.device)
+        if n > 1:
+            ph[1:] = torch.cumprod(factor, 0)
+            ph = ph / ph.abs()          # keep strictly unit modulus
+        phases = ph
+        d_real = dvec.real.cpu().clone()
+        e_real = e_abs.to(d_real.dtype).cpu()
+    else:
+        d_real = dvec.cpu().clone()
+        e_real = evec.cpu().clone()
+    return TridiagResult(d=d_real, e=e_real, band=b, n=n, vstore=vstore,
+                         offsets=offsets, nslots=counts, phases=phases)
+
+
 def band_to_tridiagonal(uplo: UpLo, band: int, mat: Matrix) -> TridiagResult:
     """Reduce the band part of ``mat`` (lower, bandwidth ``band``) to real
     tridiagonal form; returns the tridiagonal and the chase reflectors."""
@@ -66,36 +134,8 @@ def band_to_tridiagonal(uplo: UpLo, band: int, mat: Matrix) -> TridiagResult:
     store_dev = torch.zeros((n, ld), dtype=A.dtype, device=A.device)
     for dd in range(min(b, n - 1) + 1):
         store_dev[: n - dd, dd] = torch.diagonal(A, -dd)
-    is_cplx = A.is_complex()
-    store = store_dev.cpu()
-    del A, store_dev
-
-    counts = _slot_counts(n, b)
-    offsets = torch.zeros(n, dtype=torch.int64)
-    if n > 1:
-        offsets[1:] = torch.cumsum(counts, 0)[:-1]
-    total = int(counts.sum().item())
-    vstore = torch.zeros((max(total, 1), b + 1), dtype=store.dtype)
-    if n > 2:
-        get_ext().band_chase(store, b, vstore, offsets)
-
-    dvec = store[:, 0]
-    evec = store[: n - 1, 1] if n > 1 else store[:0, 1]
-    phases = None
-    if is_cplx:
-        ph = torch.ones(n, dtype=store.dtype)
-        e_abs = evec.abs()
-        for j in range(n - 1):
-            aj = e_abs[j]
-            ph[j + 1] = ph[j] * (evec[j] / aj if aj > 0 else 1.0)
-        phases = ph
-        d_real = dvec.real.clone()
-        e_real = e_abs.to(d_real.dtype)
-    else:
-        d_real = dvec.clone()
-        e_real = evec.clone()
-    return TridiagResult(d=d_real, e=e_real, band=b, n=n, vstore=vstore,
-                         offsets=offsets, nslots=counts, phases=phases)
+    del A
+    return chase_band(store_dev, b)
 
 
 def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,

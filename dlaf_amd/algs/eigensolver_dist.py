@@ -172,33 +172,10 @@ def hermitian_eigensolver_dist(uplo: UpLo, mat: Matrix, grid: CommGrid,
 
     refl = _red2band_dist(A, band, cm, group)
 
-    # band -> tridiagonal: replicated CPU chase (identical on every rank)
-    store = _extract_band_dist(A, band, cm, group).cpu()
-    from .band2tridiag import _slot_counts
-    counts = _slot_counts(n, band)
-    offsets = torch.zeros(n, dtype=torch.int64)
-    if n > 1:
-        offsets[1:] = torch.cumsum(counts, 0)[:-1]
-    total = int(counts.sum().item())
-    vstore = torch.zeros((max(total, 1), band + 1), dtype=store.dtype)
-    if n > 2:
-        from ..ops._ext import get_ext
-        get_ext().band_chase(store, band, vstore, offsets)
-    dvec = store[:, 0]
-    evec = store[: n - 1, 1] if n > 1 else store[:0, 1]
-    phases = None
-    if A.is_complex():
-        ph = torch.ones(n, dtype=A.dtype)
-        e_abs = evec.abs()
-        for j in range(n - 1):
-            aj = e_abs[j]
-            ph[j + 1] = ph[j] * (evec[j] / aj if aj > 0 else 1.0)
-        phases = ph
-        d_real, e_real = dvec.real.clone(), e_abs.to(dvec.real.dtype)
-    else:
-        d_real, e_real = dvec.clone(), evec.clone()
-    tri = TridiagResult(d=d_real, e=e_real, band=band, n=n, vstore=vstore,
-                        offsets=offsets, nslots=counts, phases=phases)
+    # band -> tridiagonal: replicated chase (deterministic, identical on
+    # every rank; GPU kernel when available — see band2tridiag.chase_band)
+    from .band2tridiag import chase_band
+    tri = chase_band(_extract_band_dist(A, band, cm, group), band)
 
     # tridiag D&C: replicated (scalar stages), GEMM merges on the local GPU
     w, E_real = tridiagonal_eigensolver(tri.d, tri.e, device=dev)

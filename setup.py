@@ -24,6 +24,7 @@ setup(
                 "csrc/panel_qr.hip",
                 "csrc/secular.hip",
                 "csrc/rocblas_batch.cpp",
+                "csrc/chase_gpu.hip",
             ],
             libraries=["rocblas"],
             extra_compile_args={
