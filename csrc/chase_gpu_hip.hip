@@ -195,7 +195,9 @@ __global__ __launch_bounds__(64) void chase_gpu_k(
             ok = 0;
             break;
           }
-          __builtin_amdgcn_s_sleep(8);
+          // far behind -> long sleeps: idle pollers otherwise saturate the
+          // fabric and starve the working wavefront
+          __builtin_amdgcn_s_sleep(spins > 64 ? 127 : 16);
         }
       }
       if (ok) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
@@ -280,9 +282,13 @@ void launch_chase(T* a, int64_t ld, int64_t size, int64_t b, T* vstore,
   if (per_cu < 1) per_cu = 1;
   hipDeviceProp_t prop;
   (void)hipGetDeviceProperties(&prop, 0);
-  int64_t W = (int64_t)per_cu * prop.multiProcessorCount;
+  int64_t W = (int64_t)per_cu * prop.multiProcessorCount;  // residency bound
+  // concurrency bound: at most ~steps(sweep0)/3 sweeps can be in flight —
+  // any extra WG is a pure poller burning fabric bandwidth
+  int64_t active = (size + 3 * b - 1) / (3 * b) + 16;
+  if (W > active) W = active;
   if (W > nsweeps) W = nsweeps;
-  if (W > 1024) W = 1024;  // residency is the correctness envelope; stay low
+  if (W < 1) W = 1;
  hipLaunchKernelGGL(( chase_gpu_k<T>), dim3(dim3((uint32_t)W)), dim3(dim3(64)), 0, stream, 
       a, ld, size, b, vstore, offsets, done, abortf, nsweeps);
 }
