@@ -197,7 +197,10 @@ __global__ __launch_bounds__(64) void chase_gpu_k(
           }
           // far behind -> long sleeps: idle pollers otherwise saturate the
           // fabric and starve the working wavefront
-          __builtin_amdgcn_s_sleep(spins > 64 ? 127 : 16);
+          if (spins > 64)
+            __builtin_amdgcn_s_sleep(127);
+          else
+            __builtin_amdgcn_s_sleep(16);
         }
       }
       if (ok) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
