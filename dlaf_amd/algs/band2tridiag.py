@@ -56,14 +56,16 @@ def chase_band(store: torch.Tensor, b: int) -> TridiagResult:
     """Run the bulge chase on a compact band ``store`` [n, 2b] (consumed in
     place) and package the TridiagResult.
 
-    GPU path (``csrc/chase_gpu.hip``): persistent wavefront workgroups chase
-    ~n/3b sweeps concurrently with the band, reflectors and flags resident in
-    HBM — the reference keeps this stage on the host
-    (``band_to_tridiag/mc.h:666-693``); here the host loop disappears
-    entirely. Deterministic (fixed wave-reduction order), so rank-replicated
-    distributed use stays lockstep-identical. Bounded spins: a scheduling
-    pathology sets an abort flag and the band is re-chased on the CPU from a
-    saved copy instead of hanging. ``DLAF_GPU_CHASE=0`` forces the CPU path.
+    GPU path (``csrc/chase_gpu.hip``, opt-in via ``DLAF_GPU_CHASE=1``):
+    persistent wavefront workgroups chase ~n/3b sweeps concurrently with the
+    band, reflectors and flags resident in HBM — the reference keeps this
+    stage on the host (``band_to_tridiag/mc.h:666-693``). Deterministic
+    (fixed wave-reduction order), so rank-replicated distributed use stays
+    lockstep-identical; bounded spins surface scheduling pathologies as a
+    CPU-fallback warning, never a hang. Currently ~1.7x SLOWER than the
+    16-thread CPU wavefront at n=20000 (the per-unit agent-scope
+    release/acquire handoff dominates the pipeline critical path — see
+    docs/DESIGN.md), so the CPU chase stays the default.
     """
     from ..ops._ext import get_ext
     n = store.shape[0]
@@ -75,7 +77,7 @@ def chase_band(store: torch.Tensor, b: int) -> TridiagResult:
     total = int(counts.sum().item())
     vstore = None
     if (n > 2 and store.is_cuda and b <= 64
-            and os.environ.get("DLAF_GPU_CHASE", "1") != "0"):
+            and os.environ.get("DLAF_GPU_CHASE", "0") == "1"):
         dev = store.device
         vstore = torch.zeros((max(total, 1), b + 1), dtype=store.dtype, device=dev)
         done = torch.zeros(n, dtype=torch.int32, device=dev)

@@ -288,7 +288,9 @@ def test_eigensolver_gpu_float32():
 def test_gpu_band_chase_matches_cpu(dtype):
     """GPU wavefront chase vs the CPU chase on the same band: d/e and
     reflector store agree to rounding."""
+    import os
     from dlaf_amd.algs.band2tridiag import chase_band
+    os.environ["DLAF_GPU_CHASE"] = "1"
     torch.manual_seed(11)
     n, b = 1500, 64
     ld = 2 * b
@@ -301,7 +303,10 @@ def test_gpu_band_chase_matches_cpu(dtype):
             col = col + 1j * torch.randn(n - d, dtype=torch.float64)
         store[: n - d, d] = col.to(dtype)
     cpu = chase_band(store.clone(), b)
-    gpu = chase_band(store.clone().cuda(), b)
+    try:
+        gpu = chase_band(store.clone().cuda(), b)
+    finally:
+        os.environ.pop("DLAF_GPU_CHASE", None)
     assert gpu.vstore.is_cuda, "GPU chase did not run (fell back to CPU)"
     de = (gpu.d.cpu() - cpu.d).abs().max().item()
     ee = (gpu.e.cpu() - cpu.e).abs().max().item()
