@@ -10,8 +10,12 @@ from dlaf_amd.types import total_ops
 
 def setup(ctx):
     a = random_spd(ctx)
+    st = {}
+    if ctx.opts.check_result != "none":
+        st["a0"] = a.clone()
     cholesky_factorization(UpLo.Lower, a, ctx.comm_grid)
-    return {"a": a}
+    st["a"] = a
+    return st
 
 
 def run(ctx, st):
@@ -23,5 +27,17 @@ def flops(ctx):
     return total_ops(ctx.dtype, n**3 / 3, n**3 / 3)
 
 
+def check(ctx, st, _):
+    """max |A0 X - I| (X = the computed inverse, lower stored)."""
+    import torch
+    x = st["a"].to_global()
+    x = torch.tril(x) + torch.tril(x, -1).mH
+    a0 = st["a0"].to_global()
+    a0 = torch.tril(a0) + torch.tril(a0, -1).mH
+    n = a0.shape[0]
+    eye = torch.eye(n, dtype=a0.dtype, device=a0.device)
+    return (a0 @ x - eye).abs().max().item()
+
+
 if __name__ == "__main__":
-    run_miniapp("miniapp_inverse_from_cholesky_factor", setup, run, flops)
+    run_miniapp("miniapp_inverse_from_cholesky_factor", setup, run, flops, check)

@@ -9,7 +9,11 @@ from dlaf_amd.types import total_ops
 
 
 def setup(ctx):
-    return {"a": random_spd(ctx)}
+    a = random_spd(ctx)
+    st = {"a": a}
+    if ctx.opts.check_result != "none":
+        st["a0"] = a.clone()
+    return st
 
 
 def run(ctx, st):
@@ -21,5 +25,15 @@ def flops(ctx):
     return total_ops(ctx.dtype, n**3 / 6, n**3 / 6)
 
 
+def check(ctx, st, _):
+    """max |tril(X) tril(L0) - I|."""
+    import torch
+    X = torch.tril(st["a"].to_global())
+    L0 = torch.tril(st["a0"].to_global())
+    n = X.shape[0]
+    eye = torch.eye(n, dtype=X.dtype, device=X.device)
+    return (X @ L0 - eye).abs().max().item()
+
+
 if __name__ == "__main__":
-    run_miniapp("miniapp_triangular_inverse", setup, run, flops)
+    run_miniapp("miniapp_triangular_inverse", setup, run, flops, check)

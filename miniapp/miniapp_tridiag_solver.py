@@ -19,5 +19,15 @@ def run(ctx, st):
     return tridiagonal_eigensolver(st["d"], st["e"], device=ctx.device)
 
 
+def check(ctx, st, result):
+    """max |T E - E diag(w)| / max(1, |w|_max)."""
+    w, E = result
+    n = st["d"].shape[0]
+    T = torch.diag(st["d"]) + torch.diag(st["e"], 1) + torch.diag(st["e"], -1)
+    T = T.to(E.device)
+    r = (T @ E - E * w.to(E.dtype).to(E.device)).abs().max()
+    return (r / max(1.0, w.abs().max().item())).item()
+
+
 if __name__ == "__main__":
-    run_miniapp("miniapp_tridiag_solver", setup, run, lambda ctx: None)
+    run_miniapp("miniapp_tridiag_solver", setup, run, lambda ctx: None, check)
