@@ -55,3 +55,26 @@ def test_cholesky_dist_gpu_path_single_rank():
     want = torch.linalg.cholesky(a)
     err = (got - want).abs().max().item()
     assert err < 1e-10 * n, f"err={err}"
+
+
+@pytest.mark.timeout(300)
+def test_cholesky_trsm_complex64_gpu():
+    """complex64 coverage of the GPU factor + solve path."""
+    from dlaf_amd import Side, Op, Diag, triangular_solver
+    n, nb = 1024, 256
+    A = Matrix.create(n, n, nb, nb, dtype=torch.complex64, device="cuda")
+    mutil.set_random_hermitian_positive_definite(A, seed=9)
+    a0 = A.to_global().cpu()
+    cholesky_factorization(UpLo.Lower, A)
+    torch.cuda.synchronize()
+    L = torch.tril(A.to_global().cpu())
+    err = (L @ L.mH - a0).abs().max().item() / a0.abs().max().item()
+    assert err < 1e-4, f"chol err={err}"
+    B = Matrix.create(n, 512, nb, nb, dtype=torch.complex64, device="cuda")
+    mutil.set_random(B, seed=10)
+    b0 = B.to_global().cpu()
+    triangular_solver(Side.Left, UpLo.Lower, Op.NoTrans, Diag.NonUnit, 1.0, A, B)
+    torch.cuda.synchronize()
+    want = torch.linalg.solve(L.to(torch.complex128), b0.to(torch.complex128))
+    err = (B.to_global().cpu().to(torch.complex128) - want).abs().max().item()
+    assert err < 1e-2, f"trsm err={err}"
