@@ -12,8 +12,11 @@ from dlaf_amd.algs.eigensolver import get_band_size
 n = int(sys.argv[1]) if len(sys.argv) > 1 else 4096
 nb = int(sys.argv[2]) if len(sys.argv) > 2 else 512
 band = int(sys.argv[3]) if len(sys.argv) > 3 else get_band_size(nb)
+dtype = {"d": torch.float64, "z": torch.complex128,
+         "s": torch.float32, "c": torch.complex64}[
+             sys.argv[4] if len(sys.argv) > 4 else "d"]
 dev = "cuda" if torch.cuda.is_available() else "cpu"
-mat = Matrix.create(n, n, nb, nb, dtype=torch.float64, device=dev)
+mat = Matrix.create(n, n, nb, nb, dtype=dtype, device=dev)
 mutil.set_random_hermitian(mat, seed=1)
 a0 = None
 if n <= 4096:
@@ -29,10 +32,12 @@ refl = reduction_to_band(mat, band); sync(); stamps.append(("red2band", time.per
 tri = band_to_tridiagonal(UpLo.Lower, band, mat); sync(); stamps.append(("band2tridiag", time.perf_counter()))
 w, E_real = tridiagonal_eigensolver(tri.d, tri.e, device=mat.device); sync(); stamps.append(("tridiag_dc", time.perf_counter()))
 E = E_real.to(mat.dtype).contiguous()
+if tri.phases is not None:
+    pass  # phases applied inside bt
 bt_band_to_tridiagonal(E, tri); sync(); stamps.append(("bt_band2tridiag", time.perf_counter()))
 bt_reduction_to_band(E, mat, refl); sync(); stamps.append(("bt_red2band", time.perf_counter()))
 tot = stamps[-1][1] - stamps[0][1]
-print(f"SYEV n={n} nb={nb} band={band} dev={dev}: total {tot:.2f}s")
+print(f"HEEV[{mat.dtype}] n={n} nb={nb} band={band} dev={dev}: total {tot:.2f}s")
 for (nm, t1), (_, t0) in zip(stamps[1:], stamps[:-1]):
     print(f"  {nm:16s} {t1-t0:8.2f}s")
 if a0 is not None:
