@@ -100,11 +100,14 @@ __device__ T wave_reflector(int n, T& x, int lane) {
   return tau;
 }
 
-// The band window ops. A = a + j*ld (element pointer), blk(p,q) at
-// A[(d0 + p - q) + q*ld]. All lanes participate; lane p owns row p (or
-// column q for the left apply). v/w broadcast through LDS.
+// The band window ops over the LDS-staged window W: column j+q of the band
+// lives at W[q*S + d] (depth d, S = ld + pad). Serial per-lane global loads
+// made the first version ~120 us/unit (latency-bound); staging the <= nn x ld
+// window through LDS once per unit makes it bandwidth-bound. blk(p, q) of a
+// block anchored at depth d0 is W[q*S + d0 + p - q]. Lane p owns row p
+// (column q for the left apply); v/w broadcast through LDS.
 template <typename T>
-__device__ void unit_two_sided(int nn, T tau, const T* vl, T* A, int64_t ld,
+__device__ void unit_two_sided(int nn, T tau, const T* vl, T* W, int S,
                                int lane, T* wl) {
   using TR = ScalarTraits<T>;
   if (is_zero(tau) || nn <= 0) return;
@@ -113,11 +116,11 @@ __device__ void unit_two_sided(int nn, T tau, const T* vl, T* A, int64_t ld,
   if (p < nn) {
     for (int q = 0; q < nn; ++q) {
       if (q < p)
-        w += A[(p - q) + (int64_t)q * ld] * vl[q];
+        w += W[q * S + (p - q)] * vl[q];
       else if (q == p)
-        w += TR::from_real(TR::real(A[(int64_t)p * ld])) * vl[p];
+        w += TR::from_real(TR::real(W[p * S])) * vl[p];
       else
-        w += TR::conj(A[(q - p) + (int64_t)p * ld]) * vl[q];
+        w += TR::conj(W[p * S + (q - p)]) * vl[q];
     }
   }
   T vhu = wave_sum_t((p < nn) ? TR::conj(vl[p]) * w : TR::zero());
@@ -129,39 +132,38 @@ __device__ void unit_two_sided(int nn, T tau, const T* vl, T* A, int64_t ld,
   wl[lane] = (p < nn) ? wp : TR::zero();
   __syncthreads();
   if (p < nn) {
-    for (int q = 0; q <= p; ++q) {
-      T* e = &A[(p - q) + (int64_t)q * ld];
-      *e -= vl[p] * TR::conj(wl[q]) + wl[p] * TR::conj(vl[q]);
-    }
+    for (int q = 0; q <= p; ++q)
+      W[q * S + (p - q)] -= vl[p] * TR::conj(wl[q]) + wl[p] * TR::conj(vl[q]);
   }
 }
 
 template <typename T>
-__device__ void unit_apply_right(int m, int nn, T tau, const T* vl, T* A,
-                                 int64_t ld, int d0, int lane) {
+__device__ void unit_apply_right(int m, int nn, T tau, const T* vl, T* W,
+                                 int S, int d0, int lane) {
   using TR = ScalarTraits<T>;
   if (is_zero(tau) || m <= 0 || nn <= 0) return;
   const int p = lane;
   if (p < m) {
     T s = TR::zero();
-    for (int q = 0; q < nn; ++q) s += A[(d0 + p - q) + (int64_t)q * ld] * vl[q];
+    for (int q = 0; q < nn; ++q) s += W[q * S + (d0 + p - q)] * vl[q];
     s = s * tau;
-    for (int q = 0; q < nn; ++q)
-      A[(d0 + p - q) + (int64_t)q * ld] -= s * TR::conj(vl[q]);
+    for (int q = 0; q < nn; ++q) W[q * S + (d0 + p - q)] -= s * TR::conj(vl[q]);
   }
 }
 
+// columns of the left apply start one band column after the window anchor:
+// blk(p, q) = W[(1 + q) * S + (dL + p - q)], dL = nn - 1.
 template <typename T>
-__device__ void unit_apply_left(int m, int nn, T tau, const T* vl, T* A,
-                                int64_t ld, int lane) {
+__device__ void unit_apply_left(int m, int nn, T tau, const T* vl, T* W, int S,
+                                int dL, int lane) {
   using TR = ScalarTraits<T>;
   if (is_zero(tau) || m <= 0 || nn <= 0) return;
   const int q = lane;
   if (q < nn) {
     T s = TR::zero();
-    for (int p = 0; p < m; ++p) s += TR::conj(vl[p]) * A[(p - q) + (int64_t)q * ld];
+    for (int p = 0; p < m; ++p) s += TR::conj(vl[p]) * W[(1 + q) * S + (dL + p - q)];
     s = s * TR::conj(tau);
-    for (int p = 0; p < m; ++p) A[(p - q) + (int64_t)q * ld] -= s * vl[p];
+    for (int p = 0; p < m; ++p) W[(1 + q) * S + (dL + p - q)] -= s * vl[p];
   }
 }
 
@@ -174,6 +176,9 @@ __global__ __launch_bounds__(64) void chase_gpu_k(
   const int64_t vstride = b + 1;
   __shared__ T vl[BMAX], wl[BMAX];
   __shared__ int ok_s;
+  extern __shared__ char smem[];
+  T* W = reinterpret_cast<T*>(smem);    // [<=b cols][S], S = ld + 2
+  const int S = (int)ld + 2;
 
   // consumer side: bounded relaxed poll, then one agent acquire
   auto wait_flag = [&](int64_t s, int32_t need) -> bool {
@@ -248,13 +253,21 @@ __global__ __launch_bounds__(64) void chase_gpu_k(
       const int64_t nnl = size - j, ml = size - b - j;
       const int nn = (int)(nnl < b ? nnl : b);
       const int m = (int)(ml < b ? ml : b);
-      unit_two_sided(nn, tau, vl, a + j * ld, ld, lane, wl);
-      if (m > 0) unit_apply_right(m, nn, tau, vl, a + j * ld, ld, nn, lane);
-      if (m <= 1) break;
-      T x2 = (lane < m) ? a[nn + lane + j * ld] : TR::zero();
-      tau = wave_reflector(m, x2, lane);
-      ++step;
-      {
+      // stage the unit's whole window (cols j..j+nn-1, full depth) into LDS
+      for (int c = 0; c < nn; ++c)
+        for (int d = lane; d < (int)ld; d += 64)
+          W[c * S + d] = a[(j + c) * ld + d];
+      __syncthreads();
+      unit_two_sided(nn, tau, vl, W, S, lane, wl);
+      __syncthreads();
+      if (m > 0) unit_apply_right(m, nn, tau, vl, W, S, nn, lane);
+      bool last = (m <= 1);
+      T x2 = TR::zero();
+      if (!last) {
+        __syncthreads();
+        x2 = (lane < m) ? W[nn + lane] : TR::zero();   // col j, depth nn+p
+        tau = wave_reflector(m, x2, lane);
+        ++step;
         T* slot = vstore + (offsets[s] + step) * vstride;
         if (lane == 0) slot[0] = tau;
         T vv = (lane == 0) ? TR::from_real(1)
@@ -262,11 +275,17 @@ __global__ __launch_bounds__(64) void chase_gpu_k(
         if (lane < b) slot[1 + lane] = (lane < m) ? vv : TR::zero();
         __syncthreads();
         vl[lane] = vv;
+        if (lane == 0 && m > 0) W[nn] = x2;
+        if (lane >= 1 && lane < m) W[nn + lane] = TR::zero();
         __syncthreads();
-        if (lane == 0 && m > 0) a[nn + j * ld] = x2;
-        if (lane >= 1 && lane < m) a[nn + lane + j * ld] = TR::zero();
+        unit_apply_left(m, nn - 1, tau, vl, W, S, nn - 1, lane);
       }
-      unit_apply_left(m, nn - 1, tau, vl, a + (nn - 1) + (j + 1) * ld, ld, lane);
+      __syncthreads();
+      // write the window back, then publish
+      for (int c = 0; c < nn; ++c)
+        for (int d = lane; d < (int)ld; d += 64)
+          a[(j + c) * ld + d] = W[c * S + d];
+      if (last) break;
       publish(s, step);
     }
     publish(s, INT32_MAX);
@@ -279,9 +298,15 @@ void launch_chase(T* a, int64_t ld, int64_t size, int64_t b, T* vstore,
                   hipStream_t stream) {
   int64_t nsweeps = size - 2;
   if (nsweeps <= 0) return;
+  const size_t shbytes = (size_t)b * (2 * b + 2) * sizeof(T);
+  if (shbytes > 65536) {
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(chase_gpu_k<T>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)shbytes);
+  }
   int per_cu = 0;
   (void)hipOccupancyMaxActiveBlocksPerMultiprocessor(
-      &per_cu, reinterpret_cast<const void*>(chase_gpu_k<T>), 64, 0);
+      &per_cu, reinterpret_cast<const void*>(chase_gpu_k<T>), 64, shbytes);
   if (per_cu < 1) per_cu = 1;
   hipDeviceProp_t prop;
   (void)hipGetDeviceProperties(&prop, 0);
@@ -292,7 +317,7 @@ void launch_chase(T* a, int64_t ld, int64_t size, int64_t b, T* vstore,
   if (W > active) W = active;
   if (W > nsweeps) W = nsweeps;
   if (W < 1) W = 1;
-  chase_gpu_k<T><<<dim3((uint32_t)W), dim3(64), 0, stream>>>(
+  chase_gpu_k<T><<<dim3((uint32_t)W), dim3(64), shbytes, stream>>>(
       a, ld, size, b, vstore, offsets, done, abortf, nsweeps);
 }
 
