@@ -253,10 +253,15 @@ __global__ __launch_bounds__(64) void chase_gpu_k(
       const int64_t nnl = size - j, ml = size - b - j;
       const int nn = (int)(nnl < b ? nnl : b);
       const int m = (int)(ml < b ? ml : b);
-      // stage the unit's whole window (cols j..j+nn-1, full depth) into LDS
-      for (int c = 0; c < nn; ++c)
-        for (int d = lane; d < (int)ld; d += 64)
-          W[c * S + d] = a[(j + c) * ld + d];
+      // stage the unit's whole window (cols j..j+nn-1, full depth) into
+      // LDS. The window is CONTIGUOUS global memory (a + j*ld, nn*ld
+      // elements) — one flat strided loop streams it at line rate.
+      {
+        const T* src = a + j * ld;
+        const int tot = nn * (int)ld;
+        for (int t = lane; t < tot; t += 64)
+          W[(t / (int)ld) * S + (t % (int)ld)] = src[t];
+      }
       __syncthreads();
       unit_two_sided(nn, tau, vl, W, S, lane, wl);
       __syncthreads();
@@ -281,10 +286,13 @@ __global__ __launch_bounds__(64) void chase_gpu_k(
         unit_apply_left(m, nn - 1, tau, vl, W, S, nn - 1, lane);
       }
       __syncthreads();
-      // write the window back, then publish
-      for (int c = 0; c < nn; ++c)
-        for (int d = lane; d < (int)ld; d += 64)
-          a[(j + c) * ld + d] = W[c * S + d];
+      // write the window back (contiguous global), then publish
+      {
+        T* dst = a + j * ld;
+        const int tot = nn * (int)ld;
+        for (int t = lane; t < tot; t += 64)
+          dst[t] = W[(t / (int)ld) * S + (t % (int)ld)];
+      }
       if (last) break;
       publish(s, step);
     }
