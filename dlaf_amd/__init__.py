@@ -24,7 +24,8 @@ Public API (mirrors the reference's free-function API):
 from .types import Side, UpLo, Op, Diag  # noqa: F401
 from .core.distribution import Distribution  # noqa: F401
 from .comm.grid import CommGrid  # noqa: F401
-from .matrix.matrix import Matrix  # noqa: F401
+from .matrix.matrix import Matrix
+from .matrix.panel import Panel  # noqa: F401
 
 from .algs.cholesky import cholesky_factorization  # noqa: F401
 from .algs.triangular import triangular_solver, triangular_multiplication  # noqa: F401
