@@ -307,6 +307,8 @@ def tridiagonal_eigensolver(d: torch.Tensor, e: torch.Tensor,
     Returns (evals [n] fp, evecs [n, n]) on ``device``. Reference:
     ``eigensolver/tridiag_solver/impl.h:198-278`` (local).
     """
+    if device is not None and not isinstance(device, torch.device):
+        device = torch.device(device)
     if leaf is None:
         import os
         leaf = int(os.environ.get("DLAF_DC_LEAF", "64"))

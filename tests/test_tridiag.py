@@ -1,0 +1,69 @@
+"""Direct D&C tridiagonal eigensolver tests (reference
+``test/unit/eigensolver/test_tridiag_solver_local.cpp`` analog): deflation-
+heavy spectra, split points (zero off-diagonal), degenerate sizes, leaf
+boundary sizes."""
+import numpy as np
+import pytest
+import scipy.linalg as sl
+import torch
+
+from dlaf_amd.algs.tridiag_dc import tridiagonal_eigensolver
+
+
+def _check(d, e, atol_scale=1e-12):
+    w, E = tridiagonal_eigensolver(d.clone(), e.clone(), device="cpu")
+    n = d.shape[0]
+    T = torch.diag(d) + torch.diag(e, 1) + torch.diag(e, -1)
+    scale = max(1.0, float(d.abs().max()), float(e.abs().max()) if n > 1 else 0.0)
+    res = (T @ E - E * w).abs().max().item()
+    orth = (E.T @ E - torch.eye(n, dtype=E.dtype)).abs().max().item()
+    assert res < atol_scale * n * scale, f"res={res}"
+    assert orth < atol_scale * n, f"orth={orth}"
+    wr = np.sort(sl.eigvalsh_tridiagonal(d.numpy(), e.numpy())) if n > 1 else d.numpy()
+    assert np.abs(np.sort(w.numpy()) - wr).max() < 1e-11 * n * scale
+    assert bool((w[1:] >= w[:-1]).all())
+
+
+@pytest.mark.parametrize("n", [1, 2, 3, 63, 64, 65, 129, 300])
+def test_tridiag_random(n):
+    g = torch.Generator().manual_seed(n)
+    d = torch.randn(n, generator=g, dtype=torch.float64)
+    e = torch.randn(max(n - 1, 0), generator=g, dtype=torch.float64)
+    _check(d, e)
+
+
+def test_tridiag_split_zero_offdiag():
+    """A zero e entry decouples the problem (reference split handling)."""
+    g = torch.Generator().manual_seed(7)
+    n = 200
+    d = torch.randn(n, generator=g, dtype=torch.float64)
+    e = torch.randn(n - 1, generator=g, dtype=torch.float64)
+    e[50] = 0.0
+    e[130] = 0.0
+    _check(d, e)
+
+
+def test_tridiag_full_deflation():
+    """Identity-like input: every rank-1 component deflates."""
+    n = 150
+    d = torch.full((n,), 3.0, dtype=torch.float64)
+    e = torch.zeros(n - 1, dtype=torch.float64)
+    _check(d, e)
+
+
+def test_tridiag_heavy_deflation():
+    """Many repeated diagonal entries and tiny couplings -> deflation path."""
+    n = 180
+    d = torch.ones(n, dtype=torch.float64)
+    d[::3] = 2.0
+    e = torch.full((n - 1,), 1e-18, dtype=torch.float64)
+    e[::2] = 1e-3
+    _check(d, e)
+
+
+def test_tridiag_wilkinson():
+    """Wilkinson W21+: clustered pairs of eigenvalues."""
+    m = 10
+    d = torch.tensor([abs(i - m) for i in range(2 * m + 1)], dtype=torch.float64)
+    e = torch.ones(2 * m, dtype=torch.float64)
+    _check(d, e)
