@@ -62,6 +62,11 @@ def hermitian_eigensolver(
     if band is None:
         band = get_band_size(d.nb)
     band = max(1, min(band, max(n - 1, 1)))
+    if n <= max(band, d.nb) or d.nb % band != 0:
+        # tiny problem (single tile, or no valid band divisor): direct dense
+        # eigendecomposition — the two-stage pipeline requires nb % band == 0
+        return _eigh_direct(mat, g, eigenvalues_index_begin,
+                            eigenvalues_index_end)
     if g is not None and g.distributed:
         from .eigensolver_dist import hermitian_eigensolver_dist
         return hermitian_eigensolver_dist(
@@ -79,6 +84,26 @@ def hermitian_eigensolver(
     bt_band_to_tridiagonal(E, tri)
     bt_reduction_to_band(E, mat, refl)
 
+    nE = E.shape[1]
+    evecs = Matrix.create(n, max(nE, 1), d.mb, d.nb, dtype=mat.dtype,
+                          device=mat.device, grid=mat.grid)
+    if nE:
+        _set_cols(evecs, E)
+    return w, evecs
+
+
+def _eigh_direct(mat: Matrix, g, ib: int, ie) -> Tuple[torch.Tensor, Matrix]:
+    """Dense fallback for problems at or below one band/tile: assemble the
+    Hermitian matrix, torch.linalg.eigh (rank-replicated when distributed),
+    slice the requested spectrum."""
+    d = mat.dist
+    n = d.m
+    ie = n if ie is None else ie
+    a = mat.to_global()
+    a = torch.tril(a) + torch.tril(a, -1).mH
+    w_all, E_all = torch.linalg.eigh(a)
+    w = w_all[ib:ie].clone()
+    E = E_all[:, ib:ie].contiguous()
     nE = E.shape[1]
     evecs = Matrix.create(n, max(nE, 1), d.mb, d.nb, dtype=mat.dtype,
                           device=mat.device, grid=mat.grid)

@@ -89,6 +89,18 @@ def _red2band_dist(A: torch.Tensor, band: int, cm: _ChunkMap, group):
         taus_all.append(taus)
         V = torch.tril(P[:, :nrefl], -1) + torch.eye(m_p, nrefl, dtype=dt, device=dev)
         T = t_factor(V, taus)
+        if bw < band:
+            # capped panel: remaining band columns [j0+bw, r0) get the LEFT
+            # factor Q^H (they are outside the trailing two-sided update);
+            # group-consistent: every rank reduces its owned-row partials
+            c1, c2 = j0 + bw, r0
+            VHB = torch.zeros((nrefl, c2 - c1), dtype=dt, device=dev)
+            for a, b in cm.owned_in(r0):
+                VHB += V[a - r0:b - r0].mH @ A[a:b, c1:c2]
+            _assemble(VHB, group)
+            Wb = T.mH @ VHB
+            for a, b in cm.owned_in(r0):
+                A[a:b, c1:c2].addmm_(V[a - r0:b - r0], Wb, beta=1, alpha=-1)
         W = V @ T                                   # replicated (m b^2)
         # Y = A22 W by owned row slabs; X assembled by all-reduce
         Y = torch.zeros((m_p, nrefl), dtype=dt, device=dev)

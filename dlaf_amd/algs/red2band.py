@@ -160,6 +160,15 @@ def reduction_to_band_dense(A: torch.Tensor, band: int):
         V = torch.tril(P[:, :nrefl], -1)
         V = V + torch.eye(m_p, nrefl, dtype=dt, device=dev)
         T = t_factor(V, taus)
+        if bw < band:
+            # capped panel (n - j0 - band < band): the remaining band columns
+            # [j0+bw, j0+band) still have rows >= r0 inside the band and must
+            # receive the LEFT factor Q^H = I - V T^H V^H (they are not part
+            # of the trailing two-sided update below)
+            Bblk = A[r0:, j0 + bw:j0 + band]
+            Wb = T.mH @ (V.mH @ Bblk)
+            Bblk.addmm_(V, Wb, beta=1, alpha=-1)
+            A[j0 + bw:j0 + band, r0:] = Bblk.mH
         A22 = A[r0:, r0:]
         Y = A22 @ (V @ T)                      # hemm (A22 kept full Hermitian)
         S = T.mH @ (V.mH @ Y)
