@@ -82,3 +82,30 @@ def test_bt_red2band_applies_q(dtype):
     bt_reduction_to_band(got, mat, refl)
     err = (got - want).abs().max().item()
     assert err < 1e-13 * n, f"err={err}"
+
+
+@pytest.mark.parametrize("dtype", [torch.float64, torch.complex128])
+@pytest.mark.parametrize("m,k", [(40, 8), (16, 16), (33, 5)])
+def test_t_factor_direct(dtype, m, k):
+    """T factor of a Householder panel (reference
+    ``test/unit/factorization/test_compute_t_factor.cpp`` analog):
+    I - V T V^H must equal the product of the elementary reflectors."""
+    from dlaf_amd.algs.red2band import t_factor
+    torch.manual_seed(m * 31 + k)
+    A = torch.randn(m, k, dtype=torch.float64).to(dtype)
+    if dtype.is_complex:
+        A = A + 1j * torch.randn(m, k, dtype=torch.float64)
+    Q, _ = torch.linalg.qr(A)
+    # build V/taus LAPACK-style via geqrf for a well-defined reflector set
+    a, taus = torch.geqrf(A.clone())
+    V = torch.tril(a, -1) + torch.eye(m, k, dtype=dtype)
+    T = t_factor(V, taus)
+    Qwy = torch.eye(m, dtype=dtype) - V @ T @ V.mH
+    Qref = torch.eye(m, dtype=dtype)
+    for j in range(k):
+        v = V[:, j].clone()
+        v[:j] = 0
+        H = torch.eye(m, dtype=dtype) - taus[j] * torch.outer(v, v.conj())
+        Qref = Qref @ H
+    err = (Qwy - Qref).abs().max().item()
+    assert err < 1e-12 * m, f"err={err}"
