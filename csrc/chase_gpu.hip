@@ -253,14 +253,17 @@ __global__ __launch_bounds__(64) void chase_gpu_k(
       const int64_t nnl = size - j, ml = size - b - j;
       const int nn = (int)(nnl < b ? nnl : b);
       const int m = (int)(ml < b ? ml : b);
-      // stage the unit's whole window (cols j..j+nn-1, full depth) into
-      // LDS. The window is CONTIGUOUS global memory (a + j*ld, nn*ld
-      // elements) — one flat strided loop streams it at line rate.
+      // stage the unit's window (cols j..j+nn-1) into LDS. Column c is only
+      // touched at depths <= nn+m-c (the bulge is a lower-triangular wedge),
+      // so the staircase load moves ~half of nn*ld — the cross-XCD payload
+      // read is the dominant hop cost.
       {
         const T* src = a + j * ld;
-        const int tot = nn * (int)ld;
-        for (int t = lane; t < tot; t += 64)
-          W[(t / (int)ld) * S + (t % (int)ld)] = src[t];
+        const int dmax0 = nn + (m > 0 ? m : 0) + 1;
+        for (int c = 0; c < nn; ++c) {
+          const int dm = min((int)ld, dmax0 - c);
+          for (int d = lane; d < dm; d += 64) W[c * S + d] = src[c * (int)ld + d];
+        }
       }
       __syncthreads();
       unit_two_sided(nn, tau, vl, W, S, lane, wl);
@@ -286,12 +289,14 @@ __global__ __launch_bounds__(64) void chase_gpu_k(
         unit_apply_left(m, nn - 1, tau, vl, W, S, nn - 1, lane);
       }
       __syncthreads();
-      // write the window back (contiguous global), then publish
+      // write the staircase back (same touched region), then publish
       {
         T* dst = a + j * ld;
-        const int tot = nn * (int)ld;
-        for (int t = lane; t < tot; t += 64)
-          dst[t] = W[(t / (int)ld) * S + (t % (int)ld)];
+        const int dmax0 = nn + (m > 0 ? m : 0) + 1;
+        for (int c = 0; c < nn; ++c) {
+          const int dm = min((int)ld, dmax0 - c);
+          for (int d = lane; d < dm; d += 64) dst[c * (int)ld + d] = W[c * S + d];
+        }
       }
       if (last) break;
       publish(s, step);
