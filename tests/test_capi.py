@@ -86,3 +86,56 @@ def test_potrf_info_non_spd():
         ctx, "L", a, capi.DLAF_descriptor(n, n, nb, nb, ld=n))
     assert info > 0
     capi.dlaf_free_grid(ctx)
+
+
+def test_capi_scalapack_shims():
+    """dlaf_pdpotrf / pdpotri / pdsyevd ScaLAPACK-suffixed entry points."""
+    import numpy as np
+    from dlaf_amd import capi
+    n, nb = 48, 16
+    ctx = capi.dlaf_create_grid(1, 1)
+    desc = capi.DLAF_descriptor(n, n, nb, nb, ld=n)
+    rng = np.random.default_rng(3)
+    a0 = rng.standard_normal((n, n))
+    a0 = (a0 + a0.T) / 2 + 2 * n * np.eye(n)
+    a = np.asfortranarray(a0)
+    info = capi.dlaf_pdpotrf(ctx, "L", n, a, 1, 1, desc)
+    assert info == 0
+    L = np.tril(a)
+    assert np.abs(L @ L.T - a0).max() < 1e-10 * n
+    info = capi.dlaf_pdpotri(ctx, "L", n, a, 1, 1, desc)
+    assert info == 0
+    x = np.tril(a) + np.tril(a, -1).T
+    assert np.abs(a0 @ x - np.eye(n)).max() < 1e-8 * n
+    a = np.asfortranarray(a0)
+    w = np.zeros(n)
+    z = np.asfortranarray(np.zeros((n, n)))
+    info = capi.dlaf_pdsyevd(ctx, "L", n, a, desc, w, z, desc)
+    assert info == 0
+    res = np.abs(a0 @ z - z * w).max()
+    assert res < 1e-10 * n * max(1.0, np.abs(w).max())
+    capi.dlaf_free_grid(ctx)
+
+
+def test_capi_partial_spectrum():
+    """dlaf_hermitian_eigensolver partial-spectrum indices."""
+    import numpy as np
+    from dlaf_amd import capi
+    n, nb, il, iu = 40, 16, 5, 20
+    ctx = capi.dlaf_create_grid(1, 1)
+    desc = capi.DLAF_descriptor(n, n, nb, nb, ld=n)
+    rng = np.random.default_rng(5)
+    a0 = rng.standard_normal((n, n))
+    a0 = (a0 + a0.T) / 2
+    a = np.asfortranarray(a0)
+    w = np.zeros(n)
+    z = np.asfortranarray(np.zeros((n, n)))
+    info = capi.dlaf_hermitian_eigensolver(ctx, "L", a, desc, w, z, desc,
+                                           il=il, iu=iu)
+    assert info == 0
+    wr = np.sort(np.linalg.eigvalsh(a0))
+    assert np.abs(w[: iu - il] - wr[il:iu]).max() < 1e-10 * n
+    for j in range(iu - il):
+        r = np.abs(a0 @ z[:, j] - w[j] * z[:, j]).max()
+        assert r < 1e-9 * n
+    capi.dlaf_free_grid(ctx)
