@@ -114,9 +114,8 @@ def generalized_to_standard(uplo: UpLo, mat_a: Matrix, mat_l: Matrix,
                              for lj in range(lj_prev)]
                     ops.gemm_items(rowpAp.storage, linv, mat_a.storage, items, nb,
                                    Op.NoTrans, Op.NoTrans, 1.0, 0.0, uniform=True)
-                    for lj in range(lj_prev):
-                        j = d.global_tile_of_local((0, lj))[1]
-                        mat_a.tile((k, j)).copy_(rowpAp.slot(lj))
+                    lrk = d.local_tile_of_global((k, d.global_tile_of_local((0, 0))[1]))[0]
+                    mat_a.storage[lrk, :lj_prev].copy_(rowpAp.storage[:lj_prev])
                 else:
                     for lj in range(lj_prev):
                         j = d.global_tile_of_local((0, lj))[1]

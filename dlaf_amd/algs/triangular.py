@@ -80,9 +80,8 @@ def _left_mul_row(mat: Matrix, k: int, M: torch.Tensor, opM: Op, rowp: Panel,
         ops.gemm_fused(rowp.storage, M, mat.storage, ops.make_descs(c, a, b),
                        nb, nb, nb, M.stride(0), nb, nb, opM, Op.NoTrans, 1.0, 0.0,
                        uniform=True)
-        for lj in range(lj0, lj1):
-            j = d.global_tile_of_local((0, lj))[1]
-            mat.tile((k, j)).copy_(rowp.slot(lj))
+        lrk = d.local_tile_of_global((k, d.global_tile_of_local((0, lj0))[1]))[0]
+        mat.storage[lrk, lj0:lj1].copy_(rowp.storage[lj0:lj1])
     else:
         for lj in range(lj0, lj1):
             j = d.global_tile_of_local((0, lj))[1]
@@ -280,8 +279,8 @@ def _trsm_lln_local_gpu(diagv: Diag, A: Matrix, B: Matrix) -> None:
             ops.gemm_fused(rowbuf, invs[k], B.storage, ops.make_descs(c, [0] * ntc, offs),
                            nb, nb, nb, nb, nb, nb, Op.NoTrans, Op.NoTrans, 1.0, 0.0,
                            uniform=True)
-            for j in range(ntc):
-                B.tile((k, j)).copy_(rowbuf[j])
+            # whole local row k is contiguous storage: ONE copy, not ntc
+            B.storage[k, :ntc].copy_(rowbuf[:ntc])
             ev_s = torch.cuda.Event()
             ev_s.record(sp)
         # all accumulations into rows > k run on su (no cross-stream writes):
