@@ -87,10 +87,11 @@ def generalized_to_standard(uplo: UpLo, mat_a: Matrix, mat_l: Matrix,
 
     # hoist the diagonal-block inversions (L is read-only here; each
     # tri_inverse_full is a multi-launch recursion off the per-k critical path)
-    linvs = []
+    lkk_tiles = []
     for k in range(nt):
         Lkk = pan.bcast_diag_to_all(mat_l, g, k, l_ws)
-        linvs.append(ops.tri_inverse_full(Lkk, lower=True))
+        lkk_tiles.append(Lkk.clone())
+    linvs = ops.tri_inverse_full_many(lkk_tiles, lower=True)
 
     for k in range(nt):
         kr, kc = d.rank_of_tile((k, k))

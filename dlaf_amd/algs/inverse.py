@@ -84,10 +84,15 @@ def triangular_inverse(uplo: UpLo, diag: Diag, mat: Matrix,
     # on GPU each tri_inverse_full is a multi-launch recursion that otherwise
     # sits on the critical path of every step
     invs = {}
+    own, own_tiles = [], []
     for k in range(nt - 1, -1, -1):
         dtile = pan.bcast_diag_to_col(mat, g, k, diag_ws)
         if d.rank_col == d.rank_of_tile_col(k):
-            invs[k] = ops.tri_inverse_full(dtile, lower=True, unit=unit)
+            own.append(k)
+            own_tiles.append(dtile.clone())
+    for k, inv in zip(own, ops.tri_inverse_full_many(own_tiles, lower=True,
+                                                     unit=unit)):
+        invs[k] = inv
 
     for k in range(nt - 1, -1, -1):
         kc = d.rank_of_tile_col(k)

@@ -248,8 +248,8 @@ def _trsm_lln_local_gpu(diagv: Diag, A: Matrix, B: Matrix) -> None:
     ntc = db.nr_tiles[1]
     unit = diagv == Diag.Unit
     # precompute all diagonal-block inverses (off the critical path)
-    invs = [ops.tri_inverse_full(A.tile((k, k)), lower=True, unit=unit)
-            for k in range(nt)]
+    invs = ops.tri_inverse_full_many((A.tile((k, k)) for k in range(nt)),
+                                     lower=True, unit=unit)
     rt = get_runtime(B.device)
     sp, su = rt.hp_streams[0], rt.np_streams[0]
     cur = torch.cuda.current_stream(B.device)
