@@ -159,3 +159,16 @@ def test_eigensolver_local_complex64():
     assert res < 1e-2, f"res={res}"
     wref = np.linalg.eigvalsh(a0.numpy())
     assert np.abs(np.sort(w.numpy().astype(np.float64)) - wref).max() < 1e-2
+
+
+def test_eigensolver_deterministic():
+    """Full-pipeline bitwise reproducibility (lockstep-replication contract)."""
+    n, nb = 150, 64
+    outs = []
+    for _ in range(2):
+        mat = Matrix.create(n, n, nb, nb, dtype=torch.float64)
+        mutil.set_random_hermitian(mat, seed=9)
+        w, E = hermitian_eigensolver(UpLo.Lower, mat)
+        outs.append((w.clone(), E.to_global().clone()))
+    assert torch.equal(outs[0][0], outs[1][0])
+    assert torch.equal(outs[0][1], outs[1][1])

@@ -67,3 +67,15 @@ def test_tridiag_wilkinson():
     d = torch.tensor([abs(i - m) for i in range(2 * m + 1)], dtype=torch.float64)
     e = torch.ones(2 * m, dtype=torch.float64)
     _check(d, e)
+
+
+def test_tridiag_deterministic():
+    """Bitwise reproducibility — the distributed design replicates D&C in
+    rank-lockstep and relies on identical inputs producing identical bits."""
+    g = torch.Generator().manual_seed(77)
+    n = 300
+    d = torch.randn(n, generator=g, dtype=torch.float64)
+    e = torch.randn(n - 1, generator=g, dtype=torch.float64)
+    w1, E1 = tridiagonal_eigensolver(d.clone(), e.clone(), device="cpu")
+    w2, E2 = tridiagonal_eigensolver(d.clone(), e.clone(), device="cpu")
+    assert torch.equal(w1, w2) and torch.equal(E1, E2)
