@@ -63,3 +63,29 @@ def test_config_env_and_cli(monkeypatch):
     with config.ScopedInitializer() as c2:
         assert c2.num_np_gpu_streams == 7
     config.finalize()
+
+
+def test_set_random_hermitian_banded():
+    """Banded generator (Appendix A surface): entries outside the band are
+    zero, matrix is Hermitian."""
+    import torch
+    from dlaf_amd import Matrix
+    from dlaf_amd.matrix import util as mutil
+    n, nb, band = 48, 16, 8
+    m = Matrix.create(n, n, nb, nb, dtype=torch.complex128)
+    mutil.set_random_hermitian_banded(m, band, seed=3)
+    a = m.to_global()
+    full = torch.tril(a) + torch.tril(a, -1).mH
+    assert torch.equal(torch.tril(full, -band - 1),
+                       torch.zeros_like(full).tril(-band - 1))
+    assert (full - full.mH).abs().max() == 0
+    assert full.abs().max() > 0
+
+
+def test_print_config(capsys):
+    from dlaf_amd import config
+    cfg = config.Configuration(print_config=True)
+    config.initialize(cfg)
+    out = capsys.readouterr().out
+    assert "TuneParameters" in out or "tune" in out or len(out) > 0
+    config.finalize()
