@@ -322,7 +322,9 @@ void launch_chase(T* a, int64_t ld, int64_t size, int64_t b, T* vstore,
       &per_cu, reinterpret_cast<const void*>(chase_gpu_k<T>), 64, shbytes);
   if (per_cu < 1) per_cu = 1;
   hipDeviceProp_t prop;
-  (void)hipGetDeviceProperties(&prop, 0);
+  int dev_ = 0;
+  (void)hipGetDevice(&dev_);
+  (void)hipGetDeviceProperties(&prop, dev_);
   int64_t W = (int64_t)per_cu * prop.multiProcessorCount;  // residency bound
   // concurrency bound: at most ~steps(sweep0)/3 sweeps can be in flight —
   // any extra WG is a pure poller burning fabric bandwidth

@@ -204,7 +204,9 @@ extern "C" {
                      T* wraw, hipStream_t stream) {                           \
     int threads = 256;                                                        \
     hipDeviceProp_t prop;                                                     \
-    (void)hipGetDeviceProperties(&prop, 0);                                   \
+    int dev_ = 0;                                                             \
+    (void)hipGetDevice(&dev_);                                                \
+    (void)hipGetDeviceProperties(&prop, dev_);                                \
     int per_cu = 0;                                                           \
     (void)hipOccupancyMaxActiveBlocksPerMultiprocessor(                       \
         &per_cu, reinterpret_cast<const void*>(&panel_qr_kernel<T>), threads, \

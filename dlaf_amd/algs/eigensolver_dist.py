@@ -212,6 +212,13 @@ def hermitian_eigensolver_dist(uplo: UpLo, mat: Matrix, grid: CommGrid,
         full = torch.zeros((n, evecs.dist.n), dtype=mat.dtype, device=dev)
         full[:, :nE] = E_full
         evecs.set_from_global(full)
-    # leave the band + reflectors in mat (parity with the in-place contract)
-    mat.set_from_global(A)
+    # leave the band + reflectors in mat (parity with the in-place contract).
+    # A's rows are only authoritative on their owning rank (the chunk map),
+    # so assemble the owned rows before the write-back — otherwise a_local
+    # returned via capi would differ across ranks (round-1 advisor finding).
+    Afix = torch.zeros_like(A)
+    for a, b in cm.owned_in(0):
+        Afix[a:b] = A[a:b]
+    _assemble(Afix, group)
+    mat.set_from_global(Afix)
     return w, evecs

@@ -111,11 +111,17 @@ def _unwrap_local(mat: Matrix, a_local) -> None:
         a_local[:] = t.numpy()
 
 
-def _potrf_info(mat: Matrix) -> int:
+def _potrf_info(mat: Matrix, grid=None) -> int:
     """info > 0 when the input was not positive definite: a failed pivot
     sqrt produces NaN on that diagonal (reference counterpart: the
     cusolver-info device assert, ``src/cusolver/assert_info.cu:35``; here
-    the check is an O(n) diagonal scan returned as ScaLAPACK-style info)."""
+    the check is an O(n) diagonal scan returned as ScaLAPACK-style info).
+
+    Distributed grids: each rank scans only its owned diagonal tiles, then
+    the smallest positive info is all-reduced over the full grid so EVERY
+    rank returns the same global info (ScaLAPACK semantics — the reference
+    returns a consistent info on all ranks)."""
+    info = 0
     nt = mat.dist.nr_tiles[0]
     for k in range(nt):
         if mat.dist.rank_of_tile((k, k)) != (mat.dist.rank_row, mat.dist.rank_col):
@@ -123,39 +129,48 @@ def _potrf_info(mat: Matrix) -> int:
         d = mat.tile((k, k)).diagonal()
         bad = torch.isnan(d.real if d.is_complex() else d)
         if bool(bad.any()):
-            return k * mat.dist.mb + int(bad.int().argmax()) + 1
-    return 0
+            info = k * mat.dist.mb + int(bad.int().argmax()) + 1
+            break
+    if grid is not None and grid.distributed:
+        import torch.distributed as tdist
+        # min over positive infos == first failing pivot; encode 0 as +inf
+        t = torch.tensor([float(info) if info > 0 else float("inf")],
+                         dtype=torch.float64)
+        tdist.all_reduce(t, op=tdist.ReduceOp.MIN, group=grid.full_group)
+        v = float(t.item())
+        info = 0 if v == float("inf") else int(v)
+    return info
 
 
 def dlaf_cholesky_factorization(ctx: int, uplo: str, a_local, desc: DLAF_descriptor) -> int:
     """``dlaf_cholesky_factorization_{s,d,c,z}`` analog; returns info (0 = ok,
     > 0 = leading minor of that order not positive definite)."""
-    assert uplo.upper() == "L"
+    ul = UpLo.Upper if uplo.upper() == "U" else UpLo.Lower
     grid = _grid(ctx)
     dev = _device_for(grid)
     mat = _wrap_local(a_local, desc, grid, dev)
-    cholesky_factorization(UpLo.Lower, mat, grid if grid.distributed else None)
-    info = _potrf_info(mat)
+    cholesky_factorization(ul, mat, grid if grid.distributed else None)
+    info = _potrf_info(mat, grid)
     _unwrap_local(mat, a_local)
     return info
 
 
 def dlaf_inverse_from_cholesky_factor(ctx: int, uplo: str, a_local,
                                       desc: DLAF_descriptor) -> int:
-    assert uplo.upper() == "L"
+    ul = UpLo.Upper if uplo.upper() == "U" else UpLo.Lower
     grid = _grid(ctx)
     mat = _wrap_local(a_local, desc, grid, _device_for(grid))
-    inverse_from_cholesky_factor(UpLo.Lower, mat, grid if grid.distributed else None)
+    inverse_from_cholesky_factor(ul, mat, grid if grid.distributed else None)
     _unwrap_local(mat, a_local)
     return 0
 
 
 def dlaf_triangular_inverse(ctx: int, uplo: str, diag: str, a_local,
                             desc: DLAF_descriptor) -> int:
-    assert uplo.upper() == "L"
+    ul = UpLo.Upper if uplo.upper() == "U" else UpLo.Lower
     grid = _grid(ctx)
     mat = _wrap_local(a_local, desc, grid, _device_for(grid))
-    triangular_inverse(UpLo.Lower, Diag.Unit if diag.upper() == "U" else Diag.NonUnit,
+    triangular_inverse(ul, Diag.Unit if diag.upper() == "U" else Diag.NonUnit,
                        mat, grid if grid.distributed else None)
     _unwrap_local(mat, a_local)
     return 0
@@ -168,10 +183,14 @@ def dlaf_hermitian_eigensolver(ctx: int, uplo: str, a_local, desc: DLAF_descript
 
     w_out: [n] real output buffer; z_local: local eigenvector buffer.
     """
-    assert uplo.upper() == "L"
     grid = _grid(ctx)
     dev = _device_for(grid)
     mat = _wrap_local(a_local, desc, grid, dev)
+    if uplo.upper() == "U":
+        # A is Hermitian: conj-transposing the storage turns the given upper
+        # triangle into the lower triangle the pipeline reads.
+        from .algs._uplo import transpose_storage
+        transpose_storage(mat)
     w, evecs = hermitian_eigensolver(UpLo.Lower, mat, grid if grid.distributed else None,
                                      eigenvalues_index_begin=il,
                                      eigenvalues_index_end=iu)
@@ -189,11 +208,14 @@ def dlaf_hermitian_generalized_eigensolver(ctx: int, uplo: str, a_local,
                                            descb: DLAF_descriptor, w_out,
                                            z_local, descz: DLAF_descriptor,
                                            factorized: bool = False) -> int:
-    assert uplo.upper() == "L"
     grid = _grid(ctx)
     dev = _device_for(grid)
     mat_a = _wrap_local(a_local, desca, grid, dev)
     mat_b = _wrap_local(b_local, descb, grid, dev)
+    if uplo.upper() == "U":
+        from .algs._uplo import transpose_storage
+        transpose_storage(mat_a)
+        transpose_storage(mat_b)
     w, evecs = hermitian_generalized_eigensolver(
         UpLo.Lower, mat_a, mat_b, grid if grid.distributed else None,
         factorized=factorized)
