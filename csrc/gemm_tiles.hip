@@ -444,10 +444,37 @@ __launch_bounds__(256) __global__ void gemm_tiles_cplx_k(
 }
 
 template <typename T>
+int v2_dispatch(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
+                int M, int N, int K, int lda, int ldb, int ldc, int opA,
+                int opB, T alpha, T beta, hipStream_t stream);
+template <>
+int v2_dispatch<double>(const GemmDesc* d, int n, const double* A,
+                        const double* B, double* C, int M, int N, int K,
+                        int lda, int ldb, int ldc, int oa, int ob,
+                        double al, double be, hipStream_t s) {
+  return gemm_tiles_v2_f64(d, n, A, B, C, M, N, K, lda, ldb, ldc, oa, ob, al,
+                           be, s);
+}
+template <>
+int v2_dispatch<float>(const GemmDesc* d, int n, const float* A,
+                       const float* B, float* C, int M, int N, int K, int lda,
+                       int ldb, int ldc, int oa, int ob, float al, float be,
+                       hipStream_t s) {
+  return gemm_tiles_v2_f32(d, n, A, B, C, M, N, K, lda, ldb, ldc, oa, ob, al,
+                           be, s);
+}
+
+template <typename T>
 void launch_real(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
                  int M, int N, int K, int lda, int ldb, int ldc, int opA,
                  int opB, T alpha, T beta, hipStream_t stream, int inplace) {
   if (ndesc <= 0 || M <= 0 || N <= 0) return;
+  // v2 glds fast path for full-tile non-inplace batches (the hot phases)
+  if (!inplace &&
+      v2_dispatch<T>(descs, ndesc, A, B, C, M, N, K, lda, ldb, ldc,
+                     (opA == OP_C) ? OP_T : opA, (opB == OP_C) ? OP_T : opB,
+                     alpha, beta, stream))
+    return;
   // kernel variant (microbench lever): 0 = BK16 single-buffer (default),
   // 1 = BK32 single-buffer, 2 = BK16 double-buffer (one barrier per step)
   static const int variant = [] {
@@ -507,10 +534,37 @@ void launch_real(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
 }
 
 template <typename T>
+int v2_dispatch_cplx(const GemmDesc* descs, int ndesc, const T* A, const T* B,
+                     T* C, int M, int N, int K, int lda, int ldb, int ldc,
+                     int opA, int opB, T ar, T ai, T br, T bi,
+                     hipStream_t stream);
+template <>
+int v2_dispatch_cplx<double>(const GemmDesc* d, int n, const double* A,
+                             const double* B, double* C, int M, int N, int K,
+                             int lda, int ldb, int ldc, int oa, int ob,
+                             double ar, double ai, double br, double bi,
+                             hipStream_t s) {
+  return gemm_tiles_v2_c128(d, n, A, B, C, M, N, K, lda, ldb, ldc, oa, ob, ar,
+                            ai, br, bi, s);
+}
+template <>
+int v2_dispatch_cplx<float>(const GemmDesc* d, int n, const float* A,
+                            const float* B, float* C, int M, int N, int K,
+                            int lda, int ldb, int ldc, int oa, int ob,
+                            float ar, float ai, float br, float bi,
+                            hipStream_t s) {
+  return gemm_tiles_v2_c64(d, n, A, B, C, M, N, K, lda, ldb, ldc, oa, ob, ar,
+                           ai, br, bi, s);
+}
+
+template <typename T>
 void launch_cplx(const GemmDesc* descs, int ndesc, const T* A, const T* B, T* C,
                  int M, int N, int K, int lda, int ldb, int ldc, int opA,
                  int opB, T ar, T ai, T br, T bi, hipStream_t stream) {
   if (ndesc <= 0 || M <= 0 || N <= 0) return;
+  if (v2_dispatch_cplx<T>(descs, ndesc, A, B, C, M, N, K, lda, ldb, ldc, opA,
+                          opB, ar, ai, br, bi, stream))
+    return;
   const int mblocks = (M + 63) / 64, nblocks = (N + 63) / 64;
   const dim3 grid(ndesc * mblocks * nblocks);
   const dim3 block(256);

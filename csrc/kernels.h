@@ -52,6 +52,26 @@ void gemm_tiles_c64(const GemmDesc* descs, int ndesc, const float* A,
                     float alpha_im, float beta_re, float beta_im,
                     hipStream_t stream);
 
+// ---- fused batched GEMM v2 (full-tile glds fast path; returns 1 if taken) ----
+int gemm_tiles_v2_f64(const GemmDesc* descs, int ndesc, const double* A,
+                      const double* B, double* C, int M, int N, int K, int lda,
+                      int ldb, int ldc, int opA, int opB, double alpha,
+                      double beta, hipStream_t stream);
+int gemm_tiles_v2_f32(const GemmDesc* descs, int ndesc, const float* A,
+                      const float* B, float* C, int M, int N, int K, int lda,
+                      int ldb, int ldc, int opA, int opB, float alpha,
+                      float beta, hipStream_t stream);
+int gemm_tiles_v2_c128(const GemmDesc* descs, int ndesc, const double* A,
+                       const double* B, double* C, int M, int N, int K,
+                       int lda, int ldb, int ldc, int opA, int opB,
+                       double alpha_re, double alpha_im, double beta_re,
+                       double beta_im, hipStream_t stream);
+int gemm_tiles_v2_c64(const GemmDesc* descs, int ndesc, const float* A,
+                      const float* B, float* C, int M, int N, int K, int lda,
+                      int ldb, int ldc, int opA, int opB, float alpha_re,
+                      float alpha_im, float beta_re, float beta_im,
+                      hipStream_t stream);
+
 // ---- single-tile factorization building blocks ----
 // Fused single-workgroup [factor +] invert of one diagonal block: factors the
 // leading n x n (if do_factor) in place and writes its inverse into the

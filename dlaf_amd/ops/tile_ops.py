@@ -87,6 +87,29 @@ def gemm_fused(
         descs = torch.from_numpy(descs).to(C_base.device, non_blocking=True)
     ar, ai = _alpha_parts(alpha)
     br, bi = _alpha_parts(beta)
+    # In-place safety: the kernels guarantee X = X*op(B) in place only while
+    # one column block spans all of N (every workgroup reads all of its A
+    # rows before any C write). Wider N would let a finished workgroup's C
+    # epilogue race another's A reads — stage op(A)'s tiles into a scratch
+    # buffer and run the ordinary non-aliased kernel instead.
+    bn_safe = 64 if C_base.dtype.is_complex else 128
+    if inplace and N > bn_safe and C_base.is_cuda:
+        assert opA is Op.NoTrans, "in-place staging implemented for opA=N"
+        dd = descs.view(-1, 6)
+        assert int(dd[:, 3].max()) <= 1, "in-place staging expects ktiles == 1"
+        nd = dd.shape[0]
+        af = A_base.reshape(-1)
+        ridx = torch.arange(M, device=C_base.device, dtype=torch.int64) * lda
+        kidx = torch.arange(K, device=C_base.device, dtype=torch.int64)
+        idx = dd[:, 1].view(nd, 1, 1) + ridx.view(1, M, 1) + kidx.view(1, 1, K)
+        scratch = af.index_select(0, idx.reshape(-1)).view(nd, M, K)
+        dd2 = dd.clone()
+        dd2[:, 1] = torch.arange(nd, device=C_base.device, dtype=torch.int64) * (M * K)
+        ext.batch_gemm(
+            C_base.reshape(-1), scratch.reshape(-1), B_base.reshape(-1), dd2,
+            M, N, K, K, ldb, ldc, _opc(opA), _opc(opB), ar, ai, br, bi, False,
+        )
+        return
     if (uniform and not inplace and C_base.is_cuda
             and C_base.dtype in (torch.float64, torch.float32)):
         dd = descs.view(-1, 6)

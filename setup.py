@@ -20,6 +20,7 @@ setup(
                 "csrc/ext.cpp",
                 "csrc/band_chase.cpp",
                 "csrc/gemm_tiles.hip",
+                "csrc/gemm_tiles_v2.hip",
                 "csrc/factor.hip",
                 "csrc/panel_qr.hip",
                 "csrc/secular.hip",

@@ -163,3 +163,79 @@ def test_trsm_panel(dtype):
         err = (panel[i].cpu().to(rdt) - ref).abs().max().item()
         scale = ref.abs().max().item() + 1
         assert err <= _tol(dtype, nb) * scale * 100, f"tile {i} err={err}"
+
+
+@pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("opA", [Op.NoTrans, Op.Trans, Op.ConjTrans])
+@pytest.mark.parametrize("opB", [Op.NoTrans, Op.Trans, Op.ConjTrans])
+def test_gemm_v2_fulltile_ktiles(dtype, opA, opB):
+    """v2 glds fast path: full-tile multi-desc batch with K-tile accumulation
+    (the hot trailing-update shape) vs fp64 torch reference."""
+    torch.manual_seed(3)
+    nb = 128 if not dtype.is_complex else 64
+    M = N = nb
+    K = nb
+    nt = 4  # operand tiles
+    ts = nb * nb
+    A = _rand((nt, nb, nb), dtype)
+    B = _rand((nt, nb, nb), dtype)
+    C = _rand((2, nb, nb), dtype)
+    C0 = C.clone()
+    # two descs: desc0 accumulates ktiles 0..1, desc1 ktiles 2..3
+    descs = ops.make_descs([0, ts], [0, 2 * ts], [0, 2 * ts], ktiles=2,
+                           a_kstride=ts, b_kstride=ts)
+    alpha, beta = (1.5 - 0.5j, 0.25 + 1j) if dtype.is_complex else (-1.0, 1.0)
+    ops.gemm_fused(C, A, B, descs, M, N, K, nb, nb, nb, opA, opB, alpha, beta)
+    torch.cuda.synchronize()
+    hp = torch.complex128 if dtype.is_complex else torch.float64
+    for d in range(2):
+        ref = beta * C0[d].cpu().to(hp)
+        for kt in range(2):
+            t = 2 * d + kt
+            ref = ref + alpha * (_ref_op(A[t].cpu().to(hp), opA)
+                                 @ _ref_op(B[t].cpu().to(hp), opB))
+        err = (C[d].cpu().to(hp) - ref).abs().max().item()
+        scale = ref.abs().max().item() + 1
+        assert err <= _tol(dtype, 2 * K) * scale, f"desc {d}: err={err}"
+
+
+@pytest.mark.parametrize("dtype", [torch.float64, torch.float32])
+def test_gemm_v2_wide(dtype):
+    """v2 at a multi-block C (512x384) with rectangular K."""
+    torch.manual_seed(4)
+    M, N, K = 512, 384, 256
+    A = _rand((M, K), dtype)
+    B = _rand((K, N), dtype)
+    C = torch.zeros((M, N), dtype=dtype, device="cuda")
+    descs = ops.make_descs([0], [0], [0])
+    ops.gemm_fused(C, A, B, descs, M, N, K, K, N, N, Op.NoTrans, Op.NoTrans, 1.0, 0.0)
+    torch.cuda.synchronize()
+    hp = torch.float64
+    ref = A.cpu().to(hp) @ B.cpu().to(hp)
+    err = (C.cpu().to(hp) - ref).abs().max().item()
+    assert err <= _tol(dtype, K) * (ref.abs().max().item() + 1), f"err={err}"
+
+
+@pytest.mark.parametrize("dtype", [torch.float64, torch.complex128])
+def test_gemm_inplace_wide_n(dtype):
+    """In-place X = X @ op(B) with N wider than one kernel column block:
+    exercises the scratch-staging path in gemm_fused (round-2 race fix)."""
+    torch.manual_seed(5)
+    nb = 256
+    nt = 3
+    X = _rand((nt, nb, nb), dtype)
+    Bm = _rand((nb, nb), dtype)
+    X0 = X.clone()
+    ts = nb * nb
+    offs = [i * ts for i in range(nt)]
+    descs = ops.make_descs(offs, offs, [0] * nt)
+    ops.gemm_fused(X, X, Bm, descs, nb, nb, nb, nb, nb, nb,
+                   Op.NoTrans, Op.ConjTrans if dtype.is_complex else Op.Trans,
+                   1.0, 0.0, inplace=True)
+    torch.cuda.synchronize()
+    hp = torch.complex128 if dtype.is_complex else torch.float64
+    for i in range(nt):
+        ref = X0[i].cpu().to(hp) @ (Bm.cpu().to(hp).mH if dtype.is_complex
+                                    else Bm.cpu().to(hp).mT)
+        err = (X[i].cpu().to(hp) - ref).abs().max().item()
+        assert err <= _tol(dtype, nb) * (ref.abs().max().item() + 1), f"{i}: {err}"
