@@ -63,74 +63,89 @@ __device__ inline int xcd_remap(int wg, int nwg) {
 }
 
 // ---------------- real kernel ----------------
-// BM=BN=128, BK=16, 256 threads; wave w owns the 64x64 quadrant
-// (wrow, wcol) = ((w>>1)*64, (w&1)*64).
-// LDS (one __shared__ object — §5 trap 4a): S[buf][A:2048 | B:2048] elems.
-//   A image: OPA==N -> row-major [i][k] (glds row 8i-group), else k-major
-//   [k][i]; B image: OPB==N -> k-major [k][c], else row-major [c][k].
-//   Every image is glds-lane-linear: instr t covers LDS elems t*128..+127.
-template <typename T, int OPA, int OPB, int SWZ>
-__launch_bounds__(256, 2) __global__ void gemm_v2_k(
+// BM=128 x BN (template 64/128), BK=16, 256 threads; 4 waves as a 2x2 grid
+// of (64 x BN/2) tiles; NFRAG = BN/32 fragments of mfma_{f64,f32}_16x16x4.
+// LDS (one __shared__ object — §5 trap 4a): S[DEPTH][A: 128*BK | B: BK*BN].
+//   A image: OPA==N -> row-major [i][k] (XOR chunk swizzle, below), else
+//   k-major [k][i]; B image: OPB==N -> k-major [k][c], else row-major
+//   [c][k] (swizzled). Images are glds-lane-linear: instr t covers LDS
+//   elems t*64*EPL..+64*EPL-1.
+//
+// Swizzle: a row-major image read column-wise is up to 8-way bank
+// conflicted (rows are 128 B = 2 banks apart). The 16-B chunk at (row, c)
+// holds global chunk c ^ (row & (CPR-1)) — applied on the per-lane glds
+// SOURCE address (LDS stays lane-linear, the glds contract) and undone in
+// the fragment index; residual conflict is 2-way.
+template <typename T, int OPA, int OPB, int SWZ, int DEPTH, int BN>
+__launch_bounds__(256, DEPTH == 2 ? 2 : 1) __global__ void gemm_v2_k(
     const GemmDesc* __restrict__ descs, const T* __restrict__ A,
     const T* __restrict__ B, T* __restrict__ C, int M, int N, int K, int lda,
     int ldb, int ldc, T alpha, T beta, int mblocks, int nblocks) {
+  constexpr int BM = 128;
   constexpr int BK = 16;
   constexpr int EPL = 16 / sizeof(T);  // elements per lane per glds
-  __shared__ T S[2][4096 * (8 / sizeof(T))];
+  constexpr int CPR = BK / EPL;        // 16-B chunks per rk-image row
+  constexpr int AEL = BM * BK;         // A image elements
+  constexpr int BEL = BK * BN;         // B image elements
+  constexpr int NFRAG = BN / 32;
+  __shared__ T S[DEPTH][AEL + BEL];
 
   int wg = blockIdx.x;
   if constexpr (SWZ) wg = xcd_remap(wg, gridDim.x);
   const int per_desc = mblocks * nblocks;
   const GemmDesc d = descs[wg / per_desc];
   const int rem = wg % per_desc;
-  const int i0 = (rem / nblocks) * 128, j0 = (rem % nblocks) * 128;
+  const int i0 = (rem / nblocks) * BM, j0 = (rem % nblocks) * BN;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63, w = tid >> 6;
-  const int wrow = (w >> 1) * 64, wcol = (w & 1) * 64;
+  const int wrow = (w >> 1) * 64, wcol = (w & 1) * (BN / 2);
   const int li = lane & 15, lk = lane >> 4;
 
   const int spt = K / BK;  // steps per K-tile
   const int total = (int)d.ktiles * spt;
-  // glds instructions per operand image (2048 elems) and per wave
-  constexpr int NAW = 2048 / (64 * EPL) / 4;  // f64: 4, f32: 2
+  // glds instructions per wave per step, per operand
+  constexpr int NAW = AEL / (64 * EPL) / 4;
+  constexpr int NBW = BEL / (64 * EPL) / 4;
 
   using acc_t = typename MfmaV2<T>::acc_t;
-  acc_t acc[4][4];
+  acc_t acc[4][NFRAG];
 #pragma unroll
   for (int a = 0; a < 4; ++a)
 #pragma unroll
-    for (int b = 0; b < 4; ++b) acc[a][b] = {0, 0, 0, 0};
+    for (int b = 0; b < NFRAG; ++b) acc[a][b] = {0, 0, 0, 0};
 
   // per-lane glds source offsets for instr j=0 of this wave (element units,
   // relative to the K-tile base) and the per-instr / per-step increments
   const T* Abase0 = A + d.a_off;
   const T* Bbase0 = B + d.b_off;
   int64_t a_l0, a_jinc, a_sinc;
-  if constexpr (OPA == OP_N) {  // [i][k] image: one instr = 1024 B of rows
-    constexpr int RPI = 64 * EPL / BK;  // rows per instr (f64: 8, f32: 16)
-    a_l0 = (int64_t)(i0 + (NAW * w) * RPI + lane / (BK / EPL)) * lda +
-           (lane % (BK / EPL)) * EPL;
+  if constexpr (OPA == OP_N) {  // swizzled rk image [i][k]
+    constexpr int RPI = 64 / CPR;  // rows per instr
+    const int r_in = lane / CPR;   // row within instr; also the swizzle mask
+    a_l0 = (int64_t)(i0 + (NAW * w) * RPI + r_in) * lda +
+           EPL * ((lane % CPR) ^ (r_in & (CPR - 1)));
     a_jinc = (int64_t)RPI * lda;
     a_sinc = BK;
-  } else {  // [k][i] image: one instr = whole 128-elem k-rows
-    constexpr int KPI = 64 * EPL / 128;  // k-rows per instr (f64: 1, f32: 2)
-    a_l0 = (int64_t)(NAW * w * KPI + lane / (128 / EPL)) * lda + i0 +
-           (lane % (128 / EPL)) * EPL;
+  } else {  // kr image [k][BM]
+    constexpr int KPI = 64 * EPL / BM;
+    a_l0 = (int64_t)(NAW * w * KPI + lane / (BM / EPL)) * lda + i0 +
+           (lane % (BM / EPL)) * EPL;
     a_jinc = (int64_t)KPI * lda;
     a_sinc = (int64_t)BK * lda;
   }
   int64_t b_l0, b_jinc, b_sinc;
-  if constexpr (OPB == OP_N) {  // [k][c] image
-    constexpr int KPI = 64 * EPL / 128;
-    b_l0 = (int64_t)(NAW * w * KPI + lane / (128 / EPL)) * ldb + j0 +
-           (lane % (128 / EPL)) * EPL;
+  if constexpr (OPB == OP_N) {  // kr image [k][BN]
+    constexpr int KPI = 64 * EPL / BN;
+    b_l0 = (int64_t)(NBW * w * KPI + lane / (BN / EPL)) * ldb + j0 +
+           (lane % (BN / EPL)) * EPL;
     b_jinc = (int64_t)KPI * ldb;
     b_sinc = (int64_t)BK * ldb;
-  } else {  // [c][k] image
-    constexpr int RPI = 64 * EPL / BK;
-    b_l0 = (int64_t)(j0 + (NAW * w) * RPI + lane / (BK / EPL)) * ldb +
-           (lane % (BK / EPL)) * EPL;
+  } else {  // swizzled rk image [c][k]
+    constexpr int RPI = 64 / CPR;
+    const int r_in = lane / CPR;
+    b_l0 = (int64_t)(j0 + (NBW * w) * RPI + r_in) * ldb +
+           EPL * ((lane % CPR) ^ (r_in & (CPR - 1)));
     b_jinc = (int64_t)RPI * ldb;
     b_sinc = BK;
   }
@@ -141,17 +156,17 @@ __launch_bounds__(256, 2) __global__ void gemm_v2_k(
   int ks_cnt = 0;
   int64_t kt = 0;
   T* const Sflat = &S[0][0];
-  constexpr int BUFE = 4096 * (8 / sizeof(T));  // elems per LDS buffer
-  constexpr int BOFF = 2048 * (8 / sizeof(T));  // B image offset in a buffer
-  const int ldsw = w * NAW * 64 * EPL;  // this wave's LDS elem offset (j=0)
+  constexpr int BUFE = AEL + BEL;
+  const int ldsa = w * NAW * 64 * EPL;  // this wave's A LDS offset (j=0)
+  const int ldsb = AEL + w * NBW * 64 * EPL;
 
   auto issue = [&](int buf) {
 #pragma unroll
     for (int j = 0; j < NAW; ++j)
-      glds16(aP + j * a_jinc, Sflat + buf * BUFE + ldsw + j * 64 * EPL);
+      glds16(aP + j * a_jinc, Sflat + buf * BUFE + ldsa + j * 64 * EPL);
 #pragma unroll
-    for (int j = 0; j < NAW; ++j)
-      glds16(bP + j * b_jinc, Sflat + buf * BUFE + BOFF + ldsw + j * 64 * EPL);
+    for (int j = 0; j < NBW; ++j)
+      glds16(bP + j * b_jinc, Sflat + buf * BUFE + ldsb + j * 64 * EPL);
     // advance to the next K step (next K tile when this one is done)
     if (++ks_cnt == spt) {
       ks_cnt = 0;
@@ -166,19 +181,24 @@ __launch_bounds__(256, 2) __global__ void gemm_v2_k(
 
   auto compute = [&](int buf) {
     const T* sa = Sflat + buf * BUFE;
-    const T* sb = sa + BOFF;
-    // fragment index helpers for the two image layouts
+    const T* sb = sa + AEL;
+    // fragment index helpers (undo the rk swizzle)
     auto aidx = [&](int row, int k) {
-      return (OPA == OP_N) ? row * BK + k : k * 128 + row;
+      return (OPA == OP_N)
+                 ? row * BK + EPL * ((k / EPL) ^ (row & (CPR - 1))) + k % EPL
+                 : k * BM + row;
     };
     auto bidx = [&](int col, int k) {
-      return (OPB == OP_N) ? k * 128 + col : col * BK + k;
+      return (OPB == OP_N)
+                 ? k * BN + col
+                 : col * BK + EPL * ((k / EPL) ^ (col & (CPR - 1))) + k % EPL;
     };
-    T af[2][4], bf[2][4];
+    T af[2][4], bf[2][NFRAG];
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) af[0][mi] = sa[aidx(wrow + mi * 16 + li, lk)];
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) bf[0][ni] = sb[bidx(wcol + ni * 16 + li, lk)];
+    for (int ni = 0; ni < NFRAG; ++ni)
+      bf[0][ni] = sb[bidx(wcol + ni * 16 + li, lk)];
 #pragma unroll
     for (int ks = 0; ks < BK / 4; ++ks) {
       const int cur = ks & 1;
@@ -189,47 +209,90 @@ __launch_bounds__(256, 2) __global__ void gemm_v2_k(
         for (int mi = 0; mi < 4; ++mi)
           af[nxt][mi] = sa[aidx(wrow + mi * 16 + li, k)];
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
+        for (int ni = 0; ni < NFRAG; ++ni)
           bf[nxt][ni] = sb[bidx(wcol + ni * 16 + li, k)];
       }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
+        for (int ni = 0; ni < NFRAG; ++ni)
           acc[mi][ni] = MfmaV2<T>::mma(af[cur][mi], bf[cur][ni], acc[mi][ni]);
     }
   };
 
-  issue(0);
+  // pipeline: DEPTH buffers, DEPTH-1 steps in flight. DEPTH==2 uses plain
+  // __syncthreads() (its vmcnt(0) drain IS the completion wait); DEPTH>=3
+  // uses counted per-wave vmcnt waits + raw barriers so DEPTH-2 steps stay
+  // in flight across each barrier (the guide's 3-buf span pattern).
+  constexpr int GPW = NAW + NBW;  // glds per wave per step
+  int ibuf = 0, cbuf = 0;
+  for (int p = 0; p < DEPTH - 1 && p < total; ++p) {
+    issue(ibuf);
+    if (++ibuf == DEPTH) ibuf = 0;
+  }
   for (int s = 0; s < total; ++s) {
-    __syncthreads();  // drains the in-flight glds (vmcnt 0) + LDS reuse
-    if (s + 1 < total) issue(1 - (s & 1));
-    compute(s & 1);
+    if constexpr (DEPTH == 2) {
+      __syncthreads();  // drains the in-flight glds (vmcnt 0) + LDS reuse
+    } else {
+      const int rem_s = total - s - 1;  // steps still to land after this one
+      if (rem_s >= DEPTH - 2)
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(GPW * (DEPTH - 2)) : "memory");
+      else if (rem_s == 1)
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(GPW) : "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+    if (s + DEPTH - 1 < total) {
+      issue(ibuf);
+      if (++ibuf == DEPTH) ibuf = 0;
+    }
+    compute(cbuf);
+    if (++cbuf == DEPTH) cbuf = 0;
   }
 
-  // epilogue: C = alpha*acc + beta*C (beta test hoisted out of the loops)
+  // epilogue: C = alpha*acc + beta*C. The beta path batches its C loads per
+  // mi-chunk (loads, then uses) — interleaved load/use would eat a full
+  // vmcnt(0) drain per element (hipcc is conservative about ordinary loads
+  // in a function that issued glds).
   T* Cb = C + d.c_off;
-  auto store = [&](auto betanz) {
+  if (beta != T(0)) {
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
+    for (int mi = 0; mi < 4; ++mi) {
+      T cv[4 * NFRAG];
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
+      for (int ni = 0; ni < NFRAG; ++ni)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = i0 + wrow + mi * 16 + MfmaV2<T>::acc_row(lk, r);
+          const int col = j0 + wcol + ni * 16 + li;
+          cv[ni * 4 + r] = Cb[(int64_t)row * ldc + col];
+        }
+#pragma unroll
+      for (int ni = 0; ni < NFRAG; ++ni) {
         const acc_t v = acc[mi][ni];
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int row = i0 + wrow + mi * 16 + MfmaV2<T>::acc_row(lk, r);
           const int col = j0 + wcol + ni * 16 + li;
-          const int64_t off = (int64_t)row * ldc + col;
-          T out = alpha * (T)v[r];
-          if constexpr (decltype(betanz)::value) out += beta * Cb[off];
-          Cb[off] = out;
+          Cb[(int64_t)row * ldc + col] = alpha * (T)v[r] + beta * cv[ni * 4 + r];
         }
       }
-  };
-  if (beta != T(0))
-    store(std::true_type{});
-  else
-    store(std::false_type{});
+    }
+  } else {
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < NFRAG; ++ni) {
+        const acc_t v = acc[mi][ni];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = i0 + wrow + mi * 16 + MfmaV2<T>::acc_row(lk, r);
+          const int col = j0 + wcol + ni * 16 + li;
+          Cb[(int64_t)row * ldc + col] = alpha * (T)v[r];
+        }
+      }
+  }
 }
 
 // ---------------- complex kernel ----------------
@@ -237,8 +300,8 @@ __launch_bounds__(256, 2) __global__ void gemm_v2_k(
 // two). BM=BN=64, BK=16; wave w owns the 32x32 quadrant; 2x2 fragments,
 // 4 MFMA per fragment pair. LDS images mirror the real kernel with complex
 // elements; fragments are read as (re,im) pairs in one ds_read_b128 (f64).
-template <typename T, int OPA, int OPB, int SWZ>
-__launch_bounds__(256, 2) __global__ void gemm_v2_cplx_k(
+template <typename T, int OPA, int OPB, int SWZ, int DEPTH = 2>
+__launch_bounds__(256, DEPTH == 2 ? 2 : 1) __global__ void gemm_v2_cplx_k(
     const GemmDesc* __restrict__ descs, const T* __restrict__ A,
     const T* __restrict__ B, T* __restrict__ C, int M, int N, int K, int lda,
     int ldb, int ldc, T alpha_re, T alpha_im, T beta_re, T beta_im,
@@ -246,7 +309,7 @@ __launch_bounds__(256, 2) __global__ void gemm_v2_cplx_k(
   constexpr int BK = 16;
   constexpr int CPL = 8 / sizeof(T);  // complex elems per lane per glds
   // S[buf][A: 64*16 | B: 16*64] complex elems, interleaved re/im
-  __shared__ T S[2][4096 * (8 / sizeof(T))];
+  __shared__ T S[DEPTH][4096 * (8 / sizeof(T))];
 
   int wg = blockIdx.x;
   if constexpr (SWZ) wg = xcd_remap(wg, gridDim.x);
@@ -371,15 +434,66 @@ __launch_bounds__(256, 2) __global__ void gemm_v2_cplx_k(
     }
   };
 
-  issue(0);
+  constexpr int GPW = 2 * NAW;
+  int ibuf = 0, cbuf = 0;
+  for (int p = 0; p < DEPTH - 1 && p < total; ++p) {
+    issue(ibuf);
+    if (++ibuf == DEPTH) ibuf = 0;
+  }
   for (int s = 0; s < total; ++s) {
-    __syncthreads();
-    if (s + 1 < total) issue(1 - (s & 1));
-    compute(s & 1);
+    if constexpr (DEPTH == 2) {
+      __syncthreads();
+    } else {
+      const int rem_s = total - s - 1;
+      if (rem_s >= DEPTH - 2)
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(GPW * (DEPTH - 2)) : "memory");
+      else if (rem_s == 1)
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(GPW) : "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+    if (s + DEPTH - 1 < total) {
+      issue(ibuf);
+      if (++ibuf == DEPTH) ibuf = 0;
+    }
+    compute(cbuf);
+    if (++cbuf == DEPTH) cbuf = 0;
   }
 
   T* Cb = C + 2 * d.c_off;
-  auto store = [&](auto betanz) {
+  if (beta_re != T(0) || beta_im != T(0)) {
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+      T cv[16];
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = i0 + wrow + mi * 16 + MfmaV2<T>::acc_row(lk, r);
+          const int col = j0 + wcol + ni * 16 + li;
+          const int64_t off = 2 * ((int64_t)row * ldc + col);
+          cv[(ni * 4 + r) * 2] = Cb[off];
+          cv[(ni * 4 + r) * 2 + 1] = Cb[off + 1];
+        }
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        const acc_t vr = accr[mi][ni];
+        const acc_t vi = acci[mi][ni];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = i0 + wrow + mi * 16 + MfmaV2<T>::acc_row(lk, r);
+          const int col = j0 + wcol + ni * 16 + li;
+          const int64_t off = 2 * ((int64_t)row * ldc + col);
+          const T cr = cv[(ni * 4 + r) * 2], ci = cv[(ni * 4 + r) * 2 + 1];
+          Cb[off] = alpha_re * (T)vr[r] - alpha_im * (T)vi[r] +
+                    beta_re * cr - beta_im * ci;
+          Cb[off + 1] = alpha_re * (T)vi[r] + alpha_im * (T)vr[r] +
+                        beta_re * ci + beta_im * cr;
+        }
+      }
+    }
+  } else {
 #pragma unroll
     for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
@@ -391,29 +505,18 @@ __launch_bounds__(256, 2) __global__ void gemm_v2_cplx_k(
           const int row = i0 + wrow + mi * 16 + MfmaV2<T>::acc_row(lk, r);
           const int col = j0 + wcol + ni * 16 + li;
           const int64_t off = 2 * ((int64_t)row * ldc + col);
-          T outr = alpha_re * (T)vr[r] - alpha_im * (T)vi[r];
-          T outi = alpha_re * (T)vi[r] + alpha_im * (T)vr[r];
-          if constexpr (decltype(betanz)::value) {
-            const T cr = Cb[off], ci = Cb[off + 1];
-            outr += beta_re * cr - beta_im * ci;
-            outi += beta_re * ci + beta_im * cr;
-          }
-          Cb[off] = outr;
-          Cb[off + 1] = outi;
+          Cb[off] = alpha_re * (T)vr[r] - alpha_im * (T)vi[r];
+          Cb[off + 1] = alpha_re * (T)vi[r] + alpha_im * (T)vr[r];
         }
       }
-  };
-  if (beta_re != T(0) || beta_im != T(0))
-    store(std::true_type{});
-  else
-    store(std::false_type{});
+  }
 }
 
 template <typename T>
 int launch_v2_real(const GemmDesc* descs, int ndesc, const T* A, const T* B,
                    T* C, int M, int N, int K, int lda, int ldb, int ldc,
                    int opA, int opB, T alpha, T beta, hipStream_t stream) {
-  if (M % 128 || N % 128 || K % 16 || K <= 0) return 0;
+  if (M % 128 || N % 64 || K % 16 || K <= 0) return 0;
   static const int enabled = [] {
     const char* v = getenv("DLAF_GEMM_V2");
     return v ? atoi(v) : 1;
@@ -422,26 +525,47 @@ int launch_v2_real(const GemmDesc* descs, int ndesc, const T* A, const T* B,
     const char* v = getenv("DLAF_GEMM_V2_SWZ");
     return v ? atoi(v) : 1;
   }();
+  static const int depth = [] {
+    const char* v = getenv("DLAF_GEMM_V2_DEPTH");
+    const int d = v ? atoi(v) : 2;
+    return d < 2 ? 2 : (d > 3 ? 3 : d);
+  }();
+  static const int bn_env = [] {
+    const char* v = getenv("DLAF_GEMM_V2_BN");
+    return v ? atoi(v) : 64;
+  }();
   if (!enabled) return 0;
-  const int mblocks = M / 128, nblocks = N / 128;
+  const int bn = (N % bn_env == 0) ? bn_env : (N % 64 == 0 ? 64 : 128);
+  if (N % bn) return 0;
+  const int mblocks = M / 128, nblocks = N / bn;
   const dim3 grid(ndesc * mblocks * nblocks);
   const dim3 block(256);
   const int oa = (opA == OP_C) ? OP_T : opA;
   const int ob = (opB == OP_C) ? OP_T : opB;
+#define LV2(OA, OB, SW, DP, BNv)                                           \
+  gemm_v2_k<T, OA, OB, SW, DP, BNv><<<grid, block, 0, stream>>>(           \
+      descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks,        \
+      nblocks)
 #define CASE(OA, OB)                                                       \
   if (oa == OA && ob == OB) {                                              \
-    if (swz)                                                               \
-      gemm_v2_k<T, OA, OB, 1><<<grid, block, 0, stream>>>(                 \
-          descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks,    \
-          nblocks);                                                        \
-    else                                                                   \
-      gemm_v2_k<T, OA, OB, 0><<<grid, block, 0, stream>>>(                 \
-          descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks,    \
-          nblocks);                                                        \
+    if (bn == 64) {                                                        \
+      if (depth == 2) {                                                    \
+        if (swz) LV2(OA, OB, 1, 2, 64); else LV2(OA, OB, 0, 2, 64);        \
+      } else {                                                             \
+        if (swz) LV2(OA, OB, 1, 3, 64); else LV2(OA, OB, 0, 3, 64);        \
+      }                                                                    \
+    } else {                                                               \
+      if (depth == 2) {                                                    \
+        if (swz) LV2(OA, OB, 1, 2, 128); else LV2(OA, OB, 0, 2, 128);      \
+      } else {                                                             \
+        if (swz) LV2(OA, OB, 1, 3, 128); else LV2(OA, OB, 0, 3, 128);      \
+      }                                                                    \
+    }                                                                      \
     return 1;                                                              \
   }
   CASE(OP_N, OP_N) CASE(OP_N, OP_T) CASE(OP_T, OP_N) CASE(OP_T, OP_T)
 #undef CASE
+#undef LV2
   return 0;
 }
 
@@ -459,26 +583,35 @@ int launch_v2_cplx(const GemmDesc* descs, int ndesc, const T* A, const T* B,
     const char* v = getenv("DLAF_GEMM_V2_SWZ");
     return v ? atoi(v) : 1;
   }();
+  static const int depth = [] {
+    const char* v = getenv("DLAF_GEMM_V2_DEPTH");
+    const int d = v ? atoi(v) : 3;
+    return d < 2 ? 2 : (d > 4 ? 4 : d);
+  }();
   if (!enabled) return 0;
   const int mblocks = M / 64, nblocks = N / 64;
   const dim3 grid(ndesc * mblocks * nblocks);
   const dim3 block(256);
+#define LV2C(OA, OB, SW, DP)                                               \
+  gemm_v2_cplx_k<T, OA, OB, SW, DP><<<grid, block, 0, stream>>>(           \
+      descs, A, B, C, M, N, K, lda, ldb, ldc, ar, ai, br, bi, mblocks,     \
+      nblocks)
 #define CASE(OA, OB)                                                       \
   if (opA == OA && opB == OB) {                                            \
-    if (swz)                                                               \
-      gemm_v2_cplx_k<T, OA, OB, 1><<<grid, block, 0, stream>>>(            \
-          descs, A, B, C, M, N, K, lda, ldb, ldc, ar, ai, br, bi, mblocks, \
-          nblocks);                                                        \
-    else                                                                   \
-      gemm_v2_cplx_k<T, OA, OB, 0><<<grid, block, 0, stream>>>(            \
-          descs, A, B, C, M, N, K, lda, ldb, ldc, ar, ai, br, bi, mblocks, \
-          nblocks);                                                        \
+    if (depth == 2) {                                                      \
+      if (swz) LV2C(OA, OB, 1, 2); else LV2C(OA, OB, 0, 2);                \
+    } else if (depth == 3) {                                               \
+      if (swz) LV2C(OA, OB, 1, 3); else LV2C(OA, OB, 0, 3);                \
+    } else {                                                               \
+      if (swz) LV2C(OA, OB, 1, 4); else LV2C(OA, OB, 0, 4);                \
+    }                                                                      \
     return 1;                                                              \
   }
   CASE(OP_N, OP_N) CASE(OP_N, OP_T) CASE(OP_N, OP_C)
   CASE(OP_T, OP_N) CASE(OP_T, OP_T) CASE(OP_T, OP_C)
   CASE(OP_C, OP_N) CASE(OP_C, OP_T) CASE(OP_C, OP_C)
 #undef CASE
+#undef LV2C
   return 0;
 }
 

@@ -76,11 +76,12 @@ def gemm_fused(
     applies) — it selects the kernel geometry that reads all of A before
     writing C.
 
-    Set ``uniform=True`` when the caller guarantees ktiles == 1 and no two
-    descs share a C block: real-dtype batches then route to rocBLAS
-    pointer-array batched GEMM (65.7 vs 45.3 TF on the POTRF trailing shape,
-    profiles/microbench_r1.log). Complex stays on the fused CDNA4 kernel
-    (rocblas_zgemm_batched measured 2.7x SLOWER than it).
+    ``uniform`` is accepted for API compatibility but no longer routes
+    anywhere: since round 2 the hand-written v2 glds kernel matches or beats
+    rocBLAS batched on the uniform nb=512 trailing shape (62.8 vs 62.5 TF
+    within one probe; round-1 v1 was 45 vs 65.7), so EVERY batch runs on the
+    in-tree CDNA4 kernels. rocBLAS remains available only as a test oracle
+    (``gemm_batched_lib``).
     """
     ext = get_ext()
     if isinstance(descs, np.ndarray):
@@ -109,16 +110,6 @@ def gemm_fused(
             C_base.reshape(-1), scratch.reshape(-1), B_base.reshape(-1), dd2,
             M, N, K, K, ldb, ldc, _opc(opA), _opc(opB), ar, ai, br, bi, False,
         )
-        return
-    if (uniform and not inplace and C_base.is_cuda
-            and C_base.dtype in (torch.float64, torch.float32)):
-        dd = descs.view(-1, 6)
-        es = C_base.element_size()
-        pc = dd[:, 0] * es + C_base.data_ptr()
-        pa = dd[:, 1] * es + A_base.data_ptr()
-        pb = dd[:, 2] * es + B_base.data_ptr()
-        ext.lib_gemm_batched(C_base, pc, pa, pb, M, N, K, lda, ldb, ldc,
-                             _opc(opA), _opc(opB), ar, ai, br, bi)
         return
     ext.batch_gemm(
         C_base.reshape(-1), A_base.reshape(-1), B_base.reshape(-1), descs,
