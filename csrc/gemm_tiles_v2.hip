@@ -76,13 +76,13 @@ __device__ inline int xcd_remap(int wg, int nwg) {
 // holds global chunk c ^ (row & (CPR-1)) — applied on the per-lane glds
 // SOURCE address (LDS stays lane-linear, the glds contract) and undone in
 // the fragment index; residual conflict is 2-way.
-template <typename T, int OPA, int OPB, int SWZ, int DEPTH, int BN>
+template <typename T, int OPA, int OPB, int SWZ, int DEPTH, int BN,
+          int BK = 16>
 __launch_bounds__(256, DEPTH == 2 ? 2 : 1) __global__ void gemm_v2_k(
     const GemmDesc* __restrict__ descs, const T* __restrict__ A,
     const T* __restrict__ B, T* __restrict__ C, int M, int N, int K, int lda,
     int ldb, int ldc, T alpha, T beta, int mblocks, int nblocks) {
   constexpr int BM = 128;
-  constexpr int BK = 16;
   constexpr int EPL = 16 / sizeof(T);  // elements per lane per glds
   constexpr int CPR = BK / EPL;        // 16-B chunks per rk-image row
   constexpr int AEL = BM * BK;         // A image elements
@@ -516,7 +516,7 @@ template <typename T>
 int launch_v2_real(const GemmDesc* descs, int ndesc, const T* A, const T* B,
                    T* C, int M, int N, int K, int lda, int ldb, int ldc,
                    int opA, int opB, T alpha, T beta, hipStream_t stream) {
-  if (M % 128 || N % 64 || K % 16 || K <= 0) return 0;
+  if (M % 128 || N % 64 || K % 16 || K <= 0) return 0;  // BK=8 needs K%16 too
   static const int enabled = [] {
     const char* v = getenv("DLAF_GEMM_V2");
     return v ? atoi(v) : 1;
@@ -532,7 +532,11 @@ int launch_v2_real(const GemmDesc* descs, int ndesc, const T* A, const T* B,
   }();
   static const int bn_env = [] {
     const char* v = getenv("DLAF_GEMM_V2_BN");
-    return v ? atoi(v) : 64;
+    return v ? atoi(v) : 128;
+  }();
+  static const int bk_env = [] {
+    const char* v = getenv("DLAF_GEMM_V2_BK");
+    return v ? atoi(v) : 16;
   }();
   if (!enabled) return 0;
   const int bn = (N % bn_env == 0) ? bn_env : (N % 64 == 0 ? 64 : 128);
@@ -546,9 +550,15 @@ int launch_v2_real(const GemmDesc* descs, int ndesc, const T* A, const T* B,
   gemm_v2_k<T, OA, OB, SW, DP, BNv><<<grid, block, 0, stream>>>(           \
       descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks,        \
       nblocks)
+#define LV2K(OA, OB, SW, DP, BNv, BKv)                                    \
+  gemm_v2_k<T, OA, OB, SW, DP, BNv, BKv><<<grid, block, 0, stream>>>(      \
+      descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks,        \
+      nblocks)
 #define CASE(OA, OB)                                                       \
   if (oa == OA && ob == OB) {                                              \
-    if (bn == 64) {                                                        \
+    if (bk_env == 8 && bn == 64) {                                         \
+      if (swz) LV2K(OA, OB, 1, 2, 64, 8); else LV2K(OA, OB, 0, 2, 64, 8);  \
+    } else if (bn == 64) {                                                 \
       if (depth == 2) {                                                    \
         if (swz) LV2(OA, OB, 1, 2, 64); else LV2(OA, OB, 0, 2, 64);        \
       } else {                                                             \
@@ -565,6 +575,7 @@ int launch_v2_real(const GemmDesc* descs, int ndesc, const T* A, const T* B,
   }
   CASE(OP_N, OP_N) CASE(OP_N, OP_T) CASE(OP_T, OP_N) CASE(OP_T, OP_T)
 #undef CASE
+#undef LV2K
 #undef LV2
   return 0;
 }
@@ -585,7 +596,7 @@ int launch_v2_cplx(const GemmDesc* descs, int ndesc, const T* A, const T* B,
   }();
   static const int depth = [] {
     const char* v = getenv("DLAF_GEMM_V2_DEPTH");
-    const int d = v ? atoi(v) : 3;
+    const int d = v ? atoi(v) : 2;
     return d < 2 ? 2 : (d > 4 ? 4 : d);
   }();
   if (!enabled) return 0;
