@@ -32,6 +32,33 @@ namespace {
 constexpr int BMAX = 64;      // wave size == max band width handled per lane
 constexpr int SPIN_LIMIT = 1 << 23;
 
+typedef int v4i_ __attribute__((ext_vector_type(4)));
+
+// Write-through (sc1) stores for the cross-WG band hand-off: the line is
+// written through to memory and DROPPED from this XCD's L2, so the publish
+// needs only a per-lane vmcnt drain + flag — no agent release fence writing
+// back the whole dirty L2 (~80 us/hop measured in round 1, the reason the
+// f64 GPU chase lost to the CPU wavefront; MI355X_MICROARCH.md
+// "publish-large": 3.0 vs 8.2 us per 64 KB). §5.7: end asm with s_nop 1.
+__device__ inline void store16_sc1(void* dst, v4i_ v) {
+  asm volatile("global_store_dwordx4 %0, %1, off sc1\n\ts_nop 1" ::"v"(
+                   (uint64_t)(uintptr_t)dst),
+               "v"(v)
+               : "memory");
+}
+__device__ inline void store8_sc1(void* dst, uint64_t v) {
+  asm volatile("global_store_dwordx2 %0, %1, off sc1\n\ts_nop 1" ::"v"(
+                   (uint64_t)(uintptr_t)dst),
+               "v"(v)
+               : "memory");
+}
+__device__ inline void store4_sc1(void* dst, uint32_t v) {
+  asm volatile("global_store_dword %0, %1, off sc1\n\ts_nop 1" ::"v"(
+                   (uint64_t)(uintptr_t)dst),
+               "v"(v)
+               : "memory");
+}
+
 template <typename T>
 __device__ inline typename ScalarTraits<T>::real_t sabs2(T v) {
   if constexpr (sizeof(T) == sizeof(typename ScalarTraits<T>::real_t))
@@ -214,15 +241,17 @@ __global__ __launch_bounds__(64) void chase_gpu_k(
     __syncthreads();
     return ok_s != 0;
   };
-  // producer side: release everything this WG stored, then publish the flag
+  // producer side: the window write-back uses write-through (sc1) stores,
+  // so publishing is one per-lane vmcnt drain + barrier + relaxed flag.
+  // (Valid form per MI355X_MICROARCH.md: sc1 payload -> asm vmcnt(0) ->
+  // flag; the only plain cross-WG stores left are unit 0's own start
+  // column, which no later sweep reads before kernel end.)
   auto publish = [&](int64_t s, int32_t val) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
-    if (lane == 0) {
-      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    if (lane == 0)
       __hip_atomic_store(&done[s], val, __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_AGENT);
-    }
     __syncthreads();
   };
 
@@ -262,7 +291,19 @@ __global__ __launch_bounds__(64) void chase_gpu_k(
         const int dmax0 = nn + (m > 0 ? m : 0) + 1;
         for (int c = 0; c < nn; ++c) {
           const int dm = min((int)ld, dmax0 - c);
-          for (int d = lane; d < dm; d += 64) W[c * S + d] = src[c * (int)ld + d];
+          const T* sc_ = src + c * (int64_t)ld;
+          T* wc = W + c * S;
+          if constexpr (sizeof(T) >= 8) {
+            // ld = 2b and S = ld + 2 keep both sides 16-B aligned
+            constexpr int EPV = 16 / sizeof(T);
+            const int nfull = dm / EPV;
+            for (int t = lane; t < nfull; t += 64)
+              reinterpret_cast<v4i_*>(wc)[t] =
+                  reinterpret_cast<const v4i_*>(sc_)[t];
+            for (int d = nfull * EPV + lane; d < dm; d += 64) wc[d] = sc_[d];
+          } else {
+            for (int d = lane; d < dm; d += 64) wc[d] = sc_[d];
+          }
         }
       }
       __syncthreads();
@@ -295,7 +336,20 @@ __global__ __launch_bounds__(64) void chase_gpu_k(
         const int dmax0 = nn + (m > 0 ? m : 0) + 1;
         for (int c = 0; c < nn; ++c) {
           const int dm = min((int)ld, dmax0 - c);
-          for (int d = lane; d < dm; d += 64) dst[c * (int)ld + d] = W[c * S + d];
+          T* dc = dst + c * (int64_t)ld;
+          const T* wc = W + c * S;
+          if constexpr (sizeof(T) >= 8) {
+            constexpr int EPV = 16 / sizeof(T);
+            const int nfull = dm / EPV;
+            for (int t = lane; t < nfull; t += 64)
+              store16_sc1(dc + t * EPV,
+                          reinterpret_cast<const v4i_*>(wc)[t]);
+            for (int d = nfull * EPV + lane; d < dm; d += 64)
+              store8_sc1(dc + d, *reinterpret_cast<const uint64_t*>(&wc[d]));
+          } else {
+            for (int d = lane; d < dm; d += 64)
+              store4_sc1(dc + d, *reinterpret_cast<const uint32_t*>(&wc[d]));
+          }
         }
       }
       if (last) break;
