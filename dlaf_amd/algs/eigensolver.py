@@ -68,8 +68,16 @@ def hermitian_eigensolver(
         return _eigh_direct(mat, g, eigenvalues_index_begin,
                             eigenvalues_index_end)
     if g is not None and g.distributed:
-        from .eigensolver_dist import hermitian_eigensolver_dist
-        return hermitian_eigensolver_dist(
+        import os
+        if os.environ.get("DLAF_DIST_EIG", "tiled") == "replicated":
+            # round-1 replicated-dense design, kept as a debug fallback
+            from .eigensolver_dist import hermitian_eigensolver_dist
+            return hermitian_eigensolver_dist(
+                uplo, mat, g, band,
+                eigenvalues_index_begin=eigenvalues_index_begin,
+                eigenvalues_index_end=eigenvalues_index_end)
+        from .eigensolver_tiled import hermitian_eigensolver_tiled
+        return hermitian_eigensolver_tiled(
             uplo, mat, g, band,
             eigenvalues_index_begin=eigenvalues_index_begin,
             eigenvalues_index_end=eigenvalues_index_end)

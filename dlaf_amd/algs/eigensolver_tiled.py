@@ -1,0 +1,667 @@
+"""Tiled distributed eigensolver: O(n^2/p) + O(n*band) working set per rank.
+
+Round-2 replacement for the replicated-dense distributed pipeline
+(``eigensolver_dist.py`` kept only as history): the Hermitian input stays in
+its 2D block-cyclic tiled ``Matrix`` throughout the first stage, and the
+eigenvector matrix lives as a 1D column stripe per rank through D&C and the
+back-transforms. No stage materializes an O(n^2) replicated tensor.
+
+Reference counterparts:
+* reduction_to_band (dist): ``eigensolver/reduction_to_band/impl.h:1150-1516``
+  — here the O(n*b) panel algebra (QR, T factor, W) is REPLICATED in rank
+  lockstep (one all-reduce assembles the panel; every rank derives identical
+  V/T/W), while the O(n^2 b) work — X = A22*W partial sums and the two-sided
+  trailing update — runs on each rank's OWN tiles only, with a single
+  O(n*b) all-reduce for X. On a 7-link xGMI node the two panel-sized
+  all-reduces per step replace the reference's col-reduce + row-reduce +
+  bcast chains; the trailing update is pure local GEMM.
+* tridiag D&C (dist): ``eigensolver/tridiag_solver/merge.h:1216-1938`` — the
+  eigenvector matrix is COLUMN-STRIPED; deflation rotations, permutations
+  and the secular eigenvector matrix are folded into one per-merge
+  coefficient matrix C (m x my_cols), so each merge is: replicated O(m)
+  scalar stage + row-block all-gathers of the old evecs + one local GEMM
+  per row block. Deflated columns become unit columns of C — no distributed
+  permutation step exists because columns are produced directly in final
+  sorted order.
+* band->tridiag: replicated bulge chase on the O(n*b) band (CPU wavefront /
+  GPU persistent kernel, ``band2tridiag.py``), assembled from tile owners.
+* back-transforms: per-panel V assembly (O(n*b) all-reduce) + local GEMMs on
+  the rank's column stripe (``bt_reduction_to_band/impl.h``).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..comm import collectives as coll
+from ..comm.grid import CommGrid
+from ..matrix.matrix import Matrix
+from ..types import UpLo
+from ..core import index as ix
+from .red2band import panel_qr_, t_factor
+from .tridiag_dc import _secular_roots, _leaf, _EPS
+
+__all__ = ["hermitian_eigensolver_tiled", "red2band_tiled", "dc_striped"]
+
+
+def _herm_full(t: torch.Tensor) -> torch.Tensor:
+    return torch.tril(t) + torch.tril(t, -1).mH
+
+
+# --------------------------------------------------------------------------
+# stage 1: tiled distributed reduction to band
+# --------------------------------------------------------------------------
+
+def _local_rows_ge(d, it0: int):
+    """Local tile-row indices whose global row tile >= it0, with globals."""
+    lr = d.local_nr_tiles[0]
+    out = []
+    for li in range(lr):
+        i = d.global_tile_of_local((li, 0))[0]
+        if i >= it0:
+            out.append((li, i))
+    return out
+
+
+def _local_cols_range(d, jt_lo: int, jt_hi: int):
+    """Local tile-col indices with jt_lo <= global < jt_hi, with globals."""
+    lc = d.local_nr_tiles[1]
+    out = []
+    for lj in range(lc):
+        j = d.global_tile_of_local((0, lj))[1]
+        if jt_lo <= j < jt_hi:
+            out.append((lj, j))
+    return out
+
+
+def red2band_tiled(mat: Matrix, band: int, grid: CommGrid):
+    """Distributed reduction to band over the tiled matrix (lower storage).
+
+    The band and the reflectors are left IN the distributed tiles
+    (LAPACK-style, below the R block of each panel); returns
+    {"taus", "panels", "band"} exactly like the local path.
+    """
+    d = mat.dist
+    n, nb = d.m, d.nb
+    assert d.m == d.n and d.mb == d.nb and nb % band == 0
+    dt, dev = mat.dtype, mat.device
+    nt = d.nr_tiles[0]
+    npad = nt * nb
+    group = grid.full_group
+    my_col = d.rank_col
+    taus_all, panels = [], []
+
+    st = mat.storage  # [lr, lc, nb, nb]
+
+    for j0 in range(0, max(n - band, 0), band):
+        r0 = j0 + band
+        bw = min(band, n - r0)
+        if bw <= 0:
+            break
+        jt0 = j0 // nb
+        joff = j0 % nb
+        it0 = r0 // nb
+        m_p = n - r0
+        nrefl = min(m_p, bw)
+        ct = d.rank_of_tile_col(jt0)
+
+        # ---- assemble the panel (padded global-row coords) ----
+        P = torch.zeros((npad, bw), dtype=dt, device=dev)
+        if my_col == ct:
+            lj = ix.local_tile_of_global(jt0, d.grid_cols)
+            for li, i in _local_rows_ge(d, it0):
+                rlo, rhi = max(i * nb, r0), min((i + 1) * nb, n)
+                if rlo < rhi:
+                    P[rlo:rhi] = st[li, lj, rlo - i * nb:rhi - i * nb,
+                                    joff:joff + bw]
+        coll.all_reduce_sum(P, group)
+        Pv = P[r0:n]
+        taus = torch.zeros(nrefl, dtype=dt, device=dev)
+        panel_qr_(Pv, taus)  # replicated, identical on every rank
+        if my_col == ct:
+            lj = ix.local_tile_of_global(jt0, d.grid_cols)
+            for li, i in _local_rows_ge(d, it0):
+                rlo, rhi = max(i * nb, r0), min((i + 1) * nb, n)
+                if rlo < rhi:
+                    st[li, lj, rlo - i * nb:rhi - i * nb,
+                       joff:joff + bw] = P[rlo:rhi]
+        panels.append((j0, bw, nrefl))
+        taus_all.append(taus)
+
+        V = torch.zeros((npad, nrefl), dtype=dt, device=dev)
+        V[r0:n] = torch.tril(Pv[:, :nrefl], -1)
+        V[r0:n] += torch.eye(m_p, nrefl, dtype=dt, device=dev)
+        T = t_factor(V[r0:n], taus)
+
+        if bw < band:
+            # capped panel: remaining band columns [j0+bw, j0+band) get the
+            # LEFT factor Q^H only (outside the trailing two-sided update)
+            c1, c2 = j0 + bw, min(j0 + band, n)
+            if c2 > c1:
+                B = torch.zeros((npad, c2 - c1), dtype=dt, device=dev)
+                if my_col == ct:
+                    lj = ix.local_tile_of_global(jt0, d.grid_cols)
+                    for li, i in _local_rows_ge(d, it0):
+                        rlo, rhi = max(i * nb, r0), min((i + 1) * nb, n)
+                        if rlo < rhi:
+                            B[rlo:rhi] = st[li, lj, rlo - i * nb:rhi - i * nb,
+                                            c1 - j0 + joff:c2 - j0 + joff]
+                coll.all_reduce_sum(B, group)
+                Wb2 = T.mH @ (V[r0:n].mH @ B[r0:n])
+                B[r0:n].addmm_(V[r0:n], Wb2, beta=1, alpha=-1)
+                if my_col == ct:
+                    lj = ix.local_tile_of_global(jt0, d.grid_cols)
+                    for li, i in _local_rows_ge(d, it0):
+                        rlo, rhi = max(i * nb, r0), min((i + 1) * nb, n)
+                        if rlo < rhi:
+                            st[li, lj, rlo - i * nb:rhi - i * nb,
+                               c1 - j0 + joff:c2 - j0 + joff] = B[rlo:rhi]
+
+        # ---- X = A22 * W from local tiles, one O(n*b) all-reduce ----
+        W = torch.zeros((npad, nrefl), dtype=dt, device=dev)
+        W[r0:n] = V[r0:n] @ T
+        Wb = W.view(nt, nb, nrefl)
+        X = torch.zeros((npad, nrefl), dtype=dt, device=dev)
+        Xb = X.view(nt, nb, nrefl)
+        for li, i in _local_rows_ge(d, it0):
+            cols = _local_cols_range(d, it0, i)  # strict lower tiles
+            if cols:
+                ljs = [lj for lj, _ in cols]
+                jgs = torch.tensor([j for _, j in cols], device=dev)
+                At = st[li, ljs[0]:ljs[-1] + 1]          # (m, nb, nb) view
+                Xb[i] += torch.bmm(At, Wb[jgs]).sum(0)
+                Xb.index_add_(0, jgs, torch.matmul(At.mH, Wb[i]))
+            # diagonal tile (i, i) if local
+            if d.rank_of_tile_col(i) == my_col:
+                lj = ix.local_tile_of_global(i, d.grid_cols)
+                Ad = _herm_full(st[li, lj])
+                Xb[i] += Ad @ Wb[i]
+        X[:r0] = 0
+        X[n:] = 0
+        coll.all_reduce_sum(X, group)
+
+        S = T.mH @ (V[r0:n].mH @ X[r0:n])
+        X[r0:n] -= 0.5 * (V[r0:n] @ S)
+
+        # ---- trailing two-sided update on local tiles only ----
+        Vb = V.view(nt, nb, nrefl)
+        for li, i in _local_rows_ge(d, it0):
+            cols = _local_cols_range(d, it0, i + 1)  # incl. diagonal
+            if not cols:
+                continue
+            ljs = [lj for lj, _ in cols]
+            jgs = torch.tensor([j for _, j in cols], device=dev)
+            At = st[li, ljs[0]:ljs[-1] + 1]
+            At -= torch.matmul(Vb[i], Xb[jgs].mH)
+            At -= torch.matmul(Xb[i], Vb[jgs].mH)
+    return {"taus": taus_all, "panels": panels, "band": band}
+
+
+def extract_band_tiled(mat: Matrix, band: int, grid: CommGrid) -> torch.Tensor:
+    """Band store [n, 2*band] (band[j][d] = A[j+d, j]) from tile owners."""
+    d = mat.dist
+    n, nb = d.m, d.nb
+    dt, dev = mat.dtype, mat.device
+    store = torch.zeros((n, 2 * band), dtype=dt, device=dev)
+    lr, lc = d.local_nr_tiles
+    for li in range(lr):
+        i = d.global_tile_of_local((li, 0))[0]
+        for lj in range(lc):
+            j = d.global_tile_of_local((0, lj))[1]
+            if j != i and j != i - 1:
+                continue
+            t = mat.tile((i, j))
+            r_lo, c_lo = i * nb, j * nb
+            tsr, tsc = t.shape
+            for dep in range(0, min(band, n - 1) + 1):
+                # elements (g, g-dep) with g in tile rows, g-dep in tile cols
+                g_lo = max(r_lo, c_lo + dep)
+                g_hi = min(r_lo + tsr, c_lo + tsc + dep, n)
+                if g_lo >= g_hi:
+                    continue
+                rows = torch.arange(g_lo, g_hi, device=dev)
+                store[rows - dep, dep] = t[rows - r_lo, rows - dep - c_lo]
+    coll.all_reduce_sum(store, grid.full_group)
+    return store
+
+
+# --------------------------------------------------------------------------
+# stage 3: column-striped distributed D&C
+# --------------------------------------------------------------------------
+
+def _stripe_bounds(n: int, rank: int, world: int) -> Tuple[int, int]:
+    return (n * rank) // world, (n * (rank + 1)) // world
+
+
+def _zhat_chunked(dk: torch.Tensor, sidx: torch.Tensor, mu: torch.Tensor,
+                  zk: torch.Tensor, rho: float, chunk: int = 4096
+                  ) -> torch.Tensor:
+    """Gu/Eisenstat z-hat with O(k*chunk) memory (no k x k temporaries)."""
+    k = dk.shape[0]
+    shift = dk[sidx]
+    num = torch.zeros(k, dtype=dk.dtype, device=dk.device)
+    den = torch.zeros(k, dtype=dk.dtype, device=dk.device)
+    for j0 in range(0, k, chunk):
+        j1 = min(j0 + chunk, k)
+        # lam_j - d_i = (shift_j - d_i) + mu_j, exact near poles
+        delta = (shift[j0:j1].unsqueeze(0) - dk.unsqueeze(1)) \
+            + mu[j0:j1].unsqueeze(0)
+        num += delta.abs().clamp_min(1e-300).log().sum(1)
+        dd = dk[j0:j1].unsqueeze(0) - dk.unsqueeze(1)
+        m = dd.abs().clamp_min(1e-300).log()
+        # exclude the diagonal j == i
+        idx = torch.arange(j0, j1, device=dk.device)
+        m[idx, idx - j0] = 0.0
+        den += m.sum(1)
+    zh = torch.exp(0.5 * (num - den - math.log(rho)))
+    return torch.where(zk < 0, -zh, zh)
+
+
+def dc_striped(d: torch.Tensor, e: torch.Tensor, group, rank: int, world: int,
+               device, leaf: Optional[int] = None,
+               row_block: int = 4096) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Distributed Cuppen D&C with column-striped eigenvectors.
+
+    Every rank executes the same recursion; O(n) scalar state (evals, z,
+    deflation) is replicated in lockstep, eigenvectors live as the rank's
+    column stripe [c0, c1). Per merge of range [lo, hi): the rotations,
+    deflation permutation and secular eigenvector columns are folded into
+    one coefficient matrix C (m x my-cols, built in Qb column coordinates),
+    and E[:, mine] = E_old[:, lo:hi] @ C is computed in row blocks gathered
+    from the stripes (in-place per block: rows are disjoint).
+
+    Returns (w [n] replicated, E_stripe [n, c1-c0]).
+    """
+    import os
+    if leaf is None:
+        leaf = int(os.environ.get("DLAF_DC_LEAF", "64"))
+    if not isinstance(device, torch.device):
+        device = torch.device(device)
+    dn = d.detach().cpu().numpy().astype(np.float64).copy()
+    en = e.detach().cpu().numpy().astype(np.float64).copy()
+    n = len(dn)
+    c0, c1 = _stripe_bounds(n, rank, world)
+    nc = c1 - c0
+    E = torch.zeros((n, nc), dtype=torch.float64, device=device)
+    wg = np.zeros(n, dtype=np.float64)
+    if n == 0:
+        return torch.zeros(0, dtype=torch.float64, device=device), E
+    stripe_w = max((n * (r + 1)) // world - (n * r) // world
+                   for r in range(world))
+
+    from ..ops._ext import get_ext
+
+    def my_range(lo, hi):
+        return max(lo, c0), min(hi, c1)
+
+    def gather_rows(r_lo, r_hi, lo, hi):
+        """All-gather E[r_lo:r_hi, lo:hi] from the column stripes."""
+        m = hi - lo
+        rb = r_hi - r_lo
+        if world == 1:
+            a, b = my_range(lo, hi)
+            out = torch.zeros((rb, m), dtype=torch.float64, device=device)
+            if a < b:
+                out[:, a - lo:b - lo] = E[r_lo:r_hi, a - c0:b - c0]
+            return out
+        buf = torch.zeros((rb, stripe_w), dtype=torch.float64, device=device)
+        a, b = my_range(lo, hi)
+        if a < b:
+            buf[:, :b - a] = E[r_lo:r_hi, a - c0:b - c0]
+        gath = [torch.empty_like(buf) for _ in range(world)]
+        import torch.distributed as dist
+        dist.all_gather(gath, buf, group=group)
+        out = torch.zeros((rb, m), dtype=torch.float64, device=device)
+        for r in range(world):
+            ra, rb_ = _stripe_bounds(n, r, world)
+            a, b = max(lo, ra), min(hi, rb_)
+            if a < b:
+                out[:, a - lo:b - lo] = gath[r][:, :b - a]
+        return out
+
+    def solve(lo, hi):
+        m = hi - lo
+        if m <= leaf:
+            w, v = _leaf(dn[lo:hi].copy(), en[lo:hi - 1].copy())
+            wg[lo:hi] = w
+            a, b = my_range(lo, hi)
+            if a < b:
+                E[lo:hi, a - c0:b - c0] = \
+                    torch.from_numpy(v[:, a - lo:b - lo]).to(device)
+            return
+        mid = lo + m // 2
+        rho = float(en[mid - 1])
+        dn[mid - 1] -= rho
+        dn[mid] -= rho
+        solve(lo, mid)
+        solve(mid, hi)
+        merge(lo, mid, hi, rho)
+
+    def merge(lo, mid, hi, rho):
+        m = hi - lo
+        # ---- z = [last row of Q1, first row of Q2] / sqrt(2), replicated ----
+        z = torch.zeros(m, dtype=torch.float64, device=device)
+        a, b = my_range(lo, mid)
+        if a < b:
+            z[a - lo:b - lo] = E[mid - 1, a - c0:b - c0]
+        a, b = my_range(mid, hi)
+        if a < b:
+            z[a - lo:b - lo] = E[mid, a - c0:b - c0]
+        if world > 1:
+            coll.all_reduce_sum(z, group)
+        dcur = torch.from_numpy(wg[lo:hi].copy()).to(device)
+        znorm2 = float(z @ z)
+        if znorm2 == 0 or rho == 0:
+            order = np.argsort(wg[lo:hi], kind="stable")
+            _apply_perm(lo, hi, order)
+            wg[lo:hi] = wg[lo:hi][order]
+            return
+        rho_eff = rho * znorm2
+        z = z / math.sqrt(znorm2)
+        negate = rho_eff < 0
+        if negate:
+            dcur = -dcur
+            rho_eff = -rho_eff
+        d_s, perm = torch.sort(dcur)
+        z_s = z[perm.to(device)]
+
+        # ---- deflation scan (replicated, host) ----
+        dn_s = d_s.cpu().numpy().astype(np.float64).copy()
+        zn_s = z_s.cpu().numpy().astype(np.float64).copy()
+        dmax = max(np.abs(dn_s).max(), rho_eff) if m else 1.0
+        tol = 8.0 * _EPS * max(dmax, 1e-300)
+        dt_ = torch.from_numpy(dn_s)
+        zt_ = torch.from_numpy(zn_s)
+        deflated_t = torch.zeros(m, dtype=torch.bool)
+        rots_t = torch.zeros((max(m, 1), 4), dtype=torch.float64)
+        nrot = get_ext().dc_deflate_scan(dt_, zt_, float(rho_eff), float(tol),
+                                         deflated_t, rots_t)
+        deflated = deflated_t.numpy()
+        rots = rots_t[:nrot]
+        nd_idx = np.nonzero(~deflated)[0]
+        df_idx = np.nonzero(deflated)[0]
+        k1 = len(nd_idx)
+
+        # ---- secular roots + new eigenvalues (replicated) ----
+        if k1 > 0:
+            sec_dev = device if (device.type == "cuda" or m >= 192) \
+                else torch.device("cpu")
+            dk = torch.from_numpy(dn_s[nd_idx]).to(sec_dev)
+            zk = torch.from_numpy(zn_s[nd_idx]).to(sec_dev)
+            sidx, mu = _secular_roots(dk, zk, rho_eff)
+            lam = (dk[sidx] + mu).cpu().numpy()
+        else:
+            lam = np.zeros(0)
+        all_vals = np.concatenate([lam, dn_s[df_idx]])
+        if negate:
+            all_vals = -all_vals
+        order = np.argsort(all_vals, kind="stable")
+        wg[lo:hi] = all_vals[order]
+
+        # ---- coefficient matrix C for MY final columns ----
+        a_my, b_my = my_range(lo, hi)
+        ncm = max(b_my - a_my, 0)
+        C = torch.zeros((m, ncm), dtype=torch.float64, device=device)
+        if ncm > 0:
+            srcs = order[a_my - lo:b_my - lo]  # matmul-order sources
+            nd_sel = srcs[srcs < k1]
+            nd_pos = np.nonzero(srcs < k1)[0]
+            df_sel = srcs[srcs >= k1] - k1
+            df_pos = np.nonzero(srcs >= k1)[0]
+            if len(nd_sel):
+                # U columns for my targets: zh_i / (d_i - lam_j), normalized
+                zh = _zhat_chunked(dk, sidx, mu, zk, rho_eff)
+                sj = torch.from_numpy(nd_sel).to(sec_dev)
+                delta = (dk.unsqueeze(1) - dk[sidx[sj]].unsqueeze(0)) \
+                    - mu[sj].unsqueeze(0)
+                U = zh.unsqueeze(1) / delta
+                U = U / torch.linalg.vector_norm(U, dim=0, keepdim=True)
+                # scatter U rows (sorted positions nd_idx) into C
+                rows = torch.from_numpy(nd_idx).to(device)
+                C[rows.unsqueeze(1),
+                  torch.from_numpy(nd_pos).to(device).unsqueeze(0)] = \
+                    U.to(device)
+            if len(df_sel):
+                rows = torch.from_numpy(df_idx[df_sel]).to(device)
+                cols = torch.from_numpy(df_pos).to(device)
+                C[rows, cols] = 1.0
+            # fold the Givens rotations in (reverse order, row ops in the
+            # sorted index space): E_new = Qb G1..Gt Chat
+            for r_i in range(nrot - 1, -1, -1):
+                i_r = int(rots[r_i, 0])
+                j_r = int(rots[r_i, 1])
+                c_r = float(rots[r_i, 2])
+                s_r = float(rots[r_i, 3])
+                ri = C[i_r].clone()
+                rj = C[j_r].clone()
+                C[i_r] = c_r * ri - s_r * rj
+                C[j_r] = s_r * ri + c_r * rj
+            # map sorted index space -> Qb column space
+            Cq = torch.zeros_like(C)
+            Cq[perm.to(device)] = C
+            C = Cq
+
+        # ---- row-block streamed GEMM, in place per block ----
+        for r_lo in range(lo, hi, row_block):
+            r_hi = min(r_lo + row_block, hi)
+            G = gather_rows(r_lo, r_hi, lo, hi)
+            if ncm > 0:
+                E[r_lo:r_hi, a_my - c0:b_my - c0] = G @ C
+
+    def _apply_perm(lo, hi, order):
+        """Column permutation for the rho==0 degenerate merge."""
+        for r_lo in range(lo, hi, row_block):
+            r_hi = min(r_lo + row_block, hi)
+            G = gather_rows(r_lo, r_hi, lo, hi)
+            a, b = my_range(lo, hi)
+            if a < b:
+                sel = torch.from_numpy(order[a - lo:b - lo]).to(device)
+                E[r_lo:r_hi, a - c0:b - c0] = G[:, sel]
+
+    solve(0, n)
+    return torch.from_numpy(wg).to(device), E
+
+
+# --------------------------------------------------------------------------
+# back-transform stage 1 (red2band reflectors) on a column stripe
+# --------------------------------------------------------------------------
+
+def bt_red2band_tiled(E: torch.Tensor, mat: Matrix, refl, grid: CommGrid
+                      ) -> None:
+    """E_stripe <- Q E_stripe; V panels assembled from tile owners."""
+    d = mat.dist
+    n, nb = d.m, d.nb
+    band = refl["band"]
+    dt, dev = mat.dtype, E.device
+    st = mat.storage
+    my_col = d.rank_col
+    for (j0, bw, nrefl), taus in zip(reversed(refl["panels"]),
+                                     reversed(refl["taus"])):
+        r0 = j0 + band
+        m_p = n - r0
+        jt0 = j0 // nb
+        joff = j0 % nb
+        it0 = r0 // nb
+        P = torch.zeros((m_p, nrefl), dtype=dt, device=dev)
+        if my_col == d.rank_of_tile_col(jt0):
+            lj = ix.local_tile_of_global(jt0, d.grid_cols)
+            for li, i in _local_rows_ge(d, it0):
+                rlo, rhi = max(i * nb, r0), min((i + 1) * nb, n)
+                if rlo < rhi:
+                    P[rlo - r0:rhi - r0] = st[li, lj,
+                                              rlo - i * nb:rhi - i * nb,
+                                              joff:joff + nrefl]
+        coll.all_reduce_sum(P, grid.full_group)
+        V = torch.tril(P, -1) + torch.eye(m_p, nrefl, dtype=dt, device=dev)
+        T = t_factor(V, taus)
+        W = T @ (V.mH @ E[r0:, :])
+        E[r0:, :].addmm_(V, W, beta=1, alpha=-1)
+
+
+# --------------------------------------------------------------------------
+# stripe -> 2D block-cyclic redistribution (packed pairwise p2p)
+# --------------------------------------------------------------------------
+
+def stripe_to_matrix(E: torch.Tensor, col_ranges, out: Matrix,
+                     grid: CommGrid) -> None:
+    """Scatter the caller's column stripe E into the 2D block-cyclic ``out``
+    Matrix via packed pairwise isend/irecv (the packed-chunk exchange of
+    ``permutations/general/impl.h:303-321``, gloo- and RCCL-compatible).
+
+    ``col_ranges[r] = (lo_r, hi_r)``: the OUTPUT-space column range held by
+    rank r; E holds this rank's columns [lo_me, hi_me)."""
+    import torch.distributed as dist
+
+    d = out.dist
+    n, nE = d.m, d.n
+    nb = d.nb
+    world = grid.world_size
+    me = grid.rank
+    lo_me, hi_me = col_ranges[me]
+
+    def rank_coords(r):
+        return r // grid.grid_cols, r % grid.grid_cols
+
+    def rows_of(pr):
+        idx = []
+        for i in range(d.nr_tiles[0]):
+            if d.rank_of_tile_row(i) == pr:
+                idx.append(torch.arange(i * nb, min((i + 1) * nb, n)))
+        return (torch.cat(idx) if idx
+                else torch.zeros(0, dtype=torch.int64))
+
+    def cols_of(pc, lo, hi):
+        idx = []
+        for j in range(d.nr_tiles[1]):
+            if d.rank_of_tile_col(j) == pc:
+                a, b = max(j * nb, lo), min((j + 1) * nb, nE, hi)
+                if a < b:
+                    idx.append(torch.arange(a, b))
+        return (torch.cat(idx) if idx
+                else torch.zeros(0, dtype=torch.int64))
+
+    my_pr, my_pc = rank_coords(me)
+
+    def write_block(buf, rows, cols):
+        if rows.numel() == 0 or cols.numel() == 0:
+            return
+        rn = rows.cpu().numpy()
+        cn = cols.cpu().numpy()
+        rt = rn // nb
+        ct = cn // nb
+        for i in np.unique(rt):
+            rsel = np.nonzero(rt == i)[0]
+            li = ix.local_tile_of_global(int(i), d.grid_rows)
+            for j in np.unique(ct):
+                csel = np.nonzero(ct == j)[0]
+                lj = ix.local_tile_of_global(int(j), d.grid_cols)
+                blk = buf[torch.from_numpy(rsel).to(buf.device)][:,
+                          torch.from_numpy(csel).to(buf.device)]
+                rr = torch.from_numpy(rn[rsel] - i * nb).to(buf.device)
+                cc = torch.from_numpy(cn[csel] - j * nb).to(buf.device)
+                out.storage[li, lj][rr.unsqueeze(1), cc.unsqueeze(0)] = \
+                    blk.to(out.storage.dtype)
+
+    if world == 1:
+        write_block(E, torch.arange(n), torch.arange(lo_me, hi_me))
+        return
+
+    rows_cache = {pr: rows_of(pr) for pr in range(grid.grid_rows)}
+    my_rows = rows_cache[my_pr]
+    rows_dev = {pr: r.to(E.device) for pr, r in rows_cache.items()}
+
+    for delta in range(world):
+        dst = (me + delta) % world
+        src = (me - delta) % world
+        # send block: my stripe cols owned by dst's rank-col, dst's rows
+        dpr, dpc = rank_coords(dst)
+        scols = cols_of(dpc, lo_me, hi_me)
+        sbuf = (E[rows_dev[dpr]][:, (scols - lo_me).to(E.device)].contiguous()
+                if scols.numel() and rows_cache[dpr].numel()
+                else torch.zeros(0, dtype=E.dtype, device=E.device))
+        if delta == 0:
+            write_block(sbuf, rows_cache[my_pr], scols)
+            continue
+        # recv block: src's stripe cols owned by my rank-col, my rows
+        slo, shi = col_ranges[src]
+        rcols = cols_of(my_pc, slo, shi)
+        rbuf = torch.zeros((my_rows.numel(), rcols.numel()),
+                           dtype=E.dtype, device=E.device)
+        reqs = []
+        if rbuf.numel():
+            reqs.append(dist.irecv(_cv(rbuf), src=src, group=grid.full_group))
+        if sbuf.numel():
+            reqs.append(dist.isend(_cv(sbuf), dst=dst, group=grid.full_group))
+        for r in reqs:
+            r.wait()
+        if rbuf.numel():
+            write_block(rbuf, my_rows, rcols)
+
+
+def _cv(t: torch.Tensor) -> torch.Tensor:
+    return torch.view_as_real(t) if t.is_complex() else t
+
+
+# --------------------------------------------------------------------------
+# full pipeline
+# --------------------------------------------------------------------------
+
+def hermitian_eigensolver_tiled(uplo: UpLo, mat: Matrix, grid: CommGrid,
+                                band: int,
+                                eigenvalues_index_begin: int = 0,
+                                eigenvalues_index_end: Optional[int] = None
+                                ) -> Tuple[torch.Tensor, Matrix]:
+    """Distributed HEEV with tiled stage 1 and striped eigenvectors.
+
+    Per-rank working set: the rank's tiles O(n^2/p) + panels O(n*band) +
+    its eigenvector stripe O(n^2/p). ``mat`` is overwritten with the band
+    and the reflectors (consistent across ranks by construction).
+    """
+    assert uplo == UpLo.Lower
+    d = mat.dist
+    n = d.m
+    dev = mat.device
+    world = grid.world_size
+    rank = grid.rank
+    group = grid.full_group
+    ib = eigenvalues_index_begin
+    ie = n if eigenvalues_index_end is None else eigenvalues_index_end
+    nE = ie - ib
+
+    refl = red2band_tiled(mat, band, grid)
+
+    from .band2tridiag import chase_band
+    tri = chase_band(extract_band_tiled(mat, band, grid), band)
+
+    import os
+    rb = int(os.environ.get("DLAF_DC_ROW_BLOCK", "4096"))
+    w_all, E = dc_striped(tri.d, tri.e, group, rank, world, dev, row_block=rb)
+    c0, c1 = _stripe_bounds(n, rank, world)
+    w = w_all[ib:ie].clone()
+
+    # restrict to the requested spectrum slice (stripe-local)
+    a, b = max(c0, ib), min(c1, ie)
+    E_sub = E[:, a - c0:b - c0].to(mat.dtype).contiguous()
+    del E
+
+    from .band2tridiag import bt_band_to_tridiagonal
+    if E_sub.shape[1]:
+        bt_band_to_tridiagonal(E_sub, tri)
+    bt_red2band_tiled(E_sub, mat, refl, grid)
+
+    evecs = Matrix.create(n, max(nE, 1), d.mb, d.nb, dtype=mat.dtype,
+                          device=dev, grid=grid)
+    if nE:
+        # output-space column ranges per rank (stripes of [0,n) cut to the
+        # requested slice, shifted by ib)
+        ranges = []
+        for r in range(world):
+            ra, rb = _stripe_bounds(n, r, world)
+            ranges.append((max(ra, ib) - ib, max(min(rb, ie) - ib,
+                                                 max(ra, ib) - ib)))
+        stripe_to_matrix(E_sub, ranges, evecs, grid)
+    return w, evecs
