@@ -62,9 +62,16 @@ def hermitian_eigensolver(
     if band is None:
         band = get_band_size(d.nb)
     band = max(1, min(band, max(n - 1, 1)))
-    if n <= max(band, d.nb) or d.nb % band != 0:
-        # tiny problem (single tile, or no valid band divisor): direct dense
-        # eigendecomposition — the two-stage pipeline requires nb % band == 0
+    distributed = g is not None and g.distributed
+    if distributed and d.nb % band != 0:
+        # the tiled distributed panels must stay inside one tile column;
+        # snap to the largest divisor of nb not above the requested band
+        # (reference: get_band_size.h guarantees divisibility by
+        # construction — this keeps arbitrary user bands off a dense cliff)
+        band = max(bb for bb in range(1, band + 1) if d.nb % bb == 0)
+    if n <= max(band, d.nb):
+        # tiny problem (at or below one tile/band): direct dense
+        # eigendecomposition
         return _eigh_direct(mat, g, eigenvalues_index_begin,
                             eigenvalues_index_end)
     if g is not None and g.distributed:
