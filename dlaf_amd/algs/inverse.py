@@ -59,6 +59,10 @@ def triangular_inverse(uplo: UpLo, diag: Diag, mat: Matrix,
     assert d.m == d.n and d.mb == d.nb
     g = _trivial_grid(grid if grid is not None else mat.grid)
     unit = diag == Diag.Unit
+    if (mat.device.type == "cuda" and not g.distributed
+            and d.nr_tiles[0] > 2):
+        _trtri_local_gpu(mat, unit)
+        return
     nt = d.nr_tiles[0]
     nb = d.nb
     dev, dt = mat.device, mat.dtype
@@ -149,6 +153,113 @@ def triangular_inverse(uplo: UpLo, diag: Diag, mat: Matrix,
             kr = d.rank_of_tile_row(k)
             if d.rank_row == kr:
                 mat.tile((k, k)).copy_(inv)
+
+    if unit:
+        for k, dg in saved_diag.items():
+            mat.tile((k, k)).diagonal().copy_(dg)
+
+
+def _trtri_local_gpu(mat: Matrix, unit: bool) -> None:
+    """Single-GPU TRTRI with the POTRF/TRSM-style two-stream lookahead.
+
+    Descending column loop X[:,k] = -inv(L_kk) ... with the k-sum split so
+    the O(nt^2) "partial" accumulation of step k (tiles j >= k+2, available
+    two steps early) runs on the update stream while the critical chain
+    (the j = k+1 fix-up + the column write) stays on the high-priority
+    stream — the round-1 backlog's sequential acc-chain is now one small
+    GEMM per step. Round-1: 23.9 TF at N=16384; reference counterpart
+    ``inverse/triangular/impl.h:183-549``.
+    """
+    from ..runtime.streams import get_runtime
+
+    d = mat.dist
+    nt = d.nr_tiles[0]
+    nb = d.nb
+    dev, dt = mat.device, mat.dtype
+    ts = nb * nb
+    st = mat.storage  # [nt, nt, nb, nb]
+
+    saved_diag = {}
+    if unit:
+        for k in range(nt):
+            saved_diag[k] = mat.tile((k, k)).diagonal().clone()
+
+    # hoisted diagonal-block inversions (never touched before their step);
+    # full padded tiles (identity pad) so every desc is nb-uniform
+    own_tiles = [st[k, k].clone() for k in range(nt)]
+    invs = ops.tri_inverse_full_many(own_tiles, lower=True, unit=unit)
+    inv_stack = torch.stack(invs)  # [nt, nb, nb]
+
+    acc = torch.zeros((2, nt, nb, nb), dtype=dt, device=dev)
+
+    rt = get_runtime(dev)
+    sp, su = rt.hp_streams[0], rt.np_streams[0]
+    cur = torch.cuda.current_stream(dev)
+    sp.wait_stream(cur)
+    su.wait_stream(cur)
+    ev_write = [None] * (nt + 1)
+    ev_partial = [None] * (nt + 1)
+
+    def tile_off(i, j):
+        return (i * nt + j) * ts
+
+    # su: partial_k = sum_{j >= k+2} X[:, j] L[j, k]  (one desc per row i,
+    # ktiles accumulation along the row; B walks DOWN column k of L)
+    def launch_partial(k, buf):
+        rows = []
+        for i in range(k + 2, nt):
+            cnt = i - (k + 1)
+            rows.append([buf * nt * ts + i * ts, tile_off(i, k + 2),
+                         tile_off(k + 2, k), cnt, ts, nt * ts])
+        if rows:
+            ops.gemm_fused(acc.view(-1), st.view(-1), st.view(-1),
+                           np.array(rows, dtype=np.int64), nb, nb, nb,
+                           nb, nb, nb, Op.NoTrans, Op.NoTrans, 1.0, 0.0)
+        e = torch.cuda.Event()
+        e.record(su)
+        return e
+
+    for k in range(nt - 1, -1, -1):
+        buf = k & 1
+        # partial for this step was launched two iterations ago; launch the
+        # next partial (k-2) as soon as this step's column write ordering
+        # allows reading columns >= k (ev_write[k] below on sp; partial k-2
+        # reads cols >= k — but column k is written THIS step, so it must
+        # wait for that write; chain via events)
+        with torch.cuda.stream(sp):
+            if ev_partial[k] is not None:
+                sp.wait_event(ev_partial[k])
+            n_rows = nt - (k + 1)
+            if n_rows > 0:
+                # partial_k covered rows >= k+2 (beta 0); row k+1 starts clean
+                acc[buf, k + 1].zero_()
+                # fix-up: acc += X[:, k+1] L[k+1, k] (column k+1 final)
+                fix = [[buf * nt * ts + i * ts, tile_off(i, k + 1),
+                        tile_off(k + 1, k), 1, 0, 0]
+                       for i in range(k + 1, nt)]
+                ops.gemm_fused(acc.view(-1), st.view(-1), st.view(-1),
+                               np.array(fix, dtype=np.int64), nb, nb, nb,
+                               nb, nb, nb, Op.NoTrans, Op.NoTrans, 1.0, 1.0)
+                # column write: X[:, k] = -acc @ inv_k
+                wr = [[tile_off(i, k), buf * nt * ts + i * ts, k * ts, 1, 0, 0]
+                      for i in range(k + 1, nt)]
+                ops.gemm_fused(st.view(-1), acc.view(-1), inv_stack.view(-1),
+                               np.array(wr, dtype=np.int64), nb, nb, nb,
+                               nb, nb, nb, Op.NoTrans, Op.NoTrans, -1.0, 0.0)
+            t = mat.tile((k, k))
+            t.copy_(inv_stack[k][:t.shape[0], :t.shape[1]])
+            e = torch.cuda.Event()
+            e.record(sp)
+            ev_write[k] = e
+        if k - 2 >= 0:
+            with torch.cuda.stream(su):
+                su.wait_event(ev_write[k])          # partial reads col k
+                if k < nt - 1 and ev_write[k + 1] is not None:
+                    su.wait_event(ev_write[k + 1])
+                ev_partial[k - 2] = launch_partial(k - 2, (k - 2) & 1)
+
+    cur.wait_stream(sp)
+    cur.wait_stream(su)
 
     if unit:
         for k, dg in saved_diag.items():
