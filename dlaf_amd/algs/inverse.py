@@ -190,7 +190,14 @@ def _trtri_local_gpu(mat: Matrix, unit: bool) -> None:
     invs = ops.tri_inverse_full_many(own_tiles, lower=True, unit=unit)
     inv_stack = torch.stack(invs)  # [nt, nb, nb]
 
-    acc = torch.zeros((2, nt, nb, nb), dtype=dt, device=dev)
+    # K-split: a row's k-sum is chunked over P accumulator planes so the
+    # fused launch is load-balanced (one desc per (row, chunk) instead of
+    # one triangular-length chain per row — the chain made every launch as
+    # long as its LONGEST row and capped round-1 TRTRI at ~23 TF)
+    kchunk = 4
+    P = max(1, -(-(nt - 2) // kchunk))
+    acc = torch.zeros((2, P, nt, nb, nb), dtype=dt, device=dev)
+    accsum = torch.zeros((nt, nb, nb), dtype=dt, device=dev)
 
     rt = get_runtime(dev)
     sp, su = rt.hp_streams[0], rt.np_streams[0]
@@ -203,14 +210,18 @@ def _trtri_local_gpu(mat: Matrix, unit: bool) -> None:
     def tile_off(i, j):
         return (i * nt + j) * ts
 
-    # su: partial_k = sum_{j >= k+2} X[:, j] L[j, k]  (one desc per row i,
-    # ktiles accumulation along the row; B walks DOWN column k of L)
+    # su: partial_k = sum_{j >= k+2} X[:, j] L[j, k], K-split into planes:
+    # desc per (row i, chunk c) -> acc[buf, c, i] (distinct C blocks)
     def launch_partial(k, buf):
         rows = []
         for i in range(k + 2, nt):
             cnt = i - (k + 1)
-            rows.append([buf * nt * ts + i * ts, tile_off(i, k + 2),
-                         tile_off(k + 2, k), cnt, ts, nt * ts])
+            for c in range(0, cnt, kchunk):
+                cc = min(kchunk, cnt - c)
+                rows.append([(buf * P + c // kchunk) * nt * ts + i * ts,
+                             tile_off(i, k + 2 + c),
+                             tile_off(k + 2 + c, k), cc, ts, nt * ts])
+        acc[buf].zero_()  # short rows leave high planes untouched
         if rows:
             ops.gemm_fused(acc.view(-1), st.view(-1), st.view(-1),
                            np.array(rows, dtype=np.int64), nb, nb, nb,
@@ -231,19 +242,18 @@ def _trtri_local_gpu(mat: Matrix, unit: bool) -> None:
                 sp.wait_event(ev_partial[k])
             n_rows = nt - (k + 1)
             if n_rows > 0:
-                # partial_k covered rows >= k+2 (beta 0); row k+1 starts clean
-                acc[buf, k + 1].zero_()
-                # fix-up: acc += X[:, k+1] L[k+1, k] (column k+1 final)
-                fix = [[buf * nt * ts + i * ts, tile_off(i, k + 1),
+                # collapse the K-split planes, then fix-up + column write
+                torch.sum(acc[buf], dim=0, out=accsum)
+                fix = [[i * ts, tile_off(i, k + 1),
                         tile_off(k + 1, k), 1, 0, 0]
                        for i in range(k + 1, nt)]
-                ops.gemm_fused(acc.view(-1), st.view(-1), st.view(-1),
+                ops.gemm_fused(accsum.view(-1), st.view(-1), st.view(-1),
                                np.array(fix, dtype=np.int64), nb, nb, nb,
                                nb, nb, nb, Op.NoTrans, Op.NoTrans, 1.0, 1.0)
                 # column write: X[:, k] = -acc @ inv_k
-                wr = [[tile_off(i, k), buf * nt * ts + i * ts, k * ts, 1, 0, 0]
+                wr = [[tile_off(i, k), i * ts, k * ts, 1, 0, 0]
                       for i in range(k + 1, nt)]
-                ops.gemm_fused(st.view(-1), acc.view(-1), inv_stack.view(-1),
+                ops.gemm_fused(st.view(-1), accsum.view(-1), inv_stack.view(-1),
                                np.array(wr, dtype=np.int64), nb, nb, nb,
                                nb, nb, nb, Op.NoTrans, Op.NoTrans, -1.0, 0.0)
             t = mat.tile((k, k))
