@@ -194,7 +194,8 @@ def _trtri_local_gpu(mat: Matrix, unit: bool) -> None:
     # fused launch is load-balanced (one desc per (row, chunk) instead of
     # one triangular-length chain per row — the chain made every launch as
     # long as its LONGEST row and capped round-1 TRTRI at ~23 TF)
-    kchunk = 4
+    import os
+    kchunk = int(os.environ.get("DLAF_TRTRI_KSPLIT", "4"))
     P = max(1, -(-(nt - 2) // kchunk))
     acc = torch.zeros((2, P, nt, nb, nb), dtype=dt, device=dev)
     accsum = torch.zeros((nt, nb, nb), dtype=dt, device=dev)

@@ -38,3 +38,18 @@ def test_redistribute_dist():
     errs = run_distributed(_worker, 2)
     for e in errs:
         assert e == 0.0
+
+
+def _worker4(rank, ws):
+    grid = CommGrid(2, 2)
+    m, n = 52, 36
+    src = Matrix.create(m, n, 8, 8, dtype=torch.complex128, grid=grid)
+    mutil.set_random(src, seed=5)
+    dst = Matrix.create(m, n, 12, 12, dtype=torch.complex128, grid=grid)
+    redistribute(src, dst)
+    return (src.to_global() - dst.to_global()).abs().max().item()
+
+
+def test_redistribute_dist_2x2_complex():
+    for e in run_distributed(_worker4, 4):
+        assert e == 0.0
