@@ -237,6 +237,147 @@ def _cholesky_local(mat: Matrix) -> None:
                                   Op.NoTrans, opc, -1.0, 1.0)
 
 
+# ---------------- native Upper (local) ----------------
+
+def _upper_plan(mat: Matrix) -> _DescTable:
+    d = mat.dist
+    key = ("chol_local_U", d.nr_tiles, d.nb, mat.dtype, str(mat.device))
+    plan = _PLAN_CACHE.get(key)
+    if plan is not None:
+        return plan
+    nt = d.nr_tiles[0]
+    nb = d.nb
+    bsz = ops.potrf_bsz(mat.dtype)
+    nblocks = (nb + bsz - 1) // bsz
+    table = _DescTable()
+    for k in range(nt):
+        cols = np.arange(k + 1, nt)
+        offs = np.array([mat.tile_offset((k, int(j))) for j in cols],
+                        dtype=np.int64)
+        if len(offs):
+            # row-panel left-solve per inner block d:
+            #   upd: X[c0:c0+bs, :] -= Td[c0:c0+bs, :c0] @ X[:c0, :]
+            #   apply: X[c0:c0+bs, :] = dinv[d] @ scratch[c0 rows]
+            for dd in range(nblocks):
+                c0 = dd * bsz
+                if dd > 0:
+                    table.add(("Uk", k, "upd", dd),
+                              _rows6(offs + c0 * nb,
+                                     np.full(len(offs), c0 * nb), offs))
+                table.add(("Uk", k, "apply", dd),
+                          _rows6(offs + c0 * nb,
+                                 np.zeros(len(offs), dtype=np.int64),
+                                 np.arange(len(offs), dtype=np.int64) * (bsz * nb)))
+
+            def _descs_for(i_range):
+                c, a, b = [], [], []
+                for i in i_range:
+                    for j in range(i, nt):
+                        c.append(mat.tile_offset((i, j)))
+                        a.append(mat.tile_offset((k, i)))
+                        b.append(mat.tile_offset((k, j)))
+                return _rows6(np.array(c, dtype=np.int64),
+                              np.array(a, dtype=np.int64),
+                              np.array(b, dtype=np.int64)) if c else None
+            h = _descs_for([k + 1])
+            if h is not None:
+                table.add(("Uk", k, "head"), h)
+            t = _descs_for(range(k + 2, nt))
+            if t is not None:
+                table.add(("Uk", k, "tail"), t)
+    table.upload(mat.device)
+    _PLAN_CACHE[key] = table
+    return table
+
+
+def _cholesky_local_upper(mat: Matrix) -> None:
+    """Native Upper factorization A = U^H U over upper-stored tiles (no
+    storage transpose; reference ``factorization/cholesky/impl.h:317``
+    call_U). The diagonal factor runs on a conj-transposed nb x nb scratch
+    (O(nb^2) per step); the row panel gets a LEFT blocked solve
+    U[k,j] = Lkk^{-1} A[k,j]; the trailing update is
+    A[i,j] -= A[k,i]^H A[k,j] with the same two-stream lookahead as Lower.
+    """
+    d = mat.dist
+    nt = d.nr_tiles[0]
+    nb = d.nb
+    gpu = mat.device.type == "cuda"
+    opc = _op_conj(mat.dtype)
+    st = mat.storage  # full padded tiles (identity diag pad)
+    if not gpu:
+        for k in range(nt):
+            Td = st[k, k].mH.contiguous()
+            ops.potrf_tile(Td, None)
+            st[k, k].copy_(torch.tril(Td).mH)
+            L = torch.tril(Td)
+            for j in range(k + 1, nt):
+                t = st[k, j]
+                t.copy_(torch.linalg.solve_triangular(L, t, upper=False))
+            for i in range(k + 1, nt):
+                for j in range(i, nt):
+                    ops.gemm_tile(st[i, j], st[k, i], st[k, j],
+                                  opc, Op.NoTrans, -1.0, 1.0)
+        return
+
+    table = _upper_plan(mat)
+    dinv = ops.dinv_workspace(nb, mat.dtype, mat.device)
+    bsz = dinv.shape[-1]
+    nblocks = (nb + bsz - 1) // bsz
+    Td = torch.empty((nb, nb), dtype=mat.dtype, device=mat.device)
+    scratch = torch.empty((nt, bsz, nb), dtype=mat.dtype, device=mat.device)
+    rt = get_runtime(mat.device)
+    sp, su = rt.hp_streams[0], rt.np_streams[0]
+    cur = torch.cuda.current_stream(mat.device)
+    sp.wait_stream(cur)
+    su.wait_stream(cur)
+    ev_tail = [None] * nt
+    for k in range(nt):
+        with torch.cuda.stream(sp):
+            if k >= 2 and ev_tail[k - 2] is not None:
+                sp.wait_event(ev_tail[k - 2])
+            Td.copy_(st[k, k].mH)
+            ops.potrf_tile(Td, dinv)
+            st[k, k].copy_(torch.tril(Td).mH)
+            # blocked LEFT solve of the row panel
+            m_cols = nt - (k + 1)
+            if m_cols > 0:
+                for dd in range(nblocks):
+                    c0 = dd * bsz
+                    bs = min(bsz, nb - c0)
+                    descs = table.get(("Uk", k, "upd", dd))
+                    if descs is not None and c0 > 0:
+                        ops.gemm_fused(st, Td, st, descs, bs, nb, c0,
+                                       nb, nb, nb, Op.NoTrans, Op.NoTrans,
+                                       -1.0, 1.0)
+                    # stage the block rows, then scratch-multiply back
+                    blk = st[k, k + 1:nt, c0:c0 + bs, :]
+                    scratch[: m_cols, :bs].copy_(blk)
+                    descs = table.get(("Uk", k, "apply", dd))
+                    if descs is not None:
+                        ops.gemm_fused(st, dinv[dd], scratch.view(-1), descs,
+                                       bs, nb, bs, bsz, nb, nb,
+                                       Op.NoTrans, Op.NoTrans, 1.0, 0.0)
+            ev_p = torch.cuda.Event()
+            ev_p.record(sp)
+            head = table.get(("Uk", k, "head"))
+            if head is not None:
+                if k >= 1 and ev_tail[k - 1] is not None:
+                    sp.wait_event(ev_tail[k - 1])
+                ops.gemm_fused(st, st, st, head, nb, nb, nb, nb, nb, nb,
+                               opc, Op.NoTrans, -1.0, 1.0)
+        tail = table.get(("Uk", k, "tail"))
+        if tail is not None:
+            with torch.cuda.stream(su):
+                su.wait_event(ev_p)
+                ops.gemm_fused(st, st, st, tail, nb, nb, nb, nb, nb, nb,
+                               opc, Op.NoTrans, -1.0, 1.0)
+                ev = torch.cuda.Event()
+                ev.record(su)
+                ev_tail[k] = ev
+    cur.wait_stream(sp)
+    cur.wait_stream(su)
+
+
 # ---------------- distributed ----------------
 
 def _dist_plan(mat: Matrix) -> _DescTable:
@@ -447,15 +588,20 @@ def cholesky_factorization(uplo: UpLo, mat: Matrix, grid: Optional[CommGrid] = N
 
     Upper runs through the U = L^H storage-transpose reduction (reference
     ``factorization/cholesky/impl.h:317`` implements call_U natively)."""
+    d = mat.dist
+    assert d.m == d.n and d.mb == d.nb, "square matrix with square tiles required"
+    g = grid if grid is not None else mat.grid
     if uplo == UpLo.Upper:
+        if g is None or not g.distributed:
+            _cholesky_local_upper(mat)   # native call_U (no transposes)
+            return
+        # distributed Upper: storage-transpose reduction (the reference's
+        # call_U is native there too — future work)
         from ._uplo import transpose_storage
         transpose_storage(mat)
         cholesky_factorization(UpLo.Lower, mat, grid)
         transpose_storage(mat)
         return
-    d = mat.dist
-    assert d.m == d.n and d.mb == d.nb, "square matrix with square tiles required"
-    g = grid if grid is not None else mat.grid
     if g is None or not g.distributed:
         _cholesky_local(mat)
     else:

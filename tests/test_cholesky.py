@@ -102,3 +102,32 @@ def test_cholesky_dist_cpu_2x4():
     errs = run_distributed(_dist_cholesky_worker, 8, args=(2, 4, 40, 8, "float64"))
     for e in errs:
         assert e < 1e-11 * 40, f"err={e}"
+
+
+def test_cholesky_upper_native_local():
+    """Native Upper path (no storage transpose): A = U^H U on upper tiles."""
+    import torch as _t
+    for n, nb in ((96, 32), (130, 48)):
+        mat = Matrix.create(n, n, nb, nb, dtype=_t.float64, device="cpu")
+        mutil.set_random_hermitian_positive_definite(mat, seed=9)
+        A = mat.to_global()
+        A = _t.triu(A.mT) + _t.triu(A.mT, 1).mT  # hermitianize via upper
+        mat.set_from_global(A)
+        cholesky_factorization(UpLo.Upper, mat)
+        U = _t.triu(mat.to_global())
+        err = (U.mH @ U - A).abs().max().item()
+        assert err < 1e-10 * n, err
+
+
+def test_cholesky_upper_native_complex():
+    import torch as _t
+    n, nb = 96, 32
+    mat = Matrix.create(n, n, nb, nb, dtype=_t.complex128, device="cpu")
+    mutil.set_random_hermitian_positive_definite(mat, seed=3)
+    A = mat.to_global()
+    A = _t.tril(A) + _t.tril(A, -1).mH
+    mat.set_from_global(A.clone())
+    cholesky_factorization(UpLo.Upper, mat)
+    U = _t.triu(mat.to_global())
+    err = (U.mH @ U - A).abs().max().item()
+    assert err < 1e-10 * n, err

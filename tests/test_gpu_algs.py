@@ -314,3 +314,21 @@ def test_gpu_band_chase_matches_cpu(dtype):
     assert de < 1e-8 * n and ee < 1e-8 * n, f"d={de} e={ee}"
     # reflector rounding compounds through ~n dependent sweeps
     assert ve < 1e-6, f"vstore diff {ve}"
+
+
+@pytest.mark.parametrize("dtype", [torch.float64, torch.complex128])
+def test_cholesky_upper_native_gpu(dtype):
+    """Native Upper POTRF on device (lookahead schedule, no transposes)."""
+    from dlaf_amd import Matrix, UpLo, cholesky_factorization
+    from dlaf_amd.matrix import util as mutil
+    n, nb = 1536, 512
+    mat = Matrix.create(n, n, nb, nb, dtype=dtype, device="cuda")
+    mutil.set_random_hermitian_positive_definite(mat, seed=13)
+    A = mat.to_global()
+    A = torch.tril(A) + torch.tril(A, -1).mH
+    mat.set_from_global(A.clone())
+    cholesky_factorization(UpLo.Upper, mat)
+    U = torch.triu(mat.to_global())
+    err = (U.mH @ U - A).abs().max().item()
+    scale = A.abs().max().item()
+    assert err < 1e-10 * n * scale, err
