@@ -104,7 +104,18 @@ class _DescTable:
         return self._dev[s : s + n]
 
 
-_PLAN_CACHE: Dict = {}
+class _BoundedCache(dict):
+    """Insertion-ordered bound on the descriptor-table cache (round-1
+    finding: a long-lived process sweeping shapes leaked device memory)."""
+    LIMIT = 24
+
+    def __setitem__(self, k, v):
+        if k not in self and len(self) >= self.LIMIT:
+            del self[next(iter(self))]
+        super().__setitem__(k, v)
+
+
+_PLAN_CACHE: Dict = _BoundedCache()
 
 
 def _trsm_plan_rows(table: _DescTable, key_prefix, offs: np.ndarray, nb: int,

@@ -32,7 +32,33 @@ class MatrixMirror:
             self.target = source
         else:
             self.target = Matrix(source.dist, source.dtype, self.device, source.grid)
-            self.target.storage.copy_(source.storage)
+            self._stage_copy(source.storage, self.target.storage)
+
+    @staticmethod
+    def _stage_copy(src: torch.Tensor, dst: torch.Tensor) -> None:
+        """H2D/D2H through a pooled PINNED bounce buffer (async-capable DMA;
+        an unpinned host tensor forces a slow pageable copy). Device->device
+        and host->host copies go direct."""
+        if src.device.type == dst.device.type:
+            dst.copy_(src)
+            return
+        from ..runtime import memory as mempool
+        host = src if src.device.type == "cpu" else dst
+        if host.is_pinned():
+            dst.copy_(src)
+            return
+        buf = mempool.acquire(src.shape, src.dtype, torch.device("cpu"),
+                              pinned=True)
+        try:
+            if src.device.type == "cpu":
+                buf.copy_(src)
+                dst.copy_(buf, non_blocking=True)
+                torch.cuda.synchronize(dst.device)
+            else:
+                buf.copy_(src)
+                dst.copy_(buf)
+        finally:
+            mempool.release(buf)
 
     def get(self) -> Matrix:
         return self.target
@@ -46,7 +72,7 @@ class MatrixMirror:
 
     def copy_back(self) -> None:
         if self.target is not self.source:
-            self.source.storage.copy_(self.target.storage)
+            self._stage_copy(self.target.storage, self.source.storage)
 
 
 class MatrixRef:
