@@ -25,6 +25,13 @@ def _worker(rank, world_size, port, fn, args, result_dir):
         with open(os.path.join(result_dir, f"rank{rank}.pkl"), "wb") as f:
             pickle.dump(out, f)
     finally:
+        # drain all in-flight gloo work before teardown: destroying the
+        # process group while a peer still communicates aborts in the gloo
+        # device thread ("terminate called without an active exception")
+        try:
+            dist.barrier()
+        except Exception:
+            pass
         dist.destroy_process_group()
 
 
