@@ -162,6 +162,8 @@ def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
     exactly those the ascending order keeps correctly ordered; the disjoint
     pairs commute).
     """
+    from contextlib import nullcontext as _nullcontext
+
     from ..config import get_tune_parameters
     if group_size is None:
         group_size = get_tune_parameters().bt_band_to_tridiag_hh_apply_group_size
@@ -183,33 +185,58 @@ def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
     Epad[:n] = E
     eyeG = torch.eye(G, dtype=E.dtype, device=dev)
     group_starts = list(range(0, nsweeps, G))
+    # Column-strip parallelism (round 2): the window chain only touches ROWS,
+    # so E column strips are independent — one HIP stream per strip fills the
+    # GPU that a single full-width chain leaves ~half empty (a window GEMM is
+    # only ~nE/64 workgroups tall). The per-group V/T prep runs on strip 0's
+    # stream; other strips wait its event. merge>1 keeps the single-stream
+    # path (tunable, default off).
+    import os as _os
+    n_strips = int(_os.environ.get("DLAF_BT_STREAMS", "2"))
+    use_strips = (dev.type == "cuda" and n_strips > 1 and merge == 1
+                  and nE >= 2 * n_strips)
+    if use_strips:
+        from ..runtime.streams import get_runtime
+        rt = get_runtime(dev)
+        streams = [rt.np_streams[i % len(rt.np_streams)]
+                   for i in range(n_strips)]
+        cur = torch.cuda.current_stream(dev)
+        for st_ in streams:
+            st_.wait_stream(cur)
+        bounds = [(nE * i) // n_strips for i in range(n_strips + 1)]
+    else:
+        streams = [None]
+        bounds = [0, nE]
     for s0 in reversed(group_starts):
         Gc = min(G, nsweeps - s0)       # sweeps in this group
         counts_g = counts[s0:s0 + Gc]
         nwin = int(counts_g.max())
         if nwin == 0:
             continue
-        # staircase V panels for all windows: Vg[k, row, g]
-        Vg = torch.zeros((nwin, H, G), dtype=E.dtype, device=dev)
-        taus_g = torch.zeros((nwin, G), dtype=E.dtype, device=dev)
-        for g in range(Gc):
-            Kg = int(counts_g[g])
-            if Kg == 0:
-                continue
-            o = int(offsets[s0 + g])
-            taus_g[:Kg, g] = V[o:o + Kg, 0]
-            Vg[:Kg, g:g + b, g] = V[o:o + Kg, 1:]
-        # batched T factors: T = inv(diag(1/tau) + striu(V^H V)); tau=0 rows/cols vanish
-        Gram = Vg.mH @ Vg                                   # [nwin, G, G]
-        zc = taus_g == 0
-        safe = torch.where(zc, torch.ones_like(taus_g), taus_g)
-        M = torch.triu(Gram, 1) + torch.diag_embed(1.0 / safe)
-        mask = zc.unsqueeze(1) | zc.unsqueeze(2)
-        M = torch.where(mask, torch.zeros_like(M), M)
-        M = M + torch.diag_embed(torch.where(zc, torch.ones_like(safe),
-                                             torch.zeros_like(safe)))
-        T = torch.linalg.solve_triangular(M, eyeG.expand(nwin, G, G).contiguous(), upper=True)
-        T = torch.where(mask, torch.zeros_like(T), T)
+        _prep_ctx = (torch.cuda.stream(streams[0]) if use_strips
+                     else _nullcontext())
+        with _prep_ctx:
+            # staircase V panels for all windows: Vg[k, row, g]
+            Vg = torch.zeros((nwin, H, G), dtype=E.dtype, device=dev)
+            taus_g = torch.zeros((nwin, G), dtype=E.dtype, device=dev)
+            for g in range(Gc):
+                Kg = int(counts_g[g])
+                if Kg == 0:
+                    continue
+                o = int(offsets[s0 + g])
+                taus_g[:Kg, g] = V[o:o + Kg, 0]
+                Vg[:Kg, g:g + b, g] = V[o:o + Kg, 1:]
+            # batched T factors: T = inv(diag(1/tau) + striu(V^H V)); tau=0 rows/cols vanish
+            Gram = Vg.mH @ Vg                                   # [nwin, G, G]
+            zc = taus_g == 0
+            safe = torch.where(zc, torch.ones_like(taus_g), taus_g)
+            M = torch.triu(Gram, 1) + torch.diag_embed(1.0 / safe)
+            mask = zc.unsqueeze(1) | zc.unsqueeze(2)
+            M = torch.where(mask, torch.zeros_like(M), M)
+            M = M + torch.diag_embed(torch.where(zc, torch.ones_like(safe),
+                                                 torch.zeros_like(safe)))
+            T = torch.linalg.solve_triangular(M, eyeG.expand(nwin, G, G).contiguous(), upper=True)
+            T = torch.where(mask, torch.zeros_like(T), T)
         # Merge m consecutive windows into one block-WY apply: the product
         # Q_{k+m-1}...Q_k of compact-WY transforms is itself compact-WY with
         #   Vc = [V_k .. V_{k+m-1}] (window j shifted j*b rows down) and
@@ -257,10 +284,31 @@ def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
                 W = Tc[km, :w, :w] @ (Vck.mH @ seg)
                 seg -= Vck @ W
         else:
-            # apply windows in ascending k (ordering constraint across overlaps)
-            for k in range(nwin):
-                base = 1 + s0 + k * b
-                seg = Epad[base:base + H]
-                W = T[k] @ (Vg[k].mH @ seg)
-                seg -= Vg[k] @ W
+            # apply windows in ascending k (ordering constraint across
+            # overlaps), per column strip on its own stream
+            def _apply_strip(cs0, cs1):
+                for k in range(nwin):
+                    base = 1 + s0 + k * b
+                    seg = Epad[base:base + H, cs0:cs1]
+                    W = T[k] @ (Vg[k].mH @ seg)
+                    seg -= Vg[k] @ W
+
+            if not use_strips:
+                _apply_strip(0, nE)
+            else:
+                ep = torch.cuda.Event()
+                ep.record(streams[0])  # V/T prep ran on strip 0's stream
+                for si in range(n_strips):
+                    with torch.cuda.stream(streams[si]):
+                        if si != 0:
+                            streams[si].wait_event(ep)
+                            # V/T were allocated on strip 0's stream; tell
+                            # the caching allocator they are consumed here
+                            Vg.record_stream(streams[si])
+                            T.record_stream(streams[si])
+                        _apply_strip(bounds[si], bounds[si + 1])
+    if use_strips:
+        cur = torch.cuda.current_stream(dev)
+        for st_ in streams:
+            cur.wait_stream(st_)
     E.copy_(Epad[:n])
