@@ -185,14 +185,13 @@ def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
     Epad[:n] = E
     eyeG = torch.eye(G, dtype=E.dtype, device=dev)
     group_starts = list(range(0, nsweeps, G))
-    # Column-strip parallelism (round 2): the window chain only touches ROWS,
-    # so E column strips are independent — one HIP stream per strip fills the
-    # GPU that a single full-width chain leaves ~half empty (a window GEMM is
-    # only ~nE/64 workgroups tall). The per-group V/T prep runs on strip 0's
-    # stream; other strips wait its event. merge>1 keeps the single-stream
-    # path (tunable, default off).
+    # Column-strip parallelism (tunable, default OFF): the window chain only
+    # touches rows, so E column strips are independent. Measured at n=20000:
+    # strips LOSE (3.6 -> 4.1 s at 2 strips) — the chain is host-launch
+    # bound (~70k small GEMMs), and strips double the launch count. Kept as
+    # DLAF_BT_STREAMS for wider-nE regimes.
     import os as _os
-    n_strips = int(_os.environ.get("DLAF_BT_STREAMS", "2"))
+    n_strips = int(_os.environ.get("DLAF_BT_STREAMS", "1"))
     use_strips = (dev.type == "cuda" and n_strips > 1 and merge == 1
                   and nE >= 2 * n_strips)
     if use_strips:
