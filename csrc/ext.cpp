@@ -355,6 +355,32 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "CPU bulge chasing band->tridiag with reflector recording",
         py::arg("band"), py::arg("b"), py::arg("vstore"), py::arg("offsets"),
         py::arg("nthreads") = 0);
+  m.def("bt_apply_group", [](torch::Tensor E, torch::Tensor V,
+                             torch::Tensor VTt, int64_t base0, int64_t b,
+                             int64_t G, int64_t R, int64_t nwin) -> bool {
+    TORCH_CHECK(E.is_cuda() && V.is_cuda() && VTt.is_cuda());
+    TORCH_CHECK(E.is_contiguous() && V.is_contiguous() && VTt.is_contiguous());
+    auto s = cur_stream();
+    const int64_t npad = E.size(0), nE = E.size(1);
+    int ok = 0;
+    switch (E.scalar_type()) {
+      case at::kDouble:
+        ok = bt_apply_group_f64(E.data_ptr<double>(), nE, npad,
+                                V.data_ptr<double>(), VTt.data_ptr<double>(),
+                                base0, (int)b, (int)G, (int)R, (int)nwin, s);
+        break;
+      case at::kComplexDouble:
+        ok = bt_apply_group_c128((double*)E.data_ptr(), nE, npad,
+                                 (const double*)V.data_ptr(),
+                                 (const double*)VTt.data_ptr(), base0, (int)b,
+                                 (int)G, (int)R, (int)nwin, s);
+        break;
+      default:
+        return false;
+    }
+    HIP_CHECK(hipGetLastError());
+    return ok != 0;
+  }, "whole-group bt window-chain apply");
   m.def("batch_gemm", &batch_gemm,
         "fused batched tile GEMM: C[d] = alpha*op(A[d])op(B[d]) + beta*C[d]");
   m.def("potrf_block", &potrf_block, "single-workgroup Cholesky block factor");

@@ -72,6 +72,14 @@ int gemm_tiles_v2_c64(const GemmDesc* descs, int ndesc, const float* A,
                       float alpha_im, float beta_re, float beta_im,
                       hipStream_t stream);
 
+// ---- bt window-chain group apply (one launch per sweep group) ----
+int bt_apply_group_f64(double* E, int64_t nE, int64_t npad, const double* V,
+                       const double* VTt, int64_t base0, int b, int G, int R,
+                       int nwin, hipStream_t stream);
+int bt_apply_group_c128(double* E, int64_t nE, int64_t npad, const double* V,
+                        const double* VTt, int64_t base0, int b, int G, int R,
+                        int nwin, hipStream_t stream);
+
 // ---- single-tile factorization building blocks ----
 // Fused single-workgroup [factor +] invert of one diagonal block: factors the
 // leading n x n (if do_factor) in place and writes its inverse into the

@@ -283,6 +283,25 @@ def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
                 W = Tc[km, :w, :w] @ (Vck.mH @ seg)
                 seg -= Vck @ W
         else:
+            # whole-group single-launch kernel (csrc/bt_apply.hip): each
+            # workgroup owns a 16-column slice and marches the full window
+            # chain in LDS — replaces ~3*nwin GEMM launches per group
+            # (launch-bound, measured) with ONE.
+            if (dev.type == "cuda"
+                    and E.dtype in (torch.float64, torch.complex128)
+                    and G % 32 == 0 and b % 16 == 0
+                    and _os.environ.get("DLAF_BT_KERNEL", "1") != "0"):
+                R = -(-H // b) * b
+                if R % 16 == 0:
+                    from ..ops._ext import get_ext
+                    Vp = torch.zeros((nwin, R, G), dtype=E.dtype, device=dev)
+                    Vp[:, :H] = Vg
+                    VTt = torch.zeros((nwin, G, R), dtype=E.dtype, device=dev)
+                    VTt[:, :, :H] = torch.bmm(Vg, T).mT
+                    if get_ext().bt_apply_group(Epad, Vp.contiguous(),
+                                                VTt.contiguous(), 1 + s0,
+                                                b, G, R, nwin):
+                        continue
             # apply windows in ascending k (ordering constraint across
             # overlaps), per column strip on its own stream
             def _apply_strip(cs0, cs1):

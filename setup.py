@@ -21,6 +21,7 @@ setup(
                 "csrc/band_chase.cpp",
                 "csrc/gemm_tiles.hip",
                 "csrc/gemm_tiles_v2.hip",
+                "csrc/bt_apply.hip",
                 "csrc/factor.hip",
                 "csrc/panel_qr.hip",
                 "csrc/secular.hip",
