@@ -100,39 +100,63 @@ __global__ __launch_bounds__(256) void bt_group_f64(
     const double* Vk = V + (int64_t)k * R * G;
     const double* VTk = VTt + (int64_t)k * G * R;
 
-    // W1[g][c] = sum_h conj(V[h][g]) * ring[h][c]
-    for (int f = w; f < G / 16; f += 4) {
-      v4d acc = {0, 0, 0, 0};
+    // W1[g][c] = sum_h conj(V[h][g]) * ring[h][c]. All of this wave's
+    // fragments run in ONE k-loop (independent accumulators interleave so
+    // the MFMA dependent-accumulator latency never serializes), and the
+    // ring read is shared across fragments.
+    {
+      constexpr int NF = 2;  // G==128: (G/16)/4 frags per wave
+      v4d acc[NF] = {{0, 0, 0, 0}, {0, 0, 0, 0}};
 #pragma unroll 4
       for (int h0 = 0; h0 < R; h0 += 4) {
         const int h = h0 + lk;
         int slot = sbase + h;
         if (slot >= R) slot -= R;
-        const double a = Vk[(int64_t)h * G + f * 16 + li];
         const double bb = ring[slot * CWP + li];
-        acc = BtMfma<double>::mma(a, bb, acc);
+#pragma unroll
+        for (int j = 0; j < NF; ++j) {
+          const int f = w + 4 * j;
+          if (f < G / 16) {
+            const double a = Vk[(int64_t)h * G + f * 16 + li];
+            acc[j] = BtMfma<double>::mma(a, bb, acc[j]);
+          }
+        }
       }
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        W1[(f * 16 + lk + 4 * r) * CWP + li] = acc[r];
+      for (int j = 0; j < NF; ++j)
+        if (w + 4 * j < G / 16)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            W1[((w + 4 * j) * 16 + lk + 4 * r) * CWP + li] = acc[j][r];
     }
     __syncthreads();
 
-    // ring[h][c] -= sum_g VTt[g][h] * W1[g][c]
-    for (int f = w; f < R / 16; f += 4) {
-      v4d acc = {0, 0, 0, 0};
+    // ring[h][c] -= sum_g VTt[g][h] * W1[g][c]; same interleaved structure
+    {
+      const int nfr = R / 16;  // <= 12 (b <= 64)
+      v4d acc[3] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+      const int nj = (nfr - w + 3) / 4;  // frags this wave owns (<= 3)
 #pragma unroll 4
       for (int g0 = 0; g0 < G; g0 += 4) {
-        const double a = VTk[(int64_t)(g0 + lk) * R + f * 16 + li];
         const double bb = W1[(g0 + lk) * CWP + li];
-        acc = BtMfma<double>::mma(a, bb, acc);
-      }
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int slot = sbase + f * 16 + lk + 4 * r;
-        if (slot >= R) slot -= R;
-        if (slot >= R) slot -= R;
-        ring[slot * CWP + li] -= acc[r];
+        for (int j = 0; j < 3; ++j) {
+          if (j < nj) {
+            const int f = w + 4 * j;
+            const double a = VTk[(int64_t)(g0 + lk) * R + f * 16 + li];
+            acc[j] = BtMfma<double>::mma(a, bb, acc[j]);
+          }
+        }
+      }
+      for (int j = 0; j < nj; ++j) {
+        const int f = w + 4 * j;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int slot = sbase + f * 16 + lk + 4 * r;
+          if (slot >= R) slot -= R;
+          if (slot >= R) slot -= R;
+          ring[slot * CWP + li] -= acc[j][r];
+        }
       }
     }
     __syncthreads();
@@ -200,51 +224,74 @@ __global__ __launch_bounds__(256) void bt_group_c128(
     const double* Vk = V + (int64_t)k * R * G * 2;
     const double* VTk = VTt + (int64_t)k * G * R * 2;
 
-    // W1 = V^H seg: (ar - i ai)(br + i bi)
-    for (int f = w; f < G / 16; f += 4) {
-      v4d ar_ = {0, 0, 0, 0}, ai_ = {0, 0, 0, 0};
-#pragma unroll 4
+    // W1 = V^H seg: (ar - i ai)(br + i bi); fragment-interleaved
+    {
+      constexpr int NF = 2;
+      v4d ar_[NF] = {{0, 0, 0, 0}, {0, 0, 0, 0}};
+      v4d ai_[NF] = {{0, 0, 0, 0}, {0, 0, 0, 0}};
+#pragma unroll 2
       for (int h0 = 0; h0 < R; h0 += 4) {
         const int h = h0 + lk;
         int slot = sbase + h;
         if (slot >= R) slot -= R;
-        const int64_t va = ((int64_t)h * G + f * 16 + li) * 2;
-        const double vr = Vk[va], vi = Vk[va + 1];
         const double br = ring[(slot * CWP + li) * 2];
         const double bi = ring[(slot * CWP + li) * 2 + 1];
-        ar_ = BtMfma<double>::mma(vr, br, ar_);
-        ar_ = BtMfma<double>::mma(vi, bi, ar_);
-        ai_ = BtMfma<double>::mma(vr, bi, ai_);
-        ai_ = BtMfma<double>::mma(-vi, br, ai_);
+#pragma unroll
+        for (int j = 0; j < NF; ++j) {
+          const int f = w + 4 * j;
+          if (f < G / 16) {
+            const int64_t va = ((int64_t)h * G + f * 16 + li) * 2;
+            const double vr = Vk[va], vi = Vk[va + 1];
+            ar_[j] = BtMfma<double>::mma(vr, br, ar_[j]);
+            ai_[j] = BtMfma<double>::mma(vr, bi, ai_[j]);
+            ar_[j] = BtMfma<double>::mma(vi, bi, ar_[j]);
+            ai_[j] = BtMfma<double>::mma(-vi, br, ai_[j]);
+          }
+        }
       }
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        W1[((f * 16 + lk + 4 * r) * CWP + li) * 2] = ar_[r];
-        W1[((f * 16 + lk + 4 * r) * CWP + li) * 2 + 1] = ai_[r];
-      }
+      for (int j = 0; j < NF; ++j)
+        if (w + 4 * j < G / 16)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            W1[(((w + 4 * j) * 16 + lk + 4 * r) * CWP + li) * 2] = ar_[j][r];
+            W1[(((w + 4 * j) * 16 + lk + 4 * r) * CWP + li) * 2 + 1] = ai_[j][r];
+          }
     }
     __syncthreads();
 
-    for (int f = w; f < R / 16; f += 4) {
-      v4d ar_ = {0, 0, 0, 0}, ai_ = {0, 0, 0, 0};
-#pragma unroll 4
+    {
+      const int nfr = R / 16;
+      v4d ar_[3] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+      v4d ai_[3] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+      const int nj = (nfr - w + 3) / 4;
+#pragma unroll 2
       for (int g0 = 0; g0 < G; g0 += 4) {
-        const int64_t va = ((int64_t)(g0 + lk) * R + f * 16 + li) * 2;
-        const double vr = VTk[va], vi = VTk[va + 1];
         const double br = W1[((g0 + lk) * CWP + li) * 2];
         const double bi = W1[((g0 + lk) * CWP + li) * 2 + 1];
-        ar_ = BtMfma<double>::mma(vr, br, ar_);
-        ar_ = BtMfma<double>::mma(-vi, bi, ar_);
-        ai_ = BtMfma<double>::mma(vr, bi, ai_);
-        ai_ = BtMfma<double>::mma(vi, br, ai_);
-      }
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int slot = sbase + f * 16 + lk + 4 * r;
-        if (slot >= R) slot -= R;
-        if (slot >= R) slot -= R;
-        ring[(slot * CWP + li) * 2] -= ar_[r];
-        ring[(slot * CWP + li) * 2 + 1] -= ai_[r];
+        for (int j = 0; j < 3; ++j) {
+          if (j < nj) {
+            const int f = w + 4 * j;
+            const int64_t va = ((int64_t)(g0 + lk) * R + f * 16 + li) * 2;
+            const double vr = VTk[va], vi = VTk[va + 1];
+            ar_[j] = BtMfma<double>::mma(vr, br, ar_[j]);
+            ai_[j] = BtMfma<double>::mma(vr, bi, ai_[j]);
+            ar_[j] = BtMfma<double>::mma(-vi, bi, ar_[j]);
+            ai_[j] = BtMfma<double>::mma(vi, br, ai_[j]);
+          }
+        }
+      }
+      for (int j = 0; j < nj; ++j) {
+        const int f = w + 4 * j;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int slot = sbase + f * 16 + lk + 4 * r;
+          if (slot >= R) slot -= R;
+          if (slot >= R) slot -= R;
+          ring[(slot * CWP + li) * 2] -= ar_[j][r];
+          ring[(slot * CWP + li) * 2 + 1] -= ai_[j][r];
+        }
       }
     }
     __syncthreads();
