@@ -198,6 +198,7 @@ def _trtri_local_gpu(mat: Matrix, unit: bool) -> None:
     kchunk = int(os.environ.get("DLAF_TRTRI_KSPLIT", "4"))
     P = max(1, -(-(nt - 2) // kchunk))
     acc = torch.zeros((2, P, nt, nb, nb), dtype=dt, device=dev)
+    accsum = torch.zeros((nt, nb, nb), dtype=dt, device=dev)
 
     rt = get_runtime(dev)
     sp, su = rt.hp_streams[0], rt.np_streams[0]
@@ -242,21 +243,18 @@ def _trtri_local_gpu(mat: Matrix, unit: bool) -> None:
                 sp.wait_event(ev_partial[k])
             n_rows = nt - (k + 1)
             if n_rows > 0:
-                # fix-up into plane 0 (always written by the partial), then
-                # write the column with the plane reduction folded into the
-                # GEMM's K-chain (ktiles = P over the planes; saves a full
-                # read+write of the plane stack per step)
-                fix = [[buf * P * nt * ts + i * ts, tile_off(i, k + 1),
+                # collapse the K-split planes, then fix-up + column write
+                torch.sum(acc[buf], dim=0, out=accsum)
+                fix = [[i * ts, tile_off(i, k + 1),
                         tile_off(k + 1, k), 1, 0, 0]
                        for i in range(k + 1, nt)]
-                ops.gemm_fused(acc.view(-1), st.view(-1), st.view(-1),
+                ops.gemm_fused(accsum.view(-1), st.view(-1), st.view(-1),
                                np.array(fix, dtype=np.int64), nb, nb, nb,
                                nb, nb, nb, Op.NoTrans, Op.NoTrans, 1.0, 1.0)
-                # X[:, k] = -sum_p plane_p @ inv_k  (B revisits inv_k: k-stride 0)
-                wr = [[tile_off(i, k), buf * P * nt * ts + i * ts, k * ts,
-                       P, nt * ts, 0]
+                # column write: X[:, k] = -acc @ inv_k
+                wr = [[tile_off(i, k), i * ts, k * ts, 1, 0, 0]
                       for i in range(k + 1, nt)]
-                ops.gemm_fused(st.view(-1), acc.view(-1), inv_stack.view(-1),
+                ops.gemm_fused(st.view(-1), accsum.view(-1), inv_stack.view(-1),
                                np.array(wr, dtype=np.int64), nb, nb, nb,
                                nb, nb, nb, Op.NoTrans, Op.NoTrans, -1.0, 0.0)
             t = mat.tile((k, k))
