@@ -176,32 +176,10 @@ def main():
 
     for _ in range(args.warmup):
         step()
-
-    # Single-GPU: capture the whole step (restore + factorization: a few
-    # thousand kernel launches over the lookahead streams) into one hipGraph
-    # and replay it — removes the per-launch host overhead from the timed
-    # loop. Multi-rank keeps eager issue (RCCL collectives inside capture
-    # are not exercised). DLAF_BENCH_GRAPH=0 disables.
-    run_step = step
-    if (world_size == 1 and device.type == "cuda"
-            and os.environ.get("DLAF_BENCH_GRAPH", "1") != "0"):
-        try:
-            g = torch.cuda.CUDAGraph()
-            torch.cuda.synchronize(device)
-            with torch.cuda.graph(g):
-                step()
-            g.replay()
-            torch.cuda.synchronize(device)
-            run_step = g.replay
-        except Exception as e:  # pragma: no cover - fallback is eager
-            print(f"# graph capture unavailable ({type(e).__name__}); eager",
-                  flush=True)
-            run_step = step
-
     barrier_sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        run_step()
+        step()
     barrier_sync()
     elapsed = time.perf_counter() - t0
 

@@ -78,35 +78,3 @@ def test_cholesky_trsm_complex64_gpu():
     want = torch.linalg.solve(L.to(torch.complex128), b0.to(torch.complex128))
     err = (B.to_global().cpu().to(torch.complex128) - want).abs().max().item()
     assert err < 1e-2, f"trsm err={err}"
-
-
-def test_cholesky_graph_capture_replay():
-    """hipGraph capture of the full factorization step replays correctly
-    (the bench.py timed-loop mechanism)."""
-    import torch
-    from dlaf_amd import Matrix, UpLo, cholesky_factorization
-    from dlaf_amd.matrix import util as mutil
-    n, nb = 4096, 512
-    mat = Matrix.create(n, n, nb, nb, dtype=torch.float64, device="cuda")
-    mutil.set_random_hermitian_positive_definite(mat, seed=17)
-    pristine = mat.storage.clone()
-    a0 = mat.to_global()
-    a0 = torch.tril(a0) + torch.tril(a0, -1).mH
-
-    def step():
-        mat.storage.copy_(pristine)
-        cholesky_factorization(UpLo.Lower, mat)
-
-    step()  # warm (allocations, desc tables)
-    eager = torch.tril(mat.to_global()).clone()
-    g = torch.cuda.CUDAGraph()
-    torch.cuda.synchronize()
-    with torch.cuda.graph(g):
-        step()
-    g.replay()
-    g.replay()
-    torch.cuda.synchronize()
-    got = torch.tril(mat.to_global())
-    assert torch.equal(got, eager), "graph replay diverged from eager"
-    res = (a0 - got @ got.mH).abs().max().item() / a0.abs().max().item()
-    assert res < 1e-12, res
