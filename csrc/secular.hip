@@ -274,8 +274,9 @@ extern "C" {
 void secular_roots_f64(const double* d, const double* z2, int k, double rho,
                        long long* sidx, double* mu, hipStream_t stream) {
   if (k <= 0) return;
-  if (k >= 1024) {
-    // wave per root: 4 roots per 256-thread workgroup
+  if (k >= 128) {
+    // wave per root: 4 roots per 256-thread workgroup (a thread-per-root
+    // grid at k = 512 is 4 workgroups on a 256-CU chip)
     const int blocks = (k + 3) / 4;
     secular_wave_kernel<double><<<blocks, 256, 0, stream>>>(d, z2, k, rho,
                                                             sidx, mu);
