@@ -38,13 +38,14 @@ __device__ inline v4d mma_f64(double a, double b, v4d c) {
 // real f64: CW = 64, LDS = (R + G) * 64 * 8 = 160 KiB at R=192 ------------
 
 template <int G, int R>
-__global__ __launch_bounds__(256, 1) void bt_group_f64(
+__global__ __launch_bounds__(512, 1) void bt_group_f64(
     double* __restrict__ E, int64_t nE, int64_t npad,
     const double* __restrict__ V, const double* __restrict__ VTt,
     int64_t base0, int b, int nwin) {
   constexpr int CW = 64;
-  constexpr int NRF1 = (G / 16) / 4;      // W1 row-frags per wave (G=128: 2)
-  constexpr int NRF2 = (R / 16 + 3) / 4;  // update row-frags per wave (<=3)
+  constexpr int NW = 8;                     // waves (512 threads)
+  constexpr int NRF1 = (G / 16 + NW - 1) / NW;   // W1 row-frags/wave (1)
+  constexpr int NRF2 = (R / 16 + NW - 1) / NW;   // update row-frags/wave (2)
   extern __shared__ double S[];  // ring[R][CW] then W1[G][CW]
   double* ring = S;
   double* W1 = S + (int64_t)R * CW;
@@ -55,7 +56,7 @@ __global__ __launch_bounds__(256, 1) void bt_group_f64(
   const int64_t col0 = (int64_t)blockIdx.x * CW;
 
   auto move_rows = [&](int64_t gbase, int sbase, int f0, int cnt, bool store) {
-    for (int f = f0 + tid / CW; f < f0 + cnt; f += 256 / CW) {
+    for (int f = f0 + tid / CW; f < f0 + cnt; f += (NW * 64) / CW) {
       const int c = tid % CW;
       int slot = sbase + f;
       if (slot >= R) slot -= R;
@@ -92,7 +93,7 @@ __global__ __launch_bounds__(256, 1) void bt_group_f64(
       for (int j = 0; j < NRF1; ++j)
 #pragma unroll
         for (int cf = 0; cf < 4; ++cf) acc[j][cf] = {0, 0, 0, 0};
-#pragma unroll 2
+#pragma unroll 4
       for (int h0 = 0; h0 < R; h0 += 4) {
         const int h = h0 + lk;
         int slot = sbase + h;
@@ -100,7 +101,7 @@ __global__ __launch_bounds__(256, 1) void bt_group_f64(
         double a[NRF1];
 #pragma unroll
         for (int j = 0; j < NRF1; ++j)
-          a[j] = Vk[(int64_t)h * G + (w + 4 * j) * 16 + li];
+          a[j] = Vk[(int64_t)h * G + (w + NW * j) * 16 + li];
 #pragma unroll
         for (int cf = 0; cf < 4; ++cf) {
           const double bb = ring[slot * CW + cf * 16 + li];
@@ -115,7 +116,7 @@ __global__ __launch_bounds__(256, 1) void bt_group_f64(
         for (int cf = 0; cf < 4; ++cf)
 #pragma unroll
           for (int r = 0; r < 4; ++r)
-            W1[((w + 4 * j) * 16 + lk + 4 * r) * CW + cf * 16 + li] =
+            W1[((w + NW * j) * 16 + lk + 4 * r) * CW + cf * 16 + li] =
                 acc[j][cf][r];
     }
     __syncthreads();
@@ -123,19 +124,19 @@ __global__ __launch_bounds__(256, 1) void bt_group_f64(
     // ring[h][c] -= sum_g VTt[g][h] * W1[g][c]
     {
       const int nfr = R / 16;
-      const int nj = (nfr - w + 3) / 4;
+      const int nj = (nfr - w + NW - 1) / NW;
       v4d acc[NRF2][4];
 #pragma unroll
       for (int j = 0; j < NRF2; ++j)
 #pragma unroll
         for (int cf = 0; cf < 4; ++cf) acc[j][cf] = {0, 0, 0, 0};
-#pragma unroll 2
+#pragma unroll 4
       for (int g0 = 0; g0 < G; g0 += 4) {
         double a[NRF2];
 #pragma unroll
         for (int j = 0; j < NRF2; ++j)
           a[j] = (j < nj)
-                     ? VTk[(int64_t)(g0 + lk) * R + (w + 4 * j) * 16 + li]
+                     ? VTk[(int64_t)(g0 + lk) * R + (w + NW * j) * 16 + li]
                      : 0.0;
 #pragma unroll
         for (int cf = 0; cf < 4; ++cf) {
@@ -146,7 +147,7 @@ __global__ __launch_bounds__(256, 1) void bt_group_f64(
         }
       }
       for (int j = 0; j < nj; ++j) {
-        const int f = w + 4 * j;
+        const int f = w + NW * j;
 #pragma unroll
         for (int cf = 0; cf < 4; ++cf)
 #pragma unroll
@@ -352,7 +353,7 @@ int bt_apply_group_f64(double* E, int64_t nE, int64_t npad, const double* V,
       (void)hipFuncSetAttribute(fp,                                        \
                                 hipFuncAttributeMaxDynamicSharedMemorySize,\
                                 (int)sh);                                  \
-    bt_group_f64<128, RT><<<blocks, 256, sh, stream>>>(E, nE, npad, V,     \
+    bt_group_f64<128, RT><<<blocks, 512, sh, stream>>>(E, nE, npad, V,     \
                                                        VTt, base0, b,      \
                                                        nwin);              \
     return 1;                                                              \
