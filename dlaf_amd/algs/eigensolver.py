@@ -144,6 +144,7 @@ def hermitian_generalized_eigensolver(
     mat_b: Matrix,
     grid: Optional[CommGrid] = None,
     factorized: bool = False,
+    band: Optional[int] = None,
     eigenvalues_index_begin: int = 0,
     eigenvalues_index_end: Optional[int] = None,
 ) -> Tuple[torch.Tensor, Matrix]:
@@ -159,7 +160,7 @@ def hermitian_generalized_eigensolver(
         cholesky_factorization(UpLo.Lower, mat_b, g)
     generalized_to_standard(UpLo.Lower, mat_a, mat_b, g)
     w, evecs = hermitian_eigensolver(
-        UpLo.Lower, mat_a, g,
+        UpLo.Lower, mat_a, g, band=band,
         eigenvalues_index_begin=eigenvalues_index_begin,
         eigenvalues_index_end=eigenvalues_index_end,
     )

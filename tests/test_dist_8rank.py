@@ -120,3 +120,25 @@ def _w_norm(rank, ws):
 def test_dist_8rank_2x4(worker):
     for e in run_distributed(worker, 8):
         assert e < 1e-10 * N, e
+
+
+def _w_gen_eig(rank, ws):
+    from dlaf_amd import hermitian_generalized_eigensolver
+    grid = CommGrid(2, 4)
+    n, nb, band = 64, 8, 4
+    a = _mk(n, nb, grid, seed=13)
+    bm = _mk(n, nb, grid, seed=14, spd=True)
+    A = _herm(a.to_global())
+    a.set_from_global(A.clone())
+    B = _herm(bm.to_global())
+    bm.set_from_global(B.clone())
+    w, evecs = hermitian_generalized_eigensolver(UpLo.Lower, a, bm, grid,
+                                                 band=band)
+    E = evecs.to_global()
+    R = A @ E - B @ E @ torch.diag(w.to(E.dtype))
+    return float(R.abs().max())
+
+
+def test_dist_8rank_gen_eigensolver():
+    for e in run_distributed(_w_gen_eig, 8):
+        assert e < 1e-9, e
