@@ -599,8 +599,11 @@ def cholesky_factorization(uplo: UpLo, mat: Matrix, grid: Optional[CommGrid] = N
 
     Upper runs through the U = L^H storage-transpose reduction (reference
     ``factorization/cholesky/impl.h:317`` implements call_U natively)."""
+    from ..core.asserts import dlaf_assert
     d = mat.dist
-    assert d.m == d.n and d.mb == d.nb, "square matrix with square tiles required"
+    dlaf_assert(d.m == d.n and d.mb == d.nb,
+                "square matrix with square tiles required", d.size,
+                d.tile_size)
     g = grid if grid is not None else mat.grid
     if uplo == UpLo.Upper:
         if g is None or not g.distributed:

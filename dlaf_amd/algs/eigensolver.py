@@ -55,8 +55,13 @@ def hermitian_eigensolver(
     eigenvector columns (back-transforms applied only to the slice, the
     reference's MatrixRef mechanism).
     """
-    assert uplo == UpLo.Lower, "only Lower implemented (as the reference miniapps)"
+    from ..core.asserts import dlaf_assert
+    dlaf_assert(uplo == UpLo.Lower,
+                "only Lower implemented (as the reference miniapps)")
     d = mat.dist
+    dlaf_assert(d.m == d.n and d.mb == d.nb,
+                "square matrix with square tiles required", d.size,
+                d.tile_size)
     n = d.m
     g = grid if grid is not None else mat.grid
     if band is None:

@@ -87,3 +87,30 @@ def test_mirror_pinned_h2d_roundtrip():
         assert dev_mat.device.type == "cuda"
         dev_mat.storage.mul_(2.0)
     assert torch.equal(mat.to_global(), 2.0 * before)
+
+
+def test_assert_levels(monkeypatch):
+    from dlaf_amd.core import asserts
+    monkeypatch.setenv("DLAF_ASSERT_LEVEL", "1")
+    asserts.dlaf_assert(True)
+    try:
+        asserts.dlaf_assert(False, "boom")
+        raise RuntimeError("not raised")
+    except asserts.DlafAssertError as e:
+        assert "boom" in str(e)
+    asserts.dlaf_assert_moderate(False)  # level 1: not checked
+    monkeypatch.setenv("DLAF_ASSERT_LEVEL", "0")
+    asserts.dlaf_assert(False)  # disabled
+    monkeypatch.setenv("DLAF_ASSERT_LEVEL", "3")
+    flag = []
+    asserts.dlaf_assert_heavy(lambda: flag.append(1) or True)
+    assert flag  # evaluated at level 3
+
+
+def test_cholesky_entry_assert():
+    import pytest as _pytest
+    from dlaf_amd import cholesky_factorization, UpLo
+    from dlaf_amd.core.asserts import DlafAssertError
+    m = Matrix.create(12, 16, 4, 4, dtype=torch.float64)
+    with _pytest.raises(DlafAssertError):
+        cholesky_factorization(UpLo.Lower, m)
