@@ -102,7 +102,8 @@ def chase_band(store: torch.Tensor, b: int) -> TridiagResult:
         store_h = store.cpu() if store.is_cuda else store
         vstore = torch.zeros((max(total, 1), b + 1), dtype=store_h.dtype)
         if n > 2:
-            get_ext().band_chase(store_h, b, vstore, offsets)
+            nthr = int(os.environ.get("DLAF_CHASE_THREADS", "0"))
+            get_ext().band_chase(store_h, b, vstore, offsets, nthr)
         store = store_h
     dvec = store[:, 0]
     evec = store[: n - 1, 1] if n > 1 else store[:0, 1]
