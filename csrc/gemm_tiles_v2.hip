@@ -77,17 +77,17 @@ __device__ inline int xcd_remap(int wg, int nwg) {
 // SOURCE address (LDS stays lane-linear, the glds contract) and undone in
 // the fragment index; residual conflict is 2-way.
 template <typename T, int OPA, int OPB, int SWZ, int DEPTH, int BN,
-          int BK = 16>
+          int BK = 16, int BM = 128>
 __launch_bounds__(256, DEPTH == 2 ? 2 : 1) __global__ void gemm_v2_k(
     const GemmDesc* __restrict__ descs, const T* __restrict__ A,
     const T* __restrict__ B, T* __restrict__ C, int M, int N, int K, int lda,
     int ldb, int ldc, T alpha, T beta, int mblocks, int nblocks) {
-  constexpr int BM = 128;
   constexpr int EPL = 16 / sizeof(T);  // elements per lane per glds
   constexpr int CPR = BK / EPL;        // 16-B chunks per rk-image row
   constexpr int AEL = BM * BK;         // A image elements
   constexpr int BEL = BK * BN;         // B image elements
-  constexpr int NFRAG = BN / 32;
+  constexpr int NFRAG = BN / 32;  // col frags per wave
+  constexpr int MFRAG = BM / 32;  // row frags per wave
   __shared__ T S[DEPTH][AEL + BEL];
 
   int wg = blockIdx.x;
@@ -99,7 +99,7 @@ __launch_bounds__(256, DEPTH == 2 ? 2 : 1) __global__ void gemm_v2_k(
 
   const int tid = threadIdx.x;
   const int lane = tid & 63, w = tid >> 6;
-  const int wrow = (w >> 1) * 64, wcol = (w & 1) * (BN / 2);
+  const int wrow = (w >> 1) * (BM / 2), wcol = (w & 1) * (BN / 2);
   const int li = lane & 15, lk = lane >> 4;
 
   const int spt = K / BK;  // steps per K-tile
@@ -109,9 +109,9 @@ __launch_bounds__(256, DEPTH == 2 ? 2 : 1) __global__ void gemm_v2_k(
   constexpr int NBW = BEL / (64 * EPL) / 4;
 
   using acc_t = typename MfmaV2<T>::acc_t;
-  acc_t acc[4][NFRAG];
+  acc_t acc[MFRAG][NFRAG];
 #pragma unroll
-  for (int a = 0; a < 4; ++a)
+  for (int a = 0; a < MFRAG; ++a)
 #pragma unroll
     for (int b = 0; b < NFRAG; ++b) acc[a][b] = {0, 0, 0, 0};
 
@@ -193,9 +193,10 @@ __launch_bounds__(256, DEPTH == 2 ? 2 : 1) __global__ void gemm_v2_k(
                  ? k * BN + col
                  : col * BK + EPL * ((k / EPL) ^ (col & (CPR - 1))) + k % EPL;
     };
-    T af[2][4], bf[2][NFRAG];
+    T af[2][MFRAG], bf[2][NFRAG];
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi) af[0][mi] = sa[aidx(wrow + mi * 16 + li, lk)];
+    for (int mi = 0; mi < MFRAG; ++mi)
+      af[0][mi] = sa[aidx(wrow + mi * 16 + li, lk)];
 #pragma unroll
     for (int ni = 0; ni < NFRAG; ++ni)
       bf[0][ni] = sb[bidx(wcol + ni * 16 + li, lk)];
@@ -206,14 +207,14 @@ __launch_bounds__(256, DEPTH == 2 ? 2 : 1) __global__ void gemm_v2_k(
         const int nxt = 1 - cur;
         const int k = (ks + 1) * 4 + lk;
 #pragma unroll
-        for (int mi = 0; mi < 4; ++mi)
+        for (int mi = 0; mi < MFRAG; ++mi)
           af[nxt][mi] = sa[aidx(wrow + mi * 16 + li, k)];
 #pragma unroll
         for (int ni = 0; ni < NFRAG; ++ni)
           bf[nxt][ni] = sb[bidx(wcol + ni * 16 + li, k)];
       }
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < MFRAG; ++mi)
 #pragma unroll
         for (int ni = 0; ni < NFRAG; ++ni)
           acc[mi][ni] = MfmaV2<T>::mma(af[cur][mi], bf[cur][ni], acc[mi][ni]);
@@ -258,7 +259,7 @@ __launch_bounds__(256, DEPTH == 2 ? 2 : 1) __global__ void gemm_v2_k(
   T* Cb = C + d.c_off;
   if (beta != T(0)) {
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi) {
+    for (int mi = 0; mi < MFRAG; ++mi) {
       T cv[4 * NFRAG];
 #pragma unroll
       for (int ni = 0; ni < NFRAG; ++ni)
@@ -281,7 +282,7 @@ __launch_bounds__(256, DEPTH == 2 ? 2 : 1) __global__ void gemm_v2_k(
     }
   } else {
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
+    for (int mi = 0; mi < MFRAG; ++mi)
 #pragma unroll
       for (int ni = 0; ni < NFRAG; ++ni) {
         const acc_t v = acc[mi][ni];
@@ -516,7 +517,12 @@ template <typename T>
 int launch_v2_real(const GemmDesc* descs, int ndesc, const T* A, const T* B,
                    T* C, int M, int N, int K, int lda, int ldb, int ldc,
                    int opA, int opB, T alpha, T beta, hipStream_t stream) {
-  if (M % 128 || N % 64 || K % 16 || K <= 0) return 0;  // BK=8 needs K%16 too
+  static const int bm_env = [] {
+    const char* v = getenv("DLAF_GEMM_V2_BM");
+    return v ? atoi(v) : 128;
+  }();
+  const int bm = (bm_env == 64 && M % 64 == 0) ? 64 : 128;
+  if (M % bm || N % 64 || K % 16 || K <= 0) return 0;  // BK=8 needs K%16 too
   static const int enabled = [] {
     const char* v = getenv("DLAF_GEMM_V2");
     return v ? atoi(v) : 1;
@@ -541,7 +547,7 @@ int launch_v2_real(const GemmDesc* descs, int ndesc, const T* A, const T* B,
   if (!enabled) return 0;
   const int bn = (N % bn_env == 0) ? bn_env : (N % 64 == 0 ? 64 : 128);
   if (N % bn) return 0;
-  const int mblocks = M / 128, nblocks = N / bn;
+  const int mblocks = M / bm, nblocks = N / bn;
   const dim3 grid(ndesc * mblocks * nblocks);
   const dim3 block(256);
   const int oa = (opA == OP_C) ? OP_T : opA;
@@ -554,9 +560,16 @@ int launch_v2_real(const GemmDesc* descs, int ndesc, const T* A, const T* B,
   gemm_v2_k<T, OA, OB, SW, DP, BNv, BKv><<<grid, block, 0, stream>>>(      \
       descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks,        \
       nblocks)
+#define LV2M(OA, OB, SW)                                                   \
+  gemm_v2_k<T, OA, OB, SW, 2, 128, 16, 64><<<grid, block, 0, stream>>>(    \
+      descs, A, B, C, M, N, K, lda, ldb, ldc, alpha, beta, mblocks,        \
+      nblocks)
 #define CASE(OA, OB)                                                       \
   if (oa == OA && ob == OB) {                                              \
-    if (bk_env == 8 && bn == 64) {                                         \
+    if (bm == 64) {                                                        \
+      if (N % 128) return 0;                                               \
+      if (swz) LV2M(OA, OB, 1); else LV2M(OA, OB, 0);                      \
+    } else if (bk_env == 8 && bn == 64) {                                         \
       if (swz) LV2K(OA, OB, 1, 2, 64, 8); else LV2K(OA, OB, 0, 2, 64, 8);  \
     } else if (bn == 64) {                                                 \
       if (depth == 2) {                                                    \
@@ -575,6 +588,7 @@ int launch_v2_real(const GemmDesc* descs, int ndesc, const T* A, const T* B,
   }
   CASE(OP_N, OP_N) CASE(OP_N, OP_T) CASE(OP_T, OP_N) CASE(OP_T, OP_T)
 #undef CASE
+#undef LV2M
 #undef LV2K
 #undef LV2
   return 0;
