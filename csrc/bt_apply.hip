@@ -37,12 +37,11 @@ __device__ inline v4d mma_f64(double a, double b, v4d c) {
 
 // real f64: CW = 64, LDS = (R + G) * 64 * 8 = 160 KiB at R=192 ------------
 
-template <int G, int R>
+template <int G, int R, int CW = 64>
 __global__ __launch_bounds__(512, 1) void bt_group_f64(
-    double* __restrict__ E, int64_t nE, int64_t npad,
+    double* __restrict__ E, int64_t nE, int64_t ldE, int64_t npad,
     const double* __restrict__ V, const double* __restrict__ VTt,
     int64_t base0, int b, int nwin) {
-  constexpr int CW = 64;
   constexpr int NW = 8;                     // waves (512 threads)
   constexpr int NRF1 = (G / 16 + NW - 1) / NW;   // W1 row-frags/wave (1)
   constexpr int NRF2 = (R / 16 + NW - 1) / NW;   // update row-frags/wave (2)
@@ -64,10 +63,10 @@ __global__ __launch_bounds__(512, 1) void bt_group_f64(
       const int64_t grow = gbase + f;
       if (store) {
         if (grow < npad && col0 + c < nE)
-          E[grow * nE + col0 + c] = ring[slot * CW + c];
+          E[grow * ldE + col0 + c] = ring[slot * CW + c];
       } else {
         double v = 0.0;
-        if (grow < npad && col0 + c < nE) v = E[grow * nE + col0 + c];
+        if (grow < npad && col0 + c < nE) v = E[grow * ldE + col0 + c];
         ring[slot * CW + c] = v;
       }
     }
@@ -88,11 +87,11 @@ __global__ __launch_bounds__(512, 1) void bt_group_f64(
     // W1[g][c] = sum_h V[h][g] * ring[h][c]; each wave: NRF1 row-frags x 4
     // col-frags; ONE V load per row-frag per k-step shared by col-frags
     {
-      v4d acc[NRF1][4];
+      v4d acc[NRF1][CW / 16];
 #pragma unroll
       for (int j = 0; j < NRF1; ++j)
 #pragma unroll
-        for (int cf = 0; cf < 4; ++cf) acc[j][cf] = {0, 0, 0, 0};
+        for (int cf = 0; cf < CW / 16; ++cf) acc[j][cf] = {0, 0, 0, 0};
 #pragma unroll 4
       for (int h0 = 0; h0 < R; h0 += 4) {
         const int h = h0 + lk;
@@ -103,7 +102,7 @@ __global__ __launch_bounds__(512, 1) void bt_group_f64(
         for (int j = 0; j < NRF1; ++j)
           a[j] = Vk[(int64_t)h * G + (w + NW * j) * 16 + li];
 #pragma unroll
-        for (int cf = 0; cf < 4; ++cf) {
+        for (int cf = 0; cf < CW / 16; ++cf) {
           const double bb = ring[slot * CW + cf * 16 + li];
 #pragma unroll
           for (int j = 0; j < NRF1; ++j)
@@ -113,7 +112,7 @@ __global__ __launch_bounds__(512, 1) void bt_group_f64(
 #pragma unroll
       for (int j = 0; j < NRF1; ++j)
 #pragma unroll
-        for (int cf = 0; cf < 4; ++cf)
+        for (int cf = 0; cf < CW / 16; ++cf)
 #pragma unroll
           for (int r = 0; r < 4; ++r)
             W1[((w + NW * j) * 16 + lk + 4 * r) * CW + cf * 16 + li] =
@@ -125,11 +124,11 @@ __global__ __launch_bounds__(512, 1) void bt_group_f64(
     {
       const int nfr = R / 16;
       const int nj = (nfr - w + NW - 1) / NW;
-      v4d acc[NRF2][4];
+      v4d acc[NRF2][CW / 16];
 #pragma unroll
       for (int j = 0; j < NRF2; ++j)
 #pragma unroll
-        for (int cf = 0; cf < 4; ++cf) acc[j][cf] = {0, 0, 0, 0};
+        for (int cf = 0; cf < CW / 16; ++cf) acc[j][cf] = {0, 0, 0, 0};
 #pragma unroll 4
       for (int g0 = 0; g0 < G; g0 += 4) {
         double a[NRF2];
@@ -139,7 +138,7 @@ __global__ __launch_bounds__(512, 1) void bt_group_f64(
                      ? VTk[(int64_t)(g0 + lk) * R + (w + NW * j) * 16 + li]
                      : 0.0;
 #pragma unroll
-        for (int cf = 0; cf < 4; ++cf) {
+        for (int cf = 0; cf < CW / 16; ++cf) {
           const double bb = W1[(g0 + lk) * CW + cf * 16 + li];
 #pragma unroll
           for (int j = 0; j < NRF2; ++j)
@@ -149,7 +148,7 @@ __global__ __launch_bounds__(512, 1) void bt_group_f64(
       for (int j = 0; j < nj; ++j) {
         const int f = w + NW * j;
 #pragma unroll
-        for (int cf = 0; cf < 4; ++cf)
+        for (int cf = 0; cf < CW / 16; ++cf)
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             int slot = sbase + f * 16 + lk + 4 * r;
@@ -343,19 +342,37 @@ int bt_apply_group_f64(double* E, int64_t nE, int64_t npad, const double* V,
                        const double* VTt, int64_t base0, int b, int G, int R,
                        int nwin, hipStream_t stream) {
   if (G != 128 || b % 16 || b > 64 || R % 16 || R % b || nwin <= 0) return 0;
-  const size_t sh = ((size_t)R + G) * 64 * sizeof(double);
-  if (sh > 160 * 1024) return 0;
-  const int blocks = (int)((nE + 63) / 64);
+  const size_t sh64 = ((size_t)R + G) * 64 * sizeof(double);
+  if (sh64 > 160 * 1024) return 0;
+  // Tail split: Q = ceil(nE/64) column chunks; the last partial CU pass
+  // would idle most of the chip, so the remainder columns run as a second
+  // CW=32 launch (2 WGs/CU, half the per-WG work) — measured: 313 chunks
+  // as 2 full CW64 passes lost to the GEMM chain; 256 + 114x32 wins.
+  const int64_t Q = (nE + 63) / 64;
+  int64_t main_cols = nE;
+  if (Q > 256 && (Q % 256) != 0) main_cols = (Q / 256) * 256 * 64;
+  const int64_t rest_cols = nE - main_cols;
+  const size_t sh32 = ((size_t)R + G) * 32 * sizeof(double);
 #define BT_CASE(RT)                                                        \
   if (R == RT) {                                                           \
-    const void* fp = (const void*)bt_group_f64<128, RT>;                   \
-    if (sh > 65536)                                                        \
-      (void)hipFuncSetAttribute(fp,                                        \
-                                hipFuncAttributeMaxDynamicSharedMemorySize,\
-                                (int)sh);                                  \
-    bt_group_f64<128, RT><<<blocks, 512, sh, stream>>>(E, nE, npad, V,     \
-                                                       VTt, base0, b,      \
-                                                       nwin);              \
+    if (main_cols > 0) {                                                   \
+      const void* fp = (const void*)bt_group_f64<128, RT, 64>;             \
+      if (sh64 > 65536)                                                    \
+        (void)hipFuncSetAttribute(                                         \
+            fp, hipFuncAttributeMaxDynamicSharedMemorySize, (int)sh64);    \
+      bt_group_f64<128, RT, 64>                                            \
+          <<<(int)((main_cols + 63) / 64), 512, sh64, stream>>>(           \
+              E, main_cols, nE, npad, V, VTt, base0, b, nwin);             \
+    }                                                                      \
+    if (rest_cols > 0) {                                                   \
+      const void* fp = (const void*)bt_group_f64<128, RT, 32>;             \
+      if (sh32 > 65536)                                                    \
+        (void)hipFuncSetAttribute(                                         \
+            fp, hipFuncAttributeMaxDynamicSharedMemorySize, (int)sh32);    \
+      bt_group_f64<128, RT, 32>                                            \
+          <<<(int)((rest_cols + 31) / 32), 512, sh32, stream>>>(           \
+              E + main_cols, rest_cols, nE, npad, V, VTt, base0, b, nwin); \
+    }                                                                      \
     return 1;                                                              \
   }
   BT_CASE(192) BT_CASE(176) BT_CASE(160) BT_CASE(144)

@@ -288,17 +288,11 @@ def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
             # workgroup owns a 16-column slice and marches the full window
             # chain in LDS — replaces ~3*nwin GEMM launches per group
             # (launch-bound, measured) with ONE.
-            # Utilization gate: one workgroup owns 64 columns; with
-            # ceil(nE/64) chunks over 256 CUs the last partial pass idles
-            # CUs (n=20000: 313 chunks = 1.22 passes, measured slower than
-            # the GEMM chain; 256 chunks = 1.0 pass, measured faster).
-            # Complex measured slower throughout (1 WG/CU, 4x MFMA) - torch
-            # path keeps it.
-            _chunks = -(-nE // 64)
-            _util = _chunks / (256 * -(-_chunks // 256))
+            # f64 only: the kernel splits partial CU passes into a CW=32
+            # tail launch itself. Complex measured slower throughout
+            # (1 WG/CU, 4x MFMA) - the torch chain keeps it.
             if (dev.type == "cuda" and E.dtype == torch.float64
                     and G % 32 == 0 and b % 16 == 0
-                    and (_chunks <= 256 or _util >= 0.9)
                     and _os.environ.get("DLAF_BT_KERNEL", "1") != "0"):
                 R = -(-H // b) * b
                 if R % 16 == 0:
