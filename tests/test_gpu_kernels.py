@@ -269,3 +269,30 @@ def test_bt_apply_group_kernel_vs_torch(tc):
     err = (outs["0"] - outs["1"]).abs().max().item()
     scale = outs["0"].abs().max().item() + 1
     assert err < 1e-11 * scale * n, err
+
+
+def test_bt_apply_group_tail_split():
+    """The CW=32 tail launch (column chunks > 256) against the torch chain."""
+    import os
+    from dlaf_amd.algs import band2tridiag as b2t
+    import sys
+    sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "tools"))
+    from bench_chase_gpu import make_band
+    n, b = 500, 16
+    os.environ["DLAF_GPU_CHASE"] = "0"
+    band = make_band(n, b, torch.float64, "cuda")
+    tri = b2t.chase_band(band, b)
+    nE = 16448  # 257 64-col chunks -> 256 CW64 + 2 CW32
+    E0 = torch.randn(n, nE, dtype=torch.float64, device="cuda")
+    outs = {}
+    for mode in ("0", "1"):
+        os.environ["DLAF_BT_KERNEL"] = mode
+        E = E0.clone()
+        b2t.bt_band_to_tridiagonal(E, tri)
+        torch.cuda.synchronize()
+        outs[mode] = E.cpu()
+    os.environ.pop("DLAF_BT_KERNEL", None)
+    err = (outs["0"] - outs["1"]).abs().max().item()
+    scale = outs["0"].abs().max().item() + 1
+    assert err < 1e-11 * scale * n, err
