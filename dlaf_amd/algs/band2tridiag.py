@@ -56,17 +56,17 @@ def chase_band(store: torch.Tensor, b: int) -> TridiagResult:
     """Run the bulge chase on a compact band ``store`` [n, 2b] (consumed in
     place) and package the TridiagResult.
 
-    GPU path (``csrc/chase_gpu.hip``, default for complex dtypes,
-    ``DLAF_GPU_CHASE=1/0`` forces):
-    persistent wavefront workgroups chase ~n/3b sweeps concurrently with the
-    band, reflectors and flags resident in HBM — the reference keeps this
-    stage on the host (``band_to_tridiag/mc.h:666-693``). Deterministic
-    (fixed wave-reduction order), so rank-replicated distributed use stays
-    lockstep-identical; bounded spins surface scheduling pathologies as a
-    CPU-fallback warning, never a hang. The pipeline is handoff-fence-bound
-    (~unit-math-independent), which makes it ~1.7x slower than the CPU
-    wavefront for f64 but ~1.8x FASTER for complex128 (4x flops per unit:
-    10.6 s CPU vs 6.0 s GPU at n=16384) — hence the dtype-dependent default.
+    GPU path (``csrc/chase_gpu.hip``, the DEFAULT on device since round 2;
+    ``DLAF_GPU_CHASE=1/0`` forces): persistent wavefront workgroups, one
+    UNIT on 256 threads (4 waves splitting the q/p loops with LDS partial
+    reductions — the round-1 single-wave form left ~1 wave per CU busy),
+    with the band, reflectors and flags resident in HBM; sc1 write-through
+    window publish. The reference keeps this stage on the host
+    (``band_to_tridiag/mc.h:666-693``). Deterministic (fixed reduction
+    order), so rank-replicated distributed use stays lockstep-identical;
+    bounded spins surface scheduling pathologies as a CPU-fallback warning,
+    never a hang. Measured n=20000 f64: 3.6 s vs 4.1 s CPU wavefront;
+    n=16384 c128: 2.9 s vs 10.8 s.
     """
     from ..ops._ext import get_ext
     n = store.shape[0]
@@ -78,11 +78,10 @@ def chase_band(store: torch.Tensor, b: int) -> TridiagResult:
     total = int(counts.sum().item())
     vstore = None
     env = os.environ.get("DLAF_GPU_CHASE")
-    # default: GPU chase for COMPLEX dtypes (4x the flops make the CPU
-    # wavefront the bottleneck: 10.6 s vs 6.0 s GPU at n=16384), CPU
-    # wavefront for real (3.9 s vs 6.7 s GPU at n=20000 — the GPU pipeline
-    # is handoff-fence-bound, see docs/DESIGN.md)
-    want_gpu = is_cplx if env is None else env == "1"
+    # default: GPU chase for every dtype (round-2 256-thread units +
+    # write-through publish: f64 3.6 s vs 4.1 CPU, c128 2.9 vs 10.8 at the
+    # BASELINE shapes; docs/DESIGN.md has the progression)
+    want_gpu = True if env is None else env == "1"
     if n > 2 and store.is_cuda and b <= 64 and want_gpu:
         dev = store.device
         vstore = torch.zeros((max(total, 1), b + 1), dtype=store.dtype, device=dev)
