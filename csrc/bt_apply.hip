@@ -170,13 +170,14 @@ __global__ __launch_bounds__(512, 1) void bt_group_f64(
 // complex c128: CW = 32, LDS = (R + G) * 32 * 16 = 160 KiB at R=192 -------
 
 template <int G, int R>
-__global__ __launch_bounds__(256, 1) void bt_group_c128(
+__global__ __launch_bounds__(512, 1) void bt_group_c128(
     double* __restrict__ E, int64_t nE, int64_t npad,
     const double* __restrict__ V, const double* __restrict__ VTt,
     int64_t base0, int b, int nwin) {
   constexpr int CW = 32;
-  constexpr int NRF1 = (G / 16) / 4;
-  constexpr int NRF2 = (R / 16 + 3) / 4;
+  constexpr int NW = 8;                          // waves (512 threads)
+  constexpr int NRF1 = (G / 16 + NW - 1) / NW;   // 1
+  constexpr int NRF2 = (R / 16 + NW - 1) / NW;   // <=2
   extern __shared__ double S[];  // ring[R][CW](re,im) then W1[G][CW]
   double* ring = S;
   double* W1 = S + (int64_t)R * CW * 2;
@@ -187,7 +188,7 @@ __global__ __launch_bounds__(256, 1) void bt_group_c128(
   const int64_t col0 = (int64_t)blockIdx.x * CW;
 
   auto move_rows = [&](int64_t gbase, int sbase, int f0, int cnt, bool store) {
-    for (int f = f0 + tid / CW; f < f0 + cnt; f += 256 / CW) {
+    for (int f = f0 + tid / CW; f < f0 + cnt; f += (NW * 64) / CW) {
       const int c = tid % CW;
       int slot = sbase + f;
       if (slot >= R) slot -= R;
@@ -240,7 +241,7 @@ __global__ __launch_bounds__(256, 1) void bt_group_c128(
         double avr[NRF1], avi[NRF1];
 #pragma unroll
         for (int j = 0; j < NRF1; ++j) {
-          const int64_t va = ((int64_t)h * G + (w + 4 * j) * 16 + li) * 2;
+          const int64_t va = ((int64_t)h * G + (w + NW * j) * 16 + li) * 2;
           avr[j] = Vk[va];
           avi[j] = Vk[va + 1];
         }
@@ -264,7 +265,7 @@ __global__ __launch_bounds__(256, 1) void bt_group_c128(
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             const int64_t o =
-                (((w + 4 * j) * 16 + lk + 4 * r) * CW + cf * 16 + li) * 2;
+                (((w + NW * j) * 16 + lk + 4 * r) * CW + cf * 16 + li) * 2;
             W1[o] = ar[j][cf][r];
             W1[o + 1] = ai[j][cf][r];
           }
@@ -273,7 +274,7 @@ __global__ __launch_bounds__(256, 1) void bt_group_c128(
 
     {
       const int nfr = R / 16;
-      const int nj = (nfr - w + 3) / 4;
+      const int nj = (nfr - w + NW - 1) / NW;
       v4d ar[NRF2][2], ai[NRF2][2];
 #pragma unroll
       for (int j = 0; j < NRF2; ++j)
@@ -289,7 +290,7 @@ __global__ __launch_bounds__(256, 1) void bt_group_c128(
         for (int j = 0; j < NRF2; ++j) {
           if (j < nj) {
             const int64_t va =
-                ((int64_t)(g0 + lk) * R + (w + 4 * j) * 16 + li) * 2;
+                ((int64_t)(g0 + lk) * R + (w + NW * j) * 16 + li) * 2;
             avr[j] = VTk[va];
             avi[j] = VTk[va + 1];
           } else {
@@ -311,7 +312,7 @@ __global__ __launch_bounds__(256, 1) void bt_group_c128(
         }
       }
       for (int j = 0; j < nj; ++j) {
-        const int f = w + 4 * j;
+        const int f = w + NW * j;
 #pragma unroll
         for (int cf = 0; cf < 2; ++cf)
 #pragma unroll
@@ -394,7 +395,7 @@ int bt_apply_group_c128(double* E, int64_t nE, int64_t npad, const double* V,
       (void)hipFuncSetAttribute(fp,                                        \
                                 hipFuncAttributeMaxDynamicSharedMemorySize,\
                                 (int)sh);                                  \
-    bt_group_c128<128, RT><<<blocks, 256, sh, stream>>>(E, nE, npad, V,    \
+    bt_group_c128<128, RT><<<blocks, 512, sh, stream>>>(E, nE, npad, V,    \
                                                         VTt, base0, b,     \
                                                         nwin);             \
     return 1;                                                              \

@@ -287,10 +287,10 @@ def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
             # workgroup owns a 16-column slice and marches the full window
             # chain in LDS — replaces ~3*nwin GEMM launches per group
             # (launch-bound, measured) with ONE.
-            # f64 only: the kernel splits partial CU passes into a CW=32
-            # tail launch itself. Complex measured slower throughout
-            # (1 WG/CU, 4x MFMA) - the torch chain keeps it.
-            if (dev.type == "cuda" and E.dtype == torch.float64
+            # f64: always (the kernel splits partial CU passes into a CW=32
+            # tail launch). c128: 512-thread round-2 kernel.
+            if (dev.type == "cuda"
+                    and E.dtype in (torch.float64, torch.complex128)
                     and G % 32 == 0 and b % 16 == 0
                     and _os.environ.get("DLAF_BT_KERNEL", "1") != "0"):
                 R = -(-H // b) * b
