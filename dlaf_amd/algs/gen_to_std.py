@@ -56,9 +56,14 @@ def _hegst_diag_tile(a: torch.Tensor, l: torch.Tensor) -> None:
 
 def generalized_to_standard(uplo: UpLo, mat_a: Matrix, mat_l: Matrix,
                             grid: Optional[CommGrid] = None) -> None:
-    """In-place HEGST (itype=1): A <- inv(L) A inv(L)^H, Lower storage."""
+    """In-place HEGST (itype=1): A <- inv(L) A inv(L)^H (Lower) or
+    A <- inv(U)^H A inv(U) (Upper storage, native on the local path)."""
     if uplo == UpLo.Upper:
-        # A <- U^-H A U^-1 == (with L = U^H) L^-1 A L^-H on transposed storage
+        g = grid if grid is not None else mat_a.grid
+        if g is None or not g.distributed:
+            _hegst_local_upper(mat_a, mat_l)   # native, no transposes
+            return
+        # distributed Upper: storage-transpose reduction (future work)
         from ._uplo import transpose_storage
         transpose_storage(mat_a)
         transpose_storage(mat_l)
@@ -180,3 +185,64 @@ def generalized_to_standard(uplo: UpLo, mat_a: Matrix, mat_l: Matrix,
             ops.gemm_items(mat_a.storage, mat_l.storage, hA,
                            [(o, lo, 0) for o, lo in zip(offs, loffs)], nb,
                            Op.NoTrans, Op.NoTrans, -0.5, 1.0, uniform=True)
+
+
+def _herm_full_u(t: torch.Tensor) -> torch.Tensor:
+    """Full Hermitian tile from UPPER storage."""
+    return torch.triu(t) + torch.triu(t, 1).mH
+
+
+def _hegst_local_upper(mat_a: Matrix, mat_u: Matrix) -> None:
+    """Native Upper HEGST (itype=1, local): A <- U^{-H} A U^{-1} on
+    upper-stored tiles, the LAPACK zhegst Upper blocked order (reference
+    counterpart ``eigensolver/gen_to_std/impl.h`` call_U; the Lower path
+    uses the deferred-TRSM reshuffle — Upper is correctness-grade per-tile
+    ops, no storage transposes).
+
+    Per tile step k (B12 = U[k, k+1:], A12 = A[k, k+1:]):
+      1. A_kk <- Ukk^{-H} herm(A_kk) Ukk^{-1}
+      2. A12 <- Ukk^{-H} A12
+      3. A12 -= 1/2 herm(A_kk_new) B12
+      4. A22 -= A12^H B12 + B12^H A12    (upper tiles only)
+      5. A12 -= 1/2 herm(A_kk_new) B12
+      6. A12 <- A12 U22^{-1}             (forward column sweep over U22)
+    """
+    da, du = mat_a.dist, mat_u.dist
+    assert da.m == da.n and da.mb == da.nb
+    assert (du.m, du.n, du.mb, du.nb) == (da.m, da.n, da.mb, da.nb)
+    nt = da.nr_tiles[0]
+    sta, stu = mat_a.storage, mat_u.storage
+
+    for k in range(nt):
+        Ukk = torch.triu(stu[k, k])
+        # 1) diagonal transform (solve twice against the upper factor)
+        Ah = _herm_full_u(sta[k, k])
+        T1 = torch.linalg.solve_triangular(Ukk.mH, Ah, upper=False)
+        sta[k, k] = torch.linalg.solve_triangular(
+            Ukk.mH, T1.mH, upper=False).mH
+        if k + 1 >= nt:
+            break
+        Akk_h = _herm_full_u(sta[k, k])
+        # 2) A12 <- Ukk^{-H} A12
+        for j in range(k + 1, nt):
+            sta[k, j] = torch.linalg.solve_triangular(Ukk.mH, sta[k, j],
+                                                      upper=False)
+        # 3) A12 -= 1/2 Akk B12
+        for j in range(k + 1, nt):
+            sta[k, j] -= 0.5 * (Akk_h @ stu[k, j])
+        # 4) A22 -= A12^H B12 + B12^H A12 (upper triangle of tiles)
+        for i in range(k + 1, nt):
+            for j in range(i, nt):
+                sta[i, j] -= sta[k, i].mH @ stu[k, j] + \
+                    stu[k, i].mH @ sta[k, j]
+        # 5) second half-update
+        for j in range(k + 1, nt):
+            sta[k, j] -= 0.5 * (Akk_h @ stu[k, j])
+        # 6) A12 <- A12 U22^{-1}: forward sweep over the trailing columns
+        for j in range(k + 1, nt):
+            acc = sta[k, j].clone()
+            for i in range(k + 1, j):
+                acc -= sta[k, i] @ stu[i, j]
+            sta[k, j] = torch.linalg.solve_triangular(
+                torch.triu(stu[j, j]).mH, acc.mH, upper=False).mH
+        del Akk_h

@@ -77,3 +77,31 @@ def test_hegst_upper_local():
     got = torch.triu(got) + torch.triu(got, 1).mH
     err = (got - want).abs().max().item()
     assert err < 1e-9 * n, f"err={err}"
+
+
+def test_hegst_upper_native_local():
+    """Native Upper HEGST (no storage transposes) vs dense reference."""
+    import torch
+    from dlaf_amd import Matrix, UpLo, generalized_to_standard
+    from dlaf_amd.matrix import util as mutil
+    for dtype in (torch.float64, torch.complex128):
+        for n, nb in ((96, 32), (130, 48)):
+            a = Matrix.create(n, n, nb, nb, dtype=dtype)
+            mutil.set_random_hermitian(a, seed=3)
+            A = a.to_global()
+            A = torch.tril(A) + torch.tril(A, -1).mH
+            a.set_from_global(A.clone())
+            u = Matrix.create(n, n, nb, nb, dtype=dtype)
+            mutil.set_random_hermitian_positive_definite(u, seed=4)
+            B = u.to_global()
+            B = torch.tril(B) + torch.tril(B, -1).mH
+            U = torch.linalg.cholesky(B, upper=True)
+            u.set_from_global(U.clone())
+            generalized_to_standard(UpLo.Upper, a, u)
+            got = a.to_global()
+            got = torch.triu(got) + torch.triu(got, 1).mH
+            Ui = torch.linalg.solve_triangular(
+                U, torch.eye(n, dtype=dtype), upper=True)
+            want = Ui.mH @ A @ Ui
+            err = (got - want).abs().max().item()
+            assert err < 1e-11 * n, (dtype, n, nb, err)

@@ -332,3 +332,30 @@ def test_cholesky_upper_native_gpu(dtype):
     err = (U.mH @ U - A).abs().max().item()
     scale = A.abs().max().item()
     assert err < 1e-10 * n * scale, err
+
+
+def test_hegst_upper_native_gpu():
+    """Native Upper HEGST on device."""
+    from dlaf_amd import Matrix, UpLo, generalized_to_standard
+    from dlaf_amd.matrix import util as mutil
+    n, nb = 1024, 256
+    dtype = torch.complex128
+    a = Matrix.create(n, n, nb, nb, dtype=dtype, device="cuda")
+    mutil.set_random_hermitian(a, seed=5)
+    A = a.to_global()
+    A = torch.tril(A) + torch.tril(A, -1).mH
+    a.set_from_global(A.clone())
+    u = Matrix.create(n, n, nb, nb, dtype=dtype, device="cuda")
+    mutil.set_random_hermitian_positive_definite(u, seed=6)
+    B = u.to_global()
+    B = torch.tril(B) + torch.tril(B, -1).mH
+    U = torch.linalg.cholesky(B, upper=True)
+    u.set_from_global(U.clone())
+    generalized_to_standard(UpLo.Upper, a, u)
+    got = a.to_global()
+    got = torch.triu(got) + torch.triu(got, 1).mH
+    Ui = torch.linalg.solve_triangular(
+        U, torch.eye(n, dtype=dtype, device="cuda"), upper=True)
+    want = Ui.mH @ A @ Ui
+    err = (got - want).abs().max().item()
+    assert err < 1e-10 * n, err
