@@ -126,8 +126,11 @@ __device__ inline void derive(R xn2, T alpha, T& tau, T& scale, R& beta,
   }
 }
 
-// Workspace: norms[nb] (real, zeroed); wraw[2 * nb * nb] (T, zeroed):
-// row j holds the w_raw sums, row nb + j stages {alpha at [j], P[j,q] at [q]}.
+// Workspace: norms[nb] (real, zeroed); wraw[3 * nb * nb] (T, zeroed):
+// row j holds the w_raw sums (parity-half 0), row nb + j stages {alpha at
+// [j], P[j,q] at [q]}, row 2nb + j the parity-half-1 sums (phase A runs TWO
+// reducer blocks per column — one per row parity — so the 128-block grid is
+// not half idle on a <=64-column panel).
 template <class T>
 __global__ void panel_qr_kernel(T* __restrict__ P, long m, int nb, long ldp,
                                 T* __restrict__ taus,
@@ -144,17 +147,22 @@ __global__ void panel_qr_kernel(T* __restrict__ P, long m, int nb, long ldp,
   for (int j = 0; j < ncols; ++j) {
     T* wj = wraw + (long)j * nb;
     T* wrow = wraw + (long)(nb + j) * nb;
+    T* whalf = wraw + (long)(2 * nb + j) * nb;
     // ---- phase A: each block OWNS whole output values (no cross-block
     // atomics — an atomics formulation serializes on <= nb addresses and
-    // measured 48 ms/panel). Block q' reduces column q = j + q' over all
-    // tail rows; q' == 0 computes the norm of column j instead. ----
+    // measured 48 ms/panel). Block task bq covers column q = j + bq/2,
+    // row-parity half h = bq & 1; h == 0 writes wj/norms, h == 1 whalf
+    // (summed in phase B). q == j computes the column norm instead. ----
     {
       __shared__ R sredr[256];
       __shared__ T sredt[256];
-      for (int q = j + (int)blockIdx.x; q < nb; q += gridDim.x) {
+      const int ntasks = 2 * (nb - j);
+      for (int bq = (int)blockIdx.x; bq < ntasks; bq += gridDim.x) {
+        const int q = j + (bq >> 1);
+        const int h = bq & 1;
         if (q == j) {
           R part = R(0);
-          for (long i = j + 1 + tid; i < m; i += nthreads)
+          for (long i = j + 1 + h + 2 * (long)tid; i < m; i += 2 * nthreads)
             part += abs2v(P[i * ldp + j]);
           sredr[tid] = part;
           __syncthreads();
@@ -162,10 +170,15 @@ __global__ void panel_qr_kernel(T* __restrict__ P, long m, int nb, long ldp,
             if (tid < s) sredr[tid] += sredr[tid + s];
             __syncthreads();
           }
-          if (tid == 0) norms[j] = sredr[0];
+          if (tid == 0) {
+            if (h == 0)
+              norms[j] = sredr[0];
+            else
+              whalf[j] = ScalarTraits<T>::from_real(sredr[0]);
+          }
         } else {
           T part = ScalarTraits<T>::zero();
-          for (long i = j + 1 + tid; i < m; i += nthreads)
+          for (long i = j + 1 + h + 2 * (long)tid; i < m; i += 2 * nthreads)
             part += conjv(P[i * ldp + j]) * P[i * ldp + q];
           sredt[tid] = part;
           __syncthreads();
@@ -173,7 +186,12 @@ __global__ void panel_qr_kernel(T* __restrict__ P, long m, int nb, long ldp,
             if (tid < s) sredt[tid] += sredt[tid + s];
             __syncthreads();
           }
-          if (tid == 0) wj[q] = sredt[0];
+          if (tid == 0) {
+            if (h == 0)
+              wj[q] = sredt[0];
+            else
+              whalf[q] = sredt[0];
+          }
         }
         __syncthreads();
       }
@@ -189,14 +207,15 @@ __global__ void panel_qr_kernel(T* __restrict__ P, long m, int nb, long ldp,
       T tau, scale;
       R beta;
       bool degen;
-      derive<T, R>(norms[j], wrow[j], tau, scale, beta, degen);
+      derive<T, R>(norms[j] + ScalarTraits<T>::real(whalf[j]), wrow[j], tau,
+                   scale, beta, degen);
       if (!degen) {
         T ctau = conjv(tau);
         T cscale = conjv(scale);
         for (long idx = gid0; idx < (m - j) * (long)(nb - j - 1); idx += gstride) {
           long i = j + idx / (nb - j - 1);
           int q = j + 1 + (int)(idx % (nb - j - 1));
-          T wq = wrow[q] + cscale * wj[q];
+          T wq = wrow[q] + cscale * (wj[q] + whalf[q]);
           T vi = (i == j) ? ScalarTraits<T>::from_real(R(1))
                           : P[i * ldp + j] * scale;
           P[i * ldp + q] = P[i * ldp + q] - ctau * vi * wq;
@@ -208,10 +227,12 @@ __global__ void panel_qr_kernel(T* __restrict__ P, long m, int nb, long ldp,
   // ---- epilogue: per-column tail scaling + beta/tau writes ----
   for (int j = 0; j < ncols; ++j) {
     T* wrow = wraw + (long)(nb + j) * nb;
+    T* whalf = wraw + (long)(2 * nb + j) * nb;
     T tau, scale;
     R beta;
     bool degen;
-    derive<T, R>(norms[j], wrow[j], tau, scale, beta, degen);
+    derive<T, R>(norms[j] + ScalarTraits<T>::real(whalf[j]), wrow[j], tau,
+                 scale, beta, degen);
     if (!degen) {
       for (long i = j + 1 + gid0; i < m; i += gstride)
         P[i * ldp + j] = P[i * ldp + j] * scale;

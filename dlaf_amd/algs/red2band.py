@@ -103,7 +103,7 @@ def panel_qr_(P: torch.Tensor, taus: torch.Tensor) -> None:
         rdt = torch.float32 if P.dtype in (torch.float32, torch.complex64) \
             else torch.float64
         norms = torch.zeros(nb, dtype=rdt, device=P.device)
-        wraw = torch.zeros(2 * nb * nb, dtype=P.dtype, device=P.device)
+        wraw = torch.zeros(3 * nb * nb, dtype=P.dtype, device=P.device)
         get_ext().panel_qr(P, taus, norms, wraw)
         return
     one = torch.ones(1, dtype=P.dtype, device=P.device)
