@@ -215,16 +215,26 @@ def bt_band_to_tridiagonal(E: torch.Tensor, tri: TridiagResult,
         _prep_ctx = (torch.cuda.stream(streams[0]) if use_strips
                      else _nullcontext())
         with _prep_ctx:
-            # staircase V panels for all windows: Vg[k, row, g]
+            # staircase V panels for all windows: Vg[k, row, g]. Assembled
+            # with ONE gather+scatter per group: the per-sweep slice loop
+            # (2 copies + 2 host syncs per sweep) was ~40k tiny device
+            # copies per f64 n=20000 solve (profiles/
+            # bench_full_r2_kernel_stats.md: 199740 __amd_rocclr_copyBuffer)
             Vg = torch.zeros((nwin, H, G), dtype=E.dtype, device=dev)
             taus_g = torch.zeros((nwin, G), dtype=E.dtype, device=dev)
-            for g in range(Gc):
-                Kg = int(counts_g[g])
-                if Kg == 0:
-                    continue
-                o = int(offsets[s0 + g])
-                taus_g[:Kg, g] = V[o:o + Kg, 0]
-                Vg[:Kg, g:g + b, g] = V[o:o + Kg, 1:]
+            cg = counts_g.long()
+            gidx = torch.repeat_interleave(torch.arange(Gc, device=dev), cg)
+            if gidx.numel():
+                starts = torch.cumsum(cg, 0) - cg
+                kidx = (torch.arange(gidx.numel(), device=dev)
+                        - starts[gidx])
+                srow = offsets[s0:s0 + Gc].long()[gidx] + kidx
+                taus_g[kidx, gidx] = V[srow, 0]
+                # Vg[kidx, gidx + r, gidx] for r in [0, b)
+                rr = torch.arange(b, device=dev)
+                dst = ((kidx * H + gidx) * G + gidx).unsqueeze(1) \
+                    + rr.unsqueeze(0) * G
+                Vg.view(-1)[dst.reshape(-1)] = V[srow, 1:].reshape(-1)
             # batched T factors: T = inv(diag(1/tau) + striu(V^H V)); tau=0 rows/cols vanish
             Gram = Vg.mH @ Vg                                   # [nwin, G, G]
             zc = taus_g == 0
