@@ -234,13 +234,36 @@ def _merge(w1, Q1, w2, Q2, rho, device):
     rot_cols = sorted({c for (i, j, _, _) in rots for c in (i, j)})
     rot_pos = {c: p for p, c in enumerate(rot_cols)}
     R = gather_cols(np.array(rot_cols, dtype=np.int64))
-    for (i, j, c, s) in rots:
-        pi, pj = rot_pos[i], rot_pos[j]
-        gi = R[:, pi].clone()
-        gj = R[:, pj].clone()
-        # R <- R G with G = [[c, -s], [s, c]]; z' = G^T z zeroes component i
-        R[:, pi] = c * gi + s * gj
-        R[:, pj] = -s * gi + c * gj
+    # R <- R G per rotation, G = [[c, -s], [s, c]] (z' = G^T z zeroes
+    # component i). Rotations sharing a column must apply in order, but
+    # disjoint ones commute: greedy-schedule into ROUNDS of column-disjoint
+    # pairs and apply each round as one batched 4-kernel update (the
+    # per-rotation loop was 4 tiny device launches each; same class of
+    # host-bound overhead as the bt prep fix, profiles/
+    # bench_full_r2_kernel_stats.md).
+    rounds: list = []
+    last: dict = {}
+    for rot in rots:
+        i, j = rot[0], rot[1]
+        r = max(last.get(i, 0), last.get(j, 0))
+        if r == len(rounds):
+            rounds.append([])
+        rounds[r].append(rot)
+        last[i] = last[j] = r + 1
+    rdt = R.real.dtype if R.is_complex() else R.dtype
+    for rnd in rounds:
+        pi = torch.tensor([rot_pos[i] for (i, _, _, _) in rnd],
+                          dtype=torch.int64, device=device)
+        pj = torch.tensor([rot_pos[j] for (_, j, _, _) in rnd],
+                          dtype=torch.int64, device=device)
+        cc = torch.tensor([c for (_, _, c, _) in rnd], dtype=rdt,
+                          device=device)
+        ss = torch.tensor([s for (_, _, _, s) in rnd], dtype=rdt,
+                          device=device)
+        gi = R[:, pi]
+        gj = R[:, pj]
+        R[:, pi] = cc * gi + ss * gj
+        R[:, pj] = -ss * gi + cc * gj
 
     def patch_rotated(out, cols_np):
         pos = [(p, rot_pos[c]) for p, c in enumerate(cols_np) if c in rot_pos]
