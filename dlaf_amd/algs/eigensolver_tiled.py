@@ -434,31 +434,42 @@ def dc_striped(d: torch.Tensor, e: torch.Tensor, group, rank: int, world: int,
             # a row must stay ordered; disjoint ones commute — greedy
             # rounds of row-disjoint pairs, one batched update per round
             # (same host-bound-loop fix as the local merge / bt prep).
-            rounds: list = []
-            last: dict = {}
-            for r_i in range(nrot - 1, -1, -1):
-                i_r = int(rots[r_i, 0])
-                j_r = int(rots[r_i, 1])
-                rr = max(last.get(i_r, 0), last.get(j_r, 0))
-                if rr == len(rounds):
-                    rounds.append([])
-                rounds[rr].append((i_r, j_r, float(rots[r_i, 2]),
-                                   float(rots[r_i, 3])))
-                last[i_r] = last[j_r] = rr + 1
-            rdt = C.real.dtype if C.is_complex() else C.dtype
-            for rnd in rounds:
-                ii = torch.tensor([i for (i, _, _, _) in rnd],
-                                  dtype=torch.int64, device=device)
-                jj = torch.tensor([j for (_, j, _, _) in rnd],
-                                  dtype=torch.int64, device=device)
-                cc = torch.tensor([c for (_, _, c, _) in rnd], dtype=rdt,
-                                  device=device).unsqueeze(1)
-                ss = torch.tensor([s for (_, _, _, s) in rnd], dtype=rdt,
-                                  device=device).unsqueeze(1)
-                ri = C[ii]
-                rj = C[jj]
-                C[ii] = cc * ri - ss * rj
-                C[jj] = ss * ri + cc * rj
+            if os.environ.get("DLAF_DC_ROT_BATCH", "0") != "0":
+                rounds: list = []
+                last: dict = {}
+                for r_i in range(nrot - 1, -1, -1):
+                    i_r = int(rots[r_i, 0])
+                    j_r = int(rots[r_i, 1])
+                    rr = max(last.get(i_r, 0), last.get(j_r, 0))
+                    if rr == len(rounds):
+                        rounds.append([])
+                    rounds[rr].append((i_r, j_r, float(rots[r_i, 2]),
+                                       float(rots[r_i, 3])))
+                    last[i_r] = last[j_r] = rr + 1
+                rdt = C.real.dtype if C.is_complex() else C.dtype
+                for rnd in rounds:
+                    ii = torch.tensor([i for (i, _, _, _) in rnd],
+                                      dtype=torch.int64, device=device)
+                    jj = torch.tensor([j for (_, j, _, _) in rnd],
+                                      dtype=torch.int64, device=device)
+                    cc = torch.tensor([c for (_, _, c, _) in rnd], dtype=rdt,
+                                      device=device).unsqueeze(1)
+                    ss = torch.tensor([s for (_, _, _, s) in rnd], dtype=rdt,
+                                      device=device).unsqueeze(1)
+                    ri = C[ii]
+                    rj = C[jj]
+                    C[ii] = cc * ri - ss * rj
+                    C[jj] = ss * ri + cc * rj
+            else:
+                for r_i in range(nrot - 1, -1, -1):
+                    i_r = int(rots[r_i, 0])
+                    j_r = int(rots[r_i, 1])
+                    c_r = float(rots[r_i, 2])
+                    s_r = float(rots[r_i, 3])
+                    ri = C[i_r].clone()
+                    rj = C[j_r].clone()
+                    C[i_r] = c_r * ri - s_r * rj
+                    C[j_r] = s_r * ri + c_r * rj
             # map sorted index space -> Qb column space
             Cq = torch.zeros_like(C)
             Cq[perm.to(device)] = C

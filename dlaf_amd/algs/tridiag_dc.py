@@ -241,29 +241,38 @@ def _merge(w1, Q1, w2, Q2, rho, device):
     # per-rotation loop was 4 tiny device launches each; same class of
     # host-bound overhead as the bt prep fix, profiles/
     # bench_full_r2_kernel_stats.md).
-    rounds: list = []
-    last: dict = {}
-    for rot in rots:
-        i, j = rot[0], rot[1]
-        r = max(last.get(i, 0), last.get(j, 0))
-        if r == len(rounds):
-            rounds.append([])
-        rounds[r].append(rot)
-        last[i] = last[j] = r + 1
-    rdt = R.real.dtype if R.is_complex() else R.dtype
-    for rnd in rounds:
-        pi = torch.tensor([rot_pos[i] for (i, _, _, _) in rnd],
-                          dtype=torch.int64, device=device)
-        pj = torch.tensor([rot_pos[j] for (_, j, _, _) in rnd],
-                          dtype=torch.int64, device=device)
-        cc = torch.tensor([c for (_, _, c, _) in rnd], dtype=rdt,
-                          device=device)
-        ss = torch.tensor([s for (_, _, _, s) in rnd], dtype=rdt,
-                          device=device)
-        gi = R[:, pi]
-        gj = R[:, pj]
-        R[:, pi] = cc * gi + ss * gj
-        R[:, pj] = -ss * gi + cc * gj
+    import os as _os
+    if _os.environ.get("DLAF_DC_ROT_BATCH", "0") != "0":
+        rounds: list = []
+        last: dict = {}
+        for rot in rots:
+            i, j = rot[0], rot[1]
+            r = max(last.get(i, 0), last.get(j, 0))
+            if r == len(rounds):
+                rounds.append([])
+            rounds[r].append(rot)
+            last[i] = last[j] = r + 1
+        rdt = R.real.dtype if R.is_complex() else R.dtype
+        for rnd in rounds:
+            pi = torch.tensor([rot_pos[i] for (i, _, _, _) in rnd],
+                              dtype=torch.int64, device=device)
+            pj = torch.tensor([rot_pos[j] for (_, j, _, _) in rnd],
+                              dtype=torch.int64, device=device)
+            cc = torch.tensor([c for (_, _, c, _) in rnd], dtype=rdt,
+                              device=device)
+            ss = torch.tensor([s for (_, _, _, s) in rnd], dtype=rdt,
+                              device=device)
+            gi = R[:, pi]
+            gj = R[:, pj]
+            R[:, pi] = cc * gi + ss * gj
+            R[:, pj] = -ss * gi + cc * gj
+    else:
+        for (i, j, c, s) in rots:
+            pi_, pj_ = rot_pos[i], rot_pos[j]
+            gi = R[:, pi_].clone()
+            gj = R[:, pj_].clone()
+            R[:, pi_] = c * gi + s * gj
+            R[:, pj_] = -s * gi + c * gj
 
     def patch_rotated(out, cols_np):
         pos = [(p, rot_pos[c]) for p, c in enumerate(cols_np) if c in rot_pos]
