@@ -79,3 +79,20 @@ def test_tridiag_deterministic():
     w1, E1 = tridiagonal_eigensolver(d.clone(), e.clone(), device="cpu")
     w2, E2 = tridiagonal_eigensolver(d.clone(), e.clone(), device="cpu")
     assert torch.equal(w1, w2) and torch.equal(E1, E2)
+
+
+def test_tridiag_rot_batch_matches_sequential(monkeypatch):
+    """DLAF_DC_ROT_BATCH=1 (rounds of disjoint Givens pairs) must produce
+    bitwise the same result as the default sequential apply: the greedy
+    round scheduler only commutes provably disjoint rotations."""
+    g = torch.Generator().manual_seed(5)
+    n = 260
+    d = torch.ones(n, dtype=torch.float64)
+    d[::2] = 2.0                       # heavy pair deflation -> many rotations
+    e = 1e-3 * torch.randn(n - 1, generator=g, dtype=torch.float64).abs()
+    monkeypatch.delenv("DLAF_DC_ROT_BATCH", raising=False)
+    w1, E1 = tridiagonal_eigensolver(d.clone(), e.clone(), device="cpu")
+    monkeypatch.setenv("DLAF_DC_ROT_BATCH", "1")
+    w2, E2 = tridiagonal_eigensolver(d.clone(), e.clone(), device="cpu")
+    assert torch.equal(w1, w2)
+    assert torch.equal(E1, E2)
