@@ -8,7 +8,8 @@
 // mirror -> algorithm -> copy back, src/c_api/eigensolver/eigensolver.h:30-73).
 //
 // The package is located relative to this .so (dladdr), overridable with
-// DLAF_AMD_PYROOT. Single-process contexts only — see include/dlaf_c.h.
+// DLAF_AMD_PYROOT. Multi-process grids rendezvous via the torchrun-style
+// environment — see include/dlaf_c.h.
 
 #include <pybind11/embed.h>
 #include <pybind11/numpy.h>
@@ -43,13 +44,19 @@ struct Gil {
   py::gil_scoped_acquire acq;
 };
 
-py::object np_view(void* ptr, const struct DLAF_descriptor& d, const char* dtype) {
+py::object desc_obj(const struct DLAF_descriptor& d);
+
+py::object np_view(int ctx, void* ptr, const struct DLAF_descriptor& d,
+                   const char* dtype) {
   Gil g;
-  auto np = py::module_::import("numpy");
   py::dtype dt(dtype);
   const long isz = dt.itemsize();
-  // column-major local buffer of a 1x1 grid: full (m, n) with stride ld
-  py::array arr(dt, {(long)d.m, (long)d.n}, {isz, (long)d.ld * isz}, ptr,
+  // column-major RANK-LOCAL block-cyclic buffer (ScaLAPACK local panel):
+  // (local_m, local_n) with stride ld; for a 1x1 grid this is the full matrix
+  auto sh = capi_->attr("dlaf_local_shape")(ctx, desc_obj(d)).cast<py::tuple>();
+  const long lm = sh[0].cast<long>();
+  const long ln = sh[1].cast<long>();
+  py::array arr(dt, {lm, ln}, {isz, (long)d.ld * isz}, ptr,
                 py::str());  // base handle => non-owning view
   return arr;
 }
@@ -70,7 +77,7 @@ int run_inplace(const char* fn, int ctx, char uplo, void* a,
                 const char* extra = nullptr) {
   Gil g;
   try {
-    auto arr = np_view(a, d, dtype);
+    auto arr = np_view(ctx, a, d, dtype);
     if (extra)
       return capi_->attr(fn)(ctx, std::string(1, uplo), std::string(1, *extra),
                              arr, desc_obj(d)).cast<int>();
@@ -86,8 +93,8 @@ int run_eig(int ctx, char uplo, void* a, const struct DLAF_descriptor& da,
             const char* adt, const char* wdt, long il, long iu) {
   Gil g;
   try {
-    auto arr = np_view(a, da, adt);
-    auto zv = np_view(z, dz, adt);
+    auto arr = np_view(ctx, a, da, adt);
+    auto zv = np_view(ctx, z, dz, adt);
     auto wv = np_vec(w, da.n, wdt);
     auto kw = py::dict();
     if (iu >= 0) {
@@ -109,9 +116,9 @@ int run_geig(int ctx, char uplo, void* a, const struct DLAF_descriptor& da,
              bool factorized) {
   Gil g;
   try {
-    auto av = np_view(a, da, adt);
-    auto bv = np_view(b, db, adt);
-    auto zv = np_view(z, dz, adt);
+    auto av = np_view(ctx, a, da, adt);
+    auto bv = np_view(ctx, b, db, adt);
+    auto zv = np_view(ctx, z, dz, adt);
     auto wv = np_vec(w, da.n, wdt);
     return capi_->attr("dlaf_hermitian_generalized_eigensolver")(
         ctx, std::string(1, uplo), av, desc_obj(da), bv, desc_obj(db), wv, zv,
@@ -182,13 +189,20 @@ void dlaf_finalize(void) {
 }
 
 int dlaf_create_grid(int nprow, int npcol, char order) {
-  if (nprow != 1 || npcol != 1) {
-    std::fprintf(stderr, "dlaf_c: only 1x1 grids (see include/dlaf_c.h)\n");
+  // Multi-process grids: the Python side initializes torch.distributed
+  // (RCCL on GPU, gloo on CPU) from the torchrun-style launcher env
+  // (RANK/WORLD_SIZE/MASTER_ADDR/MASTER_PORT); nprow*npcol must equal
+  // WORLD_SIZE. Reference counterpart: src/c_api/grid.cpp (MPI_Comm in,
+  // CommunicatorGrid out) — the rendezvous here is the launcher env
+  // instead of an MPI communicator.
+  Gil g;
+  try {
+    return capi_->attr("dlaf_create_grid")(nprow, npcol, std::string(1, order))
+        .cast<int>();
+  } catch (const std::exception& e) {
+    std::fprintf(stderr, "dlaf_c: dlaf_create_grid failed: %s\n", e.what());
     return -1;
   }
-  Gil g;
-  return capi_->attr("dlaf_create_grid")(nprow, npcol, std::string(1, order))
-      .cast<int>();
 }
 
 void dlaf_free_grid(int ctx) {

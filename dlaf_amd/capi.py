@@ -55,13 +55,42 @@ def dlaf_create_grid(nprow: int, npcol: int, order: str = "R",
 
     Row-major rank order only (the reference supports both; RCCL ranks here
     are torch.distributed ranks, which this framework orders row-major).
+
+    Multi-process grids (nprow*npcol > 1): if torch.distributed is not yet
+    initialized, it is initialized here from the torchrun/launcher
+    environment (RANK / WORLD_SIZE / MASTER_ADDR / MASTER_PORT) — RCCL when
+    a GPU is visible, gloo otherwise. This is the C-ABI entry path for
+    multi-rank callers (reference: ``src/c_api/grid.cpp`` builds the
+    CommunicatorGrid from the caller's MPI_Comm; here the rendezvous is the
+    launcher environment instead of MPI).
     """
     assert order.upper().startswith("R"), "row-major rank ordering only"
+    import torch.distributed as tdist
+    if (nprow * npcol > 1 and tdist.is_available()
+            and not tdist.is_initialized()):
+        import os
+        assert "RANK" in os.environ and "WORLD_SIZE" in os.environ, (
+            "multi-process grid: launch with torchrun-style env "
+            "(RANK/WORLD_SIZE/MASTER_ADDR/MASTER_PORT)")
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        tdist.init_process_group(backend=backend)
     grid = CommGrid(nprow, npcol, device=device)
     ctx = _next_ctx[0]
     _next_ctx[0] += 1
     _grids[ctx] = grid
     return ctx
+
+
+def dlaf_local_shape(ctx: int, desc: "DLAF_descriptor"):
+    """Rank-local (rows, cols) of the block-cyclic buffer described by
+    ``desc`` on grid ``ctx`` — used by the C ABI (csrc/capi/dlaf_c.cpp) to
+    wrap the caller's LOCAL panel with the right shape on >1x1 grids."""
+    g = _grid(ctx)
+    d = Distribution(desc.m, desc.n, desc.mb, desc.nb,
+                     g.grid_rows, g.grid_cols, g.rank_row, g.rank_col,
+                     desc.isrc, desc.jsrc)
+    lm, ln = d.local_size
+    return int(lm), int(ln)
 
 
 def dlaf_free_grid(ctx: int) -> None:

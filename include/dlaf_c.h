@@ -7,11 +7,16 @@
  * caller's buffers are wrapped without copies (column-major, leading
  * dimension ld, ScaLAPACK convention).
  *
- * Scope: single-process contexts (nprow == npcol == 1) — the reference
- * builds grids from an MPI_Comm, this framework scales multi-GPU through
- * torch.distributed (one Python process per GPU over RCCL), which has no
- * meaning inside a foreign C process. Multi-rank callers use the Python
- * API (dlaf_amd.capi) under torchrun.
+ * Grids: nprow x npcol contexts. For multi-process grids launch one
+ * process per rank with the torchrun-style rendezvous environment set
+ * (RANK, WORLD_SIZE, MASTER_ADDR, MASTER_PORT); dlaf_create_grid then
+ * initializes torch.distributed inside the embedded runtime (RCCL when a
+ * GPU is visible, gloo otherwise) and nprow*npcol must equal WORLD_SIZE.
+ * The reference builds grids from an MPI_Comm / BLACS context
+ * (src/c_api/grid.cpp); this framework's rank rendezvous is the launcher
+ * environment instead of MPI — there is no MPI/BLACS interop in the
+ * MI355X-native design. Buffers are the caller's rank-local ScaLAPACK
+ * block-cyclic panels, exactly as in the reference shims.
  */
 #ifndef DLAF_C_H
 #define DLAF_C_H
@@ -34,7 +39,8 @@ typedef struct { double re, im; } dlaf_complex_z;
 int dlaf_initialize(int argc, const char* const* argv);
 void dlaf_finalize(void);
 
-/* grid management (reference: dlaf_c/grid.h; here: local 1x1 contexts) */
+/* grid management (reference: dlaf_c/grid.h); see header comment for
+ * multi-process grids */
 int dlaf_create_grid(int nprow, int npcol, char order);
 void dlaf_free_grid(int ctx);
 
