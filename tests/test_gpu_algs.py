@@ -359,3 +359,21 @@ def test_hegst_upper_native_gpu():
     want = Ui.mH @ A @ Ui
     err = (got - want).abs().max().item()
     assert err < 1e-10 * n, err
+
+
+def test_dc_rot_batch_gpu(monkeypatch):
+    """Gated DLAF_DC_ROT_BATCH=1 path on device at a modest size (clean at
+    n=8192 in validation; faults at n=20000 — docs/DESIGN.md): result must
+    bitwise-match the default sequential rotation apply."""
+    from dlaf_amd.algs.tridiag_dc import tridiagonal_eigensolver
+    g = torch.Generator().manual_seed(3)
+    n = 2048
+    d = torch.ones(n, dtype=torch.float64)
+    d[::2] = 2.0
+    e = 1e-3 * torch.randn(n - 1, generator=g, dtype=torch.float64).abs()
+    monkeypatch.delenv("DLAF_DC_ROT_BATCH", raising=False)
+    w1, E1 = tridiagonal_eigensolver(d.clone(), e.clone(), device="cuda")
+    monkeypatch.setenv("DLAF_DC_ROT_BATCH", "1")
+    w2, E2 = tridiagonal_eigensolver(d.clone(), e.clone(), device="cuda")
+    assert torch.equal(w1, w2)
+    assert torch.equal(E1, E2)
