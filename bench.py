@@ -229,6 +229,10 @@ def main():
         }))
 
     if world_size > 1:
+        # drain before teardown: destroying while a peer is still inside its
+        # last collective is a known gloo/nccl shutdown race (same fix as
+        # tests/dist_utils.py)
+        dist.barrier()
         dist.destroy_process_group()
 
 
