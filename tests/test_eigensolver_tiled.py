@@ -163,3 +163,19 @@ def test_tiled_eigensolver_8rank_memory():
     w, E, A = outs[0]
     _check_eig(torch.from_numpy(A), torch.from_numpy(w),
                torch.from_numpy(E), 5e-12 * n)
+
+
+@pytest.mark.parametrize("dtype_name,tol", [("float32", 5e-4),
+                                            ("complex64", 5e-4)])
+def test_tiled_eigensolver_dist_single_precision(dtype_name, tol):
+    """Single-precision dtypes through the full tiled distributed pipeline
+    (the reference ETI covers all four scalar types)."""
+    n, nb, band = 128, 32, 16
+    outs = run_distributed(_run_tiled, 2,
+                           args=(n, nb, band, 1, 2, dtype_name, 0, None))
+    w0, E0, A0 = outs[0]
+    for w, E, _ in outs[1:]:
+        np.testing.assert_allclose(w, w0, rtol=0, atol=1e-6)
+        np.testing.assert_allclose(E, E0, rtol=0, atol=1e-6)
+    _check_eig(torch.from_numpy(A0), torch.from_numpy(w0),
+               torch.from_numpy(E0), tol * n)
