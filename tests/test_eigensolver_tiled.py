@@ -179,3 +179,29 @@ def test_tiled_eigensolver_dist_single_precision(dtype_name, tol):
         np.testing.assert_allclose(E, E0, rtol=0, atol=1e-6)
     _check_eig(torch.from_numpy(A0), torch.from_numpy(w0),
                torch.from_numpy(E0), tol * n)
+
+
+def _run_replicated_fallback(rank, world, n, nb, band):
+    import os
+    os.environ["DLAF_DIST_EIG"] = "replicated"
+    try:
+        from dlaf_amd.algs.eigensolver import hermitian_eigensolver
+        grid = CommGrid(1, world, device=torch.device("cpu"))
+        mat = _make_mat(n, nb, grid, torch.float64, seed=21)
+        A = _herm(mat.to_global())
+        w, evecs = hermitian_eigensolver(UpLo.Lower, mat, grid, band=band)
+        return w.numpy(), evecs.to_global().numpy(), A.numpy()
+    finally:
+        del os.environ["DLAF_DIST_EIG"]
+
+
+def test_replicated_dist_fallback():
+    """DLAF_DIST_EIG=replicated keeps the round-1 replicated-dense design
+    working as a debug fallback (eigensolver.py dispatch)."""
+    n, nb, band = 96, 32, 16
+    outs = run_distributed(_run_replicated_fallback, 2, args=(n, nb, band))
+    w0, E0, A0 = outs[0]
+    for w, E, _ in outs[1:]:
+        np.testing.assert_allclose(w, w0, rtol=0, atol=1e-12)
+    _check_eig(torch.from_numpy(A0), torch.from_numpy(w0),
+               torch.from_numpy(E0), 5e-12 * n)
