@@ -277,8 +277,9 @@ extern "C" {
         &per_cu, reinterpret_cast<const void*>(&panel_qr_kernel<T>), threads, \
         0);                                                                   \
     int blocks = prop.multiProcessorCount * (per_cu > 0 ? per_cu : 1);        \
-    /* cooperative grid.sync cost grows with the WG count (measured ~40us at \
-       256 WGs); the phase-B work is small, so favour fewer, fuller WGs */    \
+    /* cap chosen when the cg grid.sync cost grew with WG count; the agent  \
+       counter barrier is cheap at 256 WGs so a higher cap may now win —    \
+       tunable via DLAF_PANEL_QR_BLOCKS (256 untested at scale)             */ \
     static const int cap = [] {                                               \
       const char* v = getenv("DLAF_PANEL_QR_BLOCKS");                         \
       return v ? atoi(v) : 128;                                               \
