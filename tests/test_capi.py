@@ -139,3 +139,44 @@ def test_capi_partial_spectrum():
         r = np.abs(a0 @ z[:, j] - w[j] * z[:, j]).max()
         assert r < 1e-9 * n
     capi.dlaf_free_grid(ctx)
+
+
+def test_capi_potrf_upper():
+    """uplo='U' routed through the C-API entry points (advisor finding 5)."""
+    n, nb = 24, 8
+    rng = np.random.default_rng(9)
+    a = rng.standard_normal((n, n))
+    a = a @ a.T + n * np.eye(n)
+    loc = a.copy()
+    ctx = capi.dlaf_create_grid(1, 1)
+    info = capi.dlaf_cholesky_factorization(
+        ctx, "U", loc, DLAF_descriptor(n, n, nb, nb))
+    assert info == 0
+    U = np.triu(loc)
+    assert np.abs(U.T @ U - a).max() < 1e-11 * n
+    # POTRI from the Upper factor
+    info = capi.dlaf_inverse_from_cholesky_factor(
+        ctx, "U", loc, DLAF_descriptor(n, n, nb, nb))
+    capi.dlaf_free_grid(ctx)
+    assert info == 0
+    x = np.triu(loc) + np.triu(loc, 1).T
+    assert np.abs(a @ x - np.eye(n)).max() < 1e-9 * n
+
+
+def test_capi_heevd_upper():
+    n, nb = 32, 8
+    rng = np.random.default_rng(10)
+    a = rng.standard_normal((n, n)) + 1j * rng.standard_normal((n, n))
+    a = (a + a.conj().T) / 2
+    loc = np.asfortranarray(a.copy())
+    w = np.zeros(n)
+    z = np.zeros((n, n), dtype=np.complex128, order="F")
+    ctx = capi.dlaf_create_grid(1, 1)
+    info = capi.dlaf_hermitian_eigensolver(
+        ctx, "U", loc, DLAF_descriptor(n, n, nb, nb), w, z,
+        DLAF_descriptor(n, n, nb, nb))
+    capi.dlaf_free_grid(ctx)
+    assert info == 0
+    wref = np.linalg.eigvalsh(a)
+    assert np.abs(np.sort(w) - wref).max() < 1e-11 * n
+    assert np.abs(a @ z - z @ np.diag(w)).max() < 1e-10 * n
