@@ -73,3 +73,19 @@ def test_miniapp_smoke_all(app):
         assert "CSVData-2" in out or "no communication" in out, out[-2000:]
     else:
         assert "CSVData-2" in out, out[-2000:]
+
+
+def test_miniapp_cholesky_2rank():
+    """Distributed miniapp harness path (grid 1x2, gloo) through torchrun —
+    the reference miniapps run under MPI the same way."""
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29581",
+           os.path.join(root, "miniapp", "miniapp_cholesky.py"),
+           "-m", "256", "-b", "64", "--grid-rows", "1", "--grid-cols", "2",
+           "--nruns", "1", "--nwarmups", "0", "--backend", "mc",
+           "--csv-output", "--check-result", "last"]
+    out = subprocess.run(cmd, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "CSVData-2" in out.stdout and "check residual" in out.stdout
