@@ -114,3 +114,19 @@ def test_cholesky_entry_assert():
     m = Matrix.create(12, 16, 4, 4, dtype=torch.float64)
     with _pytest.raises(DlafAssertError):
         cholesky_factorization(UpLo.Lower, m)
+
+
+def test_timer_and_trace_range():
+    """utils.Timer laps/report and the rocTX trace_range scope (no-op on
+    CPU builds) — reference common/timer.h + pika instrumentation analog."""
+    from dlaf_amd.utils import Timer, trace_range
+    t = Timer()
+    with trace_range("unit-test-range"):
+        x = sum(range(1000))
+    assert x == 499500
+    d1 = t.lap("phase1")
+    assert d1 >= 0.0
+    t.lap("phase2")
+    rep = t.report()
+    assert "phase1" in rep and "phase2" in rep and "total" in rep
+    assert t.elapsed() >= 0.0
