@@ -205,3 +205,38 @@ def test_replicated_dist_fallback():
         np.testing.assert_allclose(w, w0, rtol=0, atol=1e-12)
     _check_eig(torch.from_numpy(A0), torch.from_numpy(w0),
                torch.from_numpy(E0), 5e-12 * n)
+
+
+def test_tiled_eigensolver_uneven_edges():
+    """n not divisible by nb, 3 ranks (uneven stripes), partial slice
+    crossing stripe boundaries."""
+    n, nb, band = 150, 32, 16
+    outs = run_distributed(_run_tiled, 3,
+                           args=(n, nb, band, 1, 3, "float64", 30, 120))
+    w, E, A = outs[0]
+    A = torch.from_numpy(A)
+    wf = torch.linalg.eigvalsh(A)
+    np.testing.assert_allclose(w, wf[30:120].numpy(), atol=1e-10)
+    Et = torch.from_numpy(E)
+    R = A @ Et - Et @ torch.diag(torch.from_numpy(w))
+    assert float(R.abs().max()) <= 1e-10 * n
+
+
+def _run_band_snap(rank, world):
+    # band request 24 with nb=32 -> dispatcher snaps to a divisor of nb
+    from dlaf_amd.algs.eigensolver import hermitian_eigensolver
+    grid = CommGrid(1, world, device=torch.device("cpu"))
+    n, nb = 96, 32
+    mat = _make_mat(n, nb, grid, torch.float64, seed=31)
+    A = _herm(mat.to_global())
+    w, evecs = hermitian_eigensolver(UpLo.Lower, mat, grid, band=24)
+    return w.numpy(), evecs.to_global().numpy(), A.numpy()
+
+
+def test_tiled_band_snap_dispatch():
+    """Distributed dispatch snaps a non-divisor band to the largest divisor
+    of nb (eigensolver.py) instead of a dense cliff."""
+    outs = run_distributed(_run_band_snap, 2)
+    w, E, A = outs[0]
+    _check_eig(torch.from_numpy(A), torch.from_numpy(w),
+               torch.from_numpy(E), 5e-12 * 96)
