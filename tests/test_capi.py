@@ -217,3 +217,11 @@ def test_capi_scalapack_shims_complex_and_generalized():
     assert info == 0
     assert np.abs(np.tril(a) @ l0 - np.eye(n)).max() < 1e-10 * n
     capi.dlaf_free_grid(ctx)
+
+
+def test_capi_local_shape():
+    """dlaf_local_shape: rank-local block-cyclic dims used by the C ABI."""
+    ctx = capi.dlaf_create_grid(1, 1)
+    lm, ln = capi.dlaf_local_shape(ctx, DLAF_descriptor(100, 70, 32, 16))
+    capi.dlaf_free_grid(ctx)
+    assert (lm, ln) == (100, 70)  # 1x1 grid: local == global

@@ -89,3 +89,11 @@ def test_miniapp_cholesky_2rank():
     out = subprocess.run(cmd, capture_output=True, text=True, timeout=300)
     assert out.returncode == 0, out.stderr[-2000:]
     assert "CSVData-2" in out.stdout and "check residual" in out.stdout
+
+
+def test_miniapp_complex_type():
+    """--type z through the shared harness (reference dispatchMiniapp dtype
+    axis)."""
+    out = _run_miniapp("miniapp_eigensolver.py",
+                       ["--type", "z", "--check-result", "last"])
+    assert "CSVData-2" in out and "check residual" in out
