@@ -97,3 +97,31 @@ def test_miniapp_complex_type():
     out = _run_miniapp("miniapp_eigensolver.py",
                        ["--type", "z", "--check-result", "last"])
     assert "CSVData-2" in out and "check residual" in out
+
+
+def test_usage_doc_api_sequence():
+    """The docs/USAGE.md Python-API sequence, shrunk to CPU sizes — keeps
+    the documented entry points honest."""
+    import torch
+    from dlaf_amd import (Matrix, UpLo, Side, Op, Diag,
+                          cholesky_factorization, triangular_solver,
+                          hermitian_eigensolver)
+    from dlaf_amd.matrix import util as mutil
+    A = Matrix.create(96, 96, 32, 32, dtype=torch.float64, device="cpu")
+    mutil.set_random_hermitian_positive_definite(A, seed=0)
+    cholesky_factorization(UpLo.Lower, A)
+    B = Matrix.create(96, 16, 32, 32, dtype=torch.float64, device="cpu")
+    mutil.set_random(B, seed=1)
+    triangular_solver(Side.Left, UpLo.Lower, Op.NoTrans, Diag.NonUnit,
+                      1.0, A, B)
+    H = Matrix.create(80, 80, 32, 32, dtype=torch.float64, device="cpu")
+    mutil.set_random_hermitian(H, seed=2)
+    w, E = hermitian_eigensolver(UpLo.Lower, H)
+    assert w.shape == (80,)
+    H2 = Matrix.create(80, 80, 32, 32, dtype=torch.float64, device="cpu")
+    mutil.set_random_hermitian(H2, seed=2)
+    w10, E10 = hermitian_eigensolver(UpLo.Lower, H2,
+                                     eigenvalues_index_begin=0,
+                                     eigenvalues_index_end=10)
+    assert w10.shape == (10,)
+    assert torch.allclose(w10, w[:10], atol=1e-10)
